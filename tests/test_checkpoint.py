@@ -1,0 +1,132 @@
+"""Checkpoint IO tests: legacy-pickle format, reference .pth compatibility."""
+
+import os
+import pickletools
+import sys
+
+import pytest
+import torch
+
+from tskd_amd.models import (MyCNN2, MyCNN4, MyCNN5, build_model,
+                             load_checkpoint, save_checkpoint)
+
+
+def _roundtrip(tmp_path, model, name="m.pth"):
+    p = os.path.join(str(tmp_path), name)
+    save_checkpoint(model, p)
+    return p, load_checkpoint(p)
+
+
+class TestRoundTrip:
+    @pytest.mark.parametrize("variant", ["MyCNN2", "MyCNN4", "MyCNN5"])
+    def test_state_equal(self, tmp_path, variant):
+        m = build_model(variant).eval()
+        _, m2 = _roundtrip(tmp_path, m)
+        assert type(m2).__name__ == type(m).__name__
+        for (k, a), (k2, b) in zip(m.state_dict().items(), m2.state_dict().items()):
+            assert k == k2
+            assert torch.equal(a, b)
+
+    def test_forward_equal(self, tmp_path):
+        m = MyCNN5().eval()
+        _, m2 = _roundtrip(tmp_path, m)
+        x = torch.randn(8, 10, 120, generator=torch.Generator().manual_seed(3))
+        age = torch.full((8,), 65.0)
+        with torch.no_grad():
+            assert torch.equal(m(x, age), m2(x, age))
+
+    def test_save_does_not_mutate_class(self, tmp_path):
+        m = MyCNN5()
+        save_checkpoint(m, os.path.join(str(tmp_path), "x.pth"))
+        assert m.__class__ is MyCNN5
+
+    def test_no_main_pollution(self, tmp_path):
+        main = sys.modules["__main__"]
+        had = hasattr(main, "MyCNN")
+        before = getattr(main, "MyCNN", None)
+        _roundtrip(tmp_path, MyCNN5())
+        assert hasattr(main, "MyCNN") == had
+        assert getattr(main, "MyCNN", None) is before
+
+
+class TestFormat:
+    def test_legacy_nonzip_format(self, tmp_path):
+        """The file must be a legacy (non-zipfile) torch pickle whose model
+        GLOBAL is __main__.MyCNN — exactly what the reference writes
+        (explore_torch.ipynb cell 26) and loads (predictStream.py:36)."""
+        p, _ = _roundtrip(tmp_path, MyCNN5())
+        blob = open(p, "rb").read()
+        assert blob[:2] != b"PK"  # not zipfile serialization
+        # Legacy torch files are several concatenated pickle streams (magic,
+        # protocol, sys_info, then the module pickle) + raw storage payload.
+        globals_seen = set()
+        pos = 0
+        for _ in range(4):
+            ops = []
+            try:
+                for op, arg, opos in pickletools.genops(blob[pos:]):
+                    ops.append((op.name, arg, opos))
+            except Exception:
+                break
+            if not ops:
+                break
+            globals_seen |= {a for (n, a, _p) in ops if n in ("GLOBAL", "STACK_GLOBAL") and a}
+            stops = [p for (n, _a, p) in ops if n == "STOP"]
+            if not stops:
+                break
+            pos += stops[0] + 1
+        assert any("MyCNN" in g and "__main__" in g for g in globals_seen), globals_seen
+
+    def test_float32_storages(self, tmp_path):
+        p, m2 = _roundtrip(tmp_path, MyCNN5())
+        assert all(v.dtype == torch.float32 for v in m2.state_dict().values())
+
+
+class TestReferenceCheckpoints:
+    """Loads the real upstream .pth files when /root/reference is mounted."""
+
+    EXPECT = {
+        "MyCNN2.pth": ("MyCNN2", 5240, (4, 7, 5)),
+        "MyCNN3.pth": ("MyCNN3", 5240, (4, 7, 5)),
+        "MyCNN4.pth": ("MyCNN4", 5300, (4, 10, 5)),
+        "MyCNN5.pth": ("MyCNN5", 5957, (4, 10, 10)),
+    }
+
+    @pytest.mark.parametrize("fname", sorted(EXPECT))
+    def test_load_reference(self, reference_dir, fname):
+        path = os.path.join(reference_dir, "model", fname)
+        if not os.path.exists(path):
+            pytest.skip(f"{fname} not present")
+        m = load_checkpoint(path)
+        want_cls, want_params, want_conv1 = self.EXPECT[fname]
+        # MyCNN2 and MyCNN3 share an architecture; class detection by shape
+        # cannot distinguish them — accept the arch class.
+        got = type(m).__name__
+        assert got in (want_cls, "MyCNN2") if want_cls == "MyCNN3" else got == want_cls
+        n_params = sum(p.numel() for p in m.parameters())
+        assert n_params == want_params
+        assert tuple(m.conv1.weight.shape) == want_conv1
+        assert not m.training  # loader puts model in eval mode
+
+    def test_mycnn5_forward_runs(self, reference_dir):
+        path = os.path.join(reference_dir, "model", "MyCNN5.pth")
+        if not os.path.exists(path):
+            pytest.skip("MyCNN5.pth not present")
+        m = load_checkpoint(path)
+        x = torch.randn(4, 10, 120, generator=torch.Generator().manual_seed(0))
+        with torch.no_grad():
+            y = m(x, torch.full((4,), 65.0))
+        assert y.shape == (4,)
+        assert torch.isfinite(y).all()
+
+    def test_roundtrip_reference_mycnn5(self, reference_dir, tmp_path):
+        """reference .pth -> our loader -> our saver -> our loader: bit-equal."""
+        path = os.path.join(reference_dir, "model", "MyCNN5.pth")
+        if not os.path.exists(path):
+            pytest.skip("MyCNN5.pth not present")
+        m = load_checkpoint(path)
+        p2 = os.path.join(str(tmp_path), "resaved.pth")
+        save_checkpoint(m, p2)
+        m2 = load_checkpoint(p2)
+        for a, b in zip(m.state_dict().values(), m2.state_dict().values()):
+            assert torch.equal(a, b)
