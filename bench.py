@@ -35,7 +35,7 @@ def main() -> None:
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=50)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--seqs", type=int, default=512,
+    p.add_argument("--seqs", type=int, default=2048,
                    help="concurrent patient sequences per GPU")
     p.add_argument("--batch", type=int, default=1024,
                    help="windows per sequence batch (reference batch semantics)")

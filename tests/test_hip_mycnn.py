@@ -45,14 +45,47 @@ class TestHipForward:
         got = eng.forward(x.cuda()).cpu()
         torch.testing.assert_close(got, ref, rtol=2e-4, atol=2e-5)
 
-    def test_bf16_input(self):
+    def test_mfma_fragment_map(self):
+        """Ground-truth check of the v_mfma_f32_16x16x32_bf16 A/B/C lane maps
+        (asymmetric operands so a transposed map cannot pass — §5.4 rule 16)."""
+        import ctypes
+        from tskd_amd.ops import _load_lib, _stream_ptr
+        lib = _load_lib()
+        g = torch.Generator().manual_seed(11)
+        A = (torch.randn(16, 32, generator=g) * 0.5).to(torch.bfloat16)
+        B = (torch.randn(32, 16, generator=g) * 0.5 +
+             torch.arange(16).float() * 0.01).to(torch.bfloat16)
+        Ad, Bd = A.cuda(), B.cuda()
+        Cd = torch.empty(16, 16, dtype=torch.float32, device="cuda")
+        rc = lib.tskd_debug_mfma16x16x32(
+            ctypes.c_void_p(Ad.data_ptr()), ctypes.c_void_p(Bd.data_ptr()),
+            ctypes.c_void_p(Cd.data_ptr()), _stream_ptr())
+        assert rc == 0
+        torch.cuda.synchronize()
+        ref = A.float() @ B.float()
+        torch.testing.assert_close(Cd.cpu(), ref, rtol=2e-2, atol=2e-2)
+
+    def test_bf16_input_mfma_conv(self):
+        # bf16 path runs conv1 on MFMA (im2col-GEMM); weights + inputs are
+        # bf16-rounded, accumulation fp32.
         m = build_model("MyCNN5").eval()
         eng = MyCNNEngine(m, device="cuda")
         x = _x(2, 32, 10, seed=2, dtype=torch.bfloat16)
-        # Oracle sees the same bf16-rounded windows.
+        # Oracle sees the same bf16-rounded windows (fp32 weights).
         ref = _oracle(m, x.float())
         got = eng.forward(x.cuda()).cpu()
-        torch.testing.assert_close(got, ref, rtol=3e-3, atol=3e-3)
+        torch.testing.assert_close(got, ref, rtol=2e-2, atol=2e-2)
+
+    @pytest.mark.parametrize("variant", ["MyCNN5", "MyCNN2", "MyCNN4"])
+    def test_bf16_conv_features_all_variants(self, variant):
+        m = build_model(variant).eval()
+        eng = MyCNNEngine(m, device="cuda")
+        x = _x(1, 16, m.IN_CHANNELS, seed=9, dtype=torch.bfloat16)
+        with torch.no_grad():
+            f = m.pool(torch.tanh(m.conv2(m.pool(torch.tanh(m.conv1(x[0].float()))))))
+            ref = f.view(16, -1)
+        got = eng.conv_features(x.cuda()).reshape(16, -1).cpu()
+        torch.testing.assert_close(got, ref, rtol=2e-2, atol=2e-2)
 
     def test_long_sequence_accumulation(self):
         # The LSTM scan is sequential over N=1024: accumulated fp error must

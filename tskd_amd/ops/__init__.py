@@ -15,7 +15,8 @@ from typing import Optional
 import torch
 
 from tskd_amd.ops import build as _build
-from tskd_amd.ops.pack import VARIANT_IDS, pack_weights  # noqa: F401
+from tskd_amd.ops.pack import (VARIANT_IDS, pack_conv1_mfma_bfrags,  # noqa: F401
+                               pack_weights)
 
 _lib: Optional[ctypes.CDLL] = None
 _lib_err: Optional[str] = None
@@ -36,7 +37,11 @@ def _load_lib() -> ctypes.CDLL:
     lib.tskd_conv_fwd.restype = ctypes.c_int
     lib.tskd_conv_fwd.argtypes = [
         ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p,
-        ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
+    ]
+    lib.tskd_debug_mfma16x16x32.restype = ctypes.c_int
+    lib.tskd_debug_mfma16x16x32.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
     ]
     lib.tskd_lstm_head_fwd.restype = ctypes.c_int
     lib.tskd_lstm_head_fwd.argtypes = [
@@ -82,6 +87,7 @@ class MyCNNEngine:
         self.device = torch.device(device)
         self.model = model  # CPU oracle / fallback
         self.wpack = pack_weights(model).to(self.device)
+        self.bfrag = pack_conv1_mfma_bfrags(model).to(self.device)
         if self.device.type == "cuda":
             lib = _load_lib()  # raises loudly if the HIP ext is unavailable
             expect = lib.tskd_pack_size(self.variant)
@@ -112,7 +118,8 @@ class MyCNNEngine:
         rc = lib.tskd_conv_fwd(
             ctypes.c_void_p(xf.data_ptr()), is_bf16,
             ctypes.c_void_p(feat.data_ptr()),
-            ctypes.c_void_p(self.wpack.data_ptr()), sn, self.variant,
+            ctypes.c_void_p(self.wpack.data_ptr()),
+            ctypes.c_void_p(self.bfrag.data_ptr()), sn, self.variant,
             _stream_ptr())
         if rc != 0:
             raise RuntimeError(f"tskd_conv_fwd failed: hipError {rc}")

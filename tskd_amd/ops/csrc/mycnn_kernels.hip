@@ -3,32 +3,44 @@
 // Replaces the reference's PyTorch ATen CPU ops (reference bin/models.py:22-36;
 // SURVEY.md §2.6 K1-K9) with two fused kernels:
 //
-//   1. conv_stack_kernel:  (SN, CIN, 120) windows -> (SN, LIN) features.
+//   1. conv_stack_*_kernel: (SN, CIN, 120) windows -> (SN, LIN) features.
 //      Fuses Conv1d(CIN,4,K1) + tanh + MaxPool1d(PK,PS) + Conv1d(4,1,5) +
-//      tanh + pool into ONE kernel: one 64-lane wavefront per window, the
-//      window and all intermediates staged in LDS, weights staged in LDS
-//      once per workgroup. Dropout sites are identity in eval (K5).
+//      tanh + pool into ONE kernel: one 64-lane wavefront per window.
+//      Two compute paths:
+//        - bf16 input (the serving path): conv1 as an MFMA im2col-GEMM.
+//          The window is staged TRANSPOSED in LDS as bf16 (xt[t][i]), which
+//          makes the im2col matrix A[s][kk] (kk = k*CIN + i) a simple
+//          shifted view xt_flat[s*CIN + kk] — A fragments are contiguous
+//          LDS reads, no gather. B = conv weights pre-packed host-side in
+//          MFMA fragment order (4 used columns of a 16-col tile); the
+//          v_mfma_f32_16x16x32_bf16 accumulator chains over KK/32 k-steps,
+//          epilogue adds bias + tanh and lands rows in LDS for the pool.
+//        - fp32 input (exact-numerics path): direct VALU cross-correlation.
+//      Dropout sites are identity in eval (K5).
 //
-//   2. lstm_head_kernel:   (S, N, LIN) features -> (S, N) logits/probs.
+//   2. lstm_head_kernel: (S, N, LIN) features -> (S, N) logits/probs.
 //      Fuses the 2-layer LSTM(LIN,16) *batch-axis-as-time* scan (the
 //      reference's 2-D-input quirk: hidden state flows across the N windows
-//      of a batch — SURVEY.md §2.3) with the Linear(16,1) head, the
-//      age gate relu(age*eps+1) and optional sigmoid. One wavefront per
-//      sequence: the 64 lanes are the 64 LSTM gate-units (4 gates x 16
-//      hidden), per-lane weight rows live in VGPRs, h/c in registers,
-//      cross-lane traffic via __shfl only; features prefetched into LDS in
-//      CHUNK-step blocks so the sequential scan is never global-latency
-//      bound.
+//      of a batch — SURVEY.md §2.3) with the Linear(16,1) head, the age gate
+//      relu(age*eps+1) and optional sigmoid. One wavefront per sequence:
+//      the 64 lanes are the 64 gate-unit rows (4 gates x 16 hidden),
+//      per-lane weight rows live in VGPRs, the full h vectors of both layers
+//      are collected into per-lane registers with 16 INDEPENDENT shuffles
+//      per layer (dependency-chain broken: the serial-scan critical path is
+//      ~2 shuffle round-trips per layer, not 32), features prefetched into
+//      LDS in CHUNK-step blocks.
 //
-// Numerics: fp32 accumulation throughout; input windows bf16 or fp32.
-// Wavefront size is 64 on CDNA4 (not 32) — all lane math below assumes it.
+// Numerics: fp32 accumulation throughout (MFMA accumulator is fp32).
+// Wavefront size is 64 on CDNA4 (not 32) — all lane math assumes it.
 
 #include <hip/hip_runtime.h>
-#include <hip/hip_bf16.h>
 
 #define WAVE 64
 #define WG_WAVES 4
 #define WG_THREADS (WAVE * WG_WAVES)
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 // ---------------------------------------------------------------------------
 // Geometry per model variant (lengths from SURVEY.md §2.3 forward semantics)
@@ -44,6 +56,13 @@ struct Geom {
     static constexpr int P1  = (C1 - PK) / PS + 1;    // pool1 out len
     static constexpr int C2  = P1 - 5 + 1;            // conv2 out len (k=5)
     static constexpr int LIN = (C2 - PK) / PS + 1;    // pool2 out len = LSTM in
+    // MFMA im2col geometry
+    static constexpr int KK     = CIN * K1;           // reduction length
+    static constexpr int KSTEPS = (KK + 31) / 32;     // mfma k-steps (K=32)
+    static constexpr int MTILES = (C1 + 15) / 16;     // 16-row output tiles
+    static constexpr int XTN    = CIN * L;            // transposed window elems
+    static constexpr int XTSZ   = ((MTILES * 16 - 1) * CIN + KSTEPS * 32 + 7)
+                                  / 8 * 8;            // padded LDS extent
     // fp32 weight-pack offsets (must match tskd_amd/ops/pack.py)
     static constexpr int OW1   = 0;                   // [4][CIN][K1]
     static constexpr int OB1   = OW1 + 4 * CIN * K1;  // [4]
@@ -66,6 +85,7 @@ using GeomCNN4 = Geom<10, 5, 2, 2>;   // MyCNN4: same lens as CNN2, 10 ch
 
 static_assert(GeomCNN5::LIN == 25, "MyCNN5 feature length must be 25");
 static_assert(GeomCNN2::LIN == 27, "MyCNN2 feature length must be 27");
+static_assert(GeomCNN5::KSTEPS == 4 && GeomCNN2::KSTEPS == 2, "");
 
 // Same-wave LDS read-after-write fence: LDS ops of one wave complete in
 // order once lgkmcnt drains; the asm "memory" clobber stops compiler
@@ -91,54 +111,81 @@ __device__ __forceinline__ float tanhf_(float x) {
 }
 
 // ---------------------------------------------------------------------------
-// Kernel 1: fused conv stack (K1+K2+K3+K4 of SURVEY.md §2.6)
+// Shared epilogue: pool1 -> conv2 -> tanh -> pool2 -> feature store.
+// Reads lds_c1 (4 x C1 tanh'd conv1 rows), uses lds_p1/lds_c2 scratch.
 // ---------------------------------------------------------------------------
-// One wave per window; WG_WAVES windows per workgroup; grid-stride over SN.
-// DT: input element (unsigned short = bf16 bits, or float).
-template <class G, class DT>
+template <class G>
+__device__ __forceinline__ void conv_tail(
+    int lane, const float* __restrict__ lds_w, float* c1w, float* p1w,
+    float* c2w, float* __restrict__ feat, long win)
+{
+    const float* w2 = lds_w + G::OW2;
+    const float  b2 = lds_w[G::OB2];
+    // pool1 (PK, PS)
+    for (int o = lane; o < 4 * G::P1; o += WAVE) {
+        const int c = o / G::P1, q = o % G::P1;
+        const float* src = c1w + c * G::C1 + q * G::PS;
+        float m = src[0];
+        #pragma unroll
+        for (int k = 1; k < G::PK; ++k) m = fmaxf(m, src[k]);
+        p1w[o] = m;
+    }
+    wave_sync();
+    // conv2 + tanh (k=5, 4 in-ch)
+    for (int s = lane; s < G::C2; s += WAVE) {
+        float acc = b2;
+        #pragma unroll
+        for (int c = 0; c < 4; ++c) {
+            const float* pr = p1w + c * G::P1 + s;
+            #pragma unroll
+            for (int k = 0; k < 5; ++k)
+                acc = fmaf(w2[c * 5 + k], pr[k], acc);
+        }
+        c2w[s] = tanhf_(acc);
+    }
+    wave_sync();
+    // pool2 -> feature vector
+    for (int q = lane; q < G::LIN; q += WAVE) {
+        const float* src = c2w + q * G::PS;
+        float m = src[0];
+        #pragma unroll
+        for (int k = 1; k < G::PK; ++k) m = fmaxf(m, src[k]);
+        feat[win * G::LIN + q] = m;
+    }
+    wave_sync();  // before the next window overwrites scratch
+}
+
+// ---------------------------------------------------------------------------
+// Kernel 1a (fp32 path): direct VALU conv stack — exact fp32 numerics.
+// ---------------------------------------------------------------------------
+template <class G>
 __global__ __launch_bounds__(WG_THREADS) void conv_stack_kernel(
-    const DT* __restrict__ x,    // (SN, CIN, L)
-    float* __restrict__ feat,    // (SN, LIN)
+    const float* __restrict__ x,  // (SN, CIN, L)
+    float* __restrict__ feat,     // (SN, LIN)
     const float* __restrict__ wpack,
     int SN)
 {
-    constexpr int NW = 4 * G::CIN * G::K1 + 4 + 20 + 1;  // conv weights+biases
+    constexpr int NW = 4 * G::CIN * G::K1 + 4 + 20 + 1;
     __shared__ float lds_w[NW];
     __shared__ float lds_x[WG_WAVES][G::CIN * G::L];
     __shared__ float lds_c1[WG_WAVES][4 * G::C1];
     __shared__ float lds_p1[WG_WAVES][4 * G::P1];
     __shared__ float lds_c2[WG_WAVES][G::C2];
 
-    // Stage conv weights once per workgroup.
-    for (int i = threadIdx.x; i < NW; i += WG_THREADS)
-        lds_w[i] = wpack[i];
+    for (int i = threadIdx.x; i < NW; i += WG_THREADS) lds_w[i] = wpack[i];
     __syncthreads();
 
     const float* w1 = lds_w + G::OW1;
     const float* b1 = lds_w + G::OB1;
-    const float* w2 = lds_w + G::OW2;
-    const float  b2 = lds_w[G::OB2];
-
     const int wave = threadIdx.x / WAVE;
     const int lane = threadIdx.x % WAVE;
-    float* xw  = lds_x[wave];
-    float* c1w = lds_c1[wave];
-    float* p1w = lds_p1[wave];
-    float* c2w = lds_c2[wave];
+    float* xw = lds_x[wave];
 
-    for (int win = blockIdx.x * WG_WAVES + wave; win < SN;
-         win += gridDim.x * WG_WAVES) {
-        // --- stage window into LDS as fp32 ---
-        const DT* xin = x + (long)win * (G::CIN * G::L);
-        for (int i = lane; i < G::CIN * G::L; i += WAVE) {
-            if constexpr (sizeof(DT) == 2)
-                xw[i] = bf16_to_f32((unsigned short)xin[i]);
-            else
-                xw[i] = (float)xin[i];
-        }
+    for (long win = blockIdx.x * WG_WAVES + wave; win < SN;
+         win += (long)gridDim.x * WG_WAVES) {
+        const float* xin = x + win * (G::CIN * G::L);
+        for (int i = lane; i < G::CIN * G::L; i += WAVE) xw[i] = xin[i];
         wave_sync();
-
-        // --- conv1 + tanh: 4 x C1 outputs ---
         for (int o = lane; o < 4 * G::C1; o += WAVE) {
             const int c = o / G::C1, s = o % G::C1;
             float acc = b1[c];
@@ -150,45 +197,115 @@ __global__ __launch_bounds__(WG_THREADS) void conv_stack_kernel(
                 for (int k = 0; k < G::K1; ++k)
                     acc = fmaf(wr[i * G::K1 + k], xr[k], acc);
             }
-            c1w[o] = tanhf_(acc);
+            lds_c1[wave][o] = tanhf_(acc);
         }
         wave_sync();
+        conv_tail<G>(lane, lds_w, lds_c1[wave], lds_p1[wave], lds_c2[wave],
+                     feat, win);
+    }
+}
 
-        // --- pool1 (PK, PS) ---
-        for (int o = lane; o < 4 * G::P1; o += WAVE) {
-            const int c = o / G::P1, q = o % G::P1;
-            const float* src = c1w + c * G::C1 + q * G::PS;
-            float m = src[0];
+// ---------------------------------------------------------------------------
+// Kernel 1b (bf16 path): conv1 as MFMA im2col-GEMM (16x16x32 bf16).
+//
+// Fragment maps (gfx950 v_mfma_f32_16x16x32_bf16; verified on-device by
+// tskd_debug_mfma16x16x32 + tests/test_hip_mycnn.py):
+//   A: lane l holds A[row = l&15][k = (l>>4)*8 + j], j = 0..7
+//   B: lane l holds B[k = (l>>4)*8 + j][col = l&15]
+//   C: lane l holds C[row = (l>>4)*4 + r][col = l&15], r = 0..3
+// ---------------------------------------------------------------------------
+template <class G>
+__global__ __launch_bounds__(WG_THREADS) void conv_stack_mfma_kernel(
+    const unsigned short* __restrict__ x,   // (SN, CIN, L) bf16 bits
+    float* __restrict__ feat,               // (SN, LIN)
+    const float* __restrict__ wpack,
+    const unsigned short* __restrict__ bfrag,  // [KSTEPS][64][8] bf16 bits
+    int SN)
+{
+    constexpr int NW = 4 * G::CIN * G::K1 + 4 + 20 + 1;
+    __shared__ float lds_w[NW];
+    __shared__ unsigned short lds_xt[WG_WAVES][G::XTSZ];  // transposed window
+    __shared__ float lds_c1[WG_WAVES][4 * G::C1];
+    __shared__ float lds_p1[WG_WAVES][4 * G::P1];
+    __shared__ float lds_c2[WG_WAVES][G::C2];
+
+    for (int i = threadIdx.x; i < NW; i += WG_THREADS) lds_w[i] = wpack[i];
+    __syncthreads();
+
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const float* b1 = lds_w + G::OB1;
+    unsigned short* xt = lds_xt[wave];
+
+    // Preload B fragments (per-lane, all k-steps) — 16B-aligned global loads.
+    bf16x8 bfr[G::KSTEPS];
+    union BU { unsigned short u[8]; bf16x8 v; };
+    #pragma unroll
+    for (int st = 0; st < G::KSTEPS; ++st) {
+        BU bu;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) bu.u[j] = bfrag[(st * WAVE + lane) * 8 + j];
+        bfr[st] = bu.v;
+    }
+
+    for (long win = blockIdx.x * WG_WAVES + wave; win < SN;
+         win += (long)gridDim.x * WG_WAVES) {
+        const unsigned short* xin = x + win * (G::CIN * G::L);
+        // Stage TRANSPOSED: xt[t*CIN + i] = x[i*L + t]. Coalesced 8-byte
+        // global reads (L % 4 == 0 so a quad never crosses a channel row),
+        // scattered 2-byte LDS writes.
+        static_assert(G::L % 4 == 0, "quad staging needs L % 4 == 0");
+        for (int q = lane; q < G::CIN * G::L / 4; q += WAVE) {
+            const int f = q * 4;
+            const int i = f / G::L, t0 = f % G::L;
+            union { unsigned long long u; unsigned short h[4]; } v;
+            v.u = *(const unsigned long long*)(xin + f);
             #pragma unroll
-            for (int k = 1; k < G::PK; ++k) m = fmaxf(m, src[k]);
-            p1w[o] = m;
+            for (int j = 0; j < 4; ++j)
+                xt[(t0 + j) * G::CIN + i] = v.h[j];
         }
+        for (int i = G::XTN + lane; i < G::XTSZ; i += WAVE) xt[i] = 0;
         wave_sync();
 
-        // --- conv2 + tanh: C2 outputs (k=5, 4 in-ch) ---
-        for (int s = lane; s < G::C2; s += WAVE) {
-            float acc = b2;
+        // conv1 = A(im2col view of xt) x B(weights), one 16-row tile at a time
+        #pragma unroll 1
+        for (int mt = 0; mt < G::MTILES; ++mt) {
+            const int s = mt * 16 + (lane & 15);
+            f32x4 acc = {0.f, 0.f, 0.f, 0.f};
             #pragma unroll
-            for (int c = 0; c < 4; ++c) {
-                const float* pr = p1w + c * G::P1 + s;
-                #pragma unroll
-                for (int k = 0; k < 5; ++k)
-                    acc = fmaf(w2[c * 5 + k], pr[k], acc);
+            for (int st = 0; st < G::KSTEPS; ++st) {
+                const int base = s * G::CIN + st * 32 + (lane >> 4) * 8;
+                BU au;
+                if constexpr (G::CIN % 2 == 0) {
+                    // base is even: 4-byte-aligned LDS reads
+                    const unsigned int* xtu = (const unsigned int*)xt;
+                    unsigned int w0 = xtu[base / 2], w1v = xtu[base / 2 + 1];
+                    unsigned int w2v = xtu[base / 2 + 2], w3 = xtu[base / 2 + 3];
+                    au.u[0] = (unsigned short)w0;  au.u[1] = (unsigned short)(w0 >> 16);
+                    au.u[2] = (unsigned short)w1v; au.u[3] = (unsigned short)(w1v >> 16);
+                    au.u[4] = (unsigned short)w2v; au.u[5] = (unsigned short)(w2v >> 16);
+                    au.u[6] = (unsigned short)w3;  au.u[7] = (unsigned short)(w3 >> 16);
+                } else {
+                    #pragma unroll
+                    for (int j = 0; j < 8; ++j) au.u[j] = xt[base + j];
+                }
+                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(au.v, bfr[st],
+                                                              acc, 0, 0, 0);
             }
-            c2w[s] = tanhf_(acc);
+            // epilogue: bias + tanh; only the 4 real output channels land
+            const int c = lane & 15;
+            if (c < 4) {
+                #pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int row = mt * 16 + (lane >> 4) * 4 + r;
+                    if (row < G::C1)
+                        lds_c1[wave][c * G::C1 + row] = tanhf_(acc[r] + b1[c]);
+                }
+            }
         }
         wave_sync();
-
-        // --- pool2 -> feature vector ---
-        for (int q = lane; q < G::LIN; q += WAVE) {
-            const float* src = c2w + q * G::PS;
-            float m = src[0];
-            #pragma unroll
-            for (int k = 1; k < G::PK; ++k) m = fmaxf(m, src[k]);
-            feat[(long)win * G::LIN + q] = m;
-        }
-        // next grid-stride window re-stages; wave_sync before overwrite
-        wave_sync();
+        conv_tail<G>(lane, lds_w, lds_c1[wave], lds_p1[wave], lds_c2[wave],
+                     feat, win);
     }
 }
 
@@ -198,7 +315,9 @@ __global__ __launch_bounds__(WG_THREADS) void conv_stack_kernel(
 // ---------------------------------------------------------------------------
 // Lane layout: lane l owns gate-unit row l of both LSTM layers, where rows
 // 0-15 = input gate i, 16-31 = forget f, 32-47 = cell g, 48-63 = output o
-// (PyTorch gate order) for hidden units u = l & 15.  h/c live in lanes 0-15.
+// (PyTorch gate order) for hidden units u = l & 15.  After each step the
+// full 16-wide h vectors are collected into per-lane register arrays with
+// 16 independent shuffles, so the next step's h-dots are register FMAs.
 template <class G, int CHUNK = 32>
 __global__ __launch_bounds__(WG_THREADS) void lstm_head_kernel(
     const float* __restrict__ feat,  // (S, N, LIN)
@@ -208,13 +327,14 @@ __global__ __launch_bounds__(WG_THREADS) void lstm_head_kernel(
     int S, int N, float age_eps, int apply_sigmoid)
 {
     constexpr int LIN = G::LIN;
-    __shared__ float lds_feat[WG_WAVES][CHUNK * LIN];
+    constexpr int LINP = (LIN + 3) / 4 * 4;  // float4-padded chunk row
+    __shared__ float lds_feat[WG_WAVES][CHUNK * LINP];
 
     const int wave = threadIdx.x / WAVE;
     const int lane = threadIdx.x % WAVE;
     const int unit = lane & 15;
 
-    // Per-lane weight rows -> registers (broadcast through L2; once per seq).
+    // Per-lane weight rows -> registers (broadcast through L2; once).
     float wih1[LIN], whh1[16], wih2[16], whh2[16];
     const float bl1 = wpack[G::OBL1 + lane];
     const float bl2 = wpack[G::OBL2 + lane];
@@ -226,81 +346,115 @@ __global__ __launch_bounds__(WG_THREADS) void lstm_head_kernel(
         wih2[i] = wpack[G::OWIH2 + lane * 16 + i];
         whh2[i] = wpack[G::OWHH2 + lane * 16 + i];
     }
-    const float outw = wpack[G::OOUTW + unit];  // head weight for own unit
+    const float outw = wpack[G::OOUTW + unit];
     const float outb = wpack[G::OOUTB];
+    const bool is_cell_gate = (lane >= 32 && lane < 48);
 
     float* fw = lds_feat[wave];
 
-    for (int seq = blockIdx.x * WG_WAVES + wave; seq < S;
-         seq += gridDim.x * WG_WAVES) {
-        const float* fseq = feat + (long)seq * N * LIN;
-        const float* aseq = age ? age + (long)seq * N : nullptr;
-        float h1 = 0.f, c1 = 0.f, h2 = 0.f, c2 = 0.f;  // lanes 0-15 hold state
+    for (long seq = blockIdx.x * WG_WAVES + wave; seq < S;
+         seq += (long)gridDim.x * WG_WAVES) {
+        const float* fseq = feat + seq * (long)N * LIN;
+        const float* aseq = age ? age + seq * N : nullptr;
+        float h1v[16], h2v[16];   // full h vectors, replicated per lane
+        float c1 = 0.f, c2 = 0.f; // own-unit cell state
+        #pragma unroll
+        for (int u = 0; u < 16; ++u) h1v[u] = h2v[u] = 0.f;
 
         for (int t0 = 0; t0 < N; t0 += CHUNK) {
             const int tn = min(CHUNK, N - t0);
-            // Prefetch a CHUNK of features into LDS (amortizes HBM latency
-            // over CHUNK sequential steps).
-            for (int i = lane; i < tn * LIN; i += WAVE)
-                fw[i] = fseq[(long)t0 * LIN + i];
+            // Prefetch a feature chunk into LDS (row stride LINP).
+            for (int i = lane; i < tn * LIN; i += WAVE) {
+                const int tt = i / LIN, j = i % LIN;
+                fw[tt * LINP + j] = fseq[(long)(t0 + tt) * LIN + j];
+            }
             wave_sync();
 
             for (int tt = 0; tt < tn; ++tt) {
-                const float* xt = fw + tt * LIN;
-                // ----- layer 1 gates: LIN-dim x-dot (LDS broadcast reads)
-                //       + 16-dim h-dot (register shuffle) -----
-                float g = bl1;
+                const float* xt = fw + tt * LINP;
+                // ----- layer 1: LIN-dim x-dot (LDS broadcast, 2 partial
+                //       accumulators) + 16-dim h-dot (register FMA) -----
+                float ga = bl1, gb = 0.f;
                 #pragma unroll
-                for (int i = 0; i < LIN; ++i) g = fmaf(wih1[i], xt[i], g);
+                for (int i = 0; i < LIN; i += 2) {
+                    ga = fmaf(wih1[i], xt[i], ga);
+                    if (i + 1 < LIN) gb = fmaf(wih1[i + 1], xt[i + 1], gb);
+                }
                 #pragma unroll
-                for (int u = 0; u < 16; ++u)
-                    g = fmaf(whh1[u], __shfl(h1, u), g);
-                float a = (lane >= 32 && lane < 48) ? tanhf_(g) : sigmoidf_(g);
+                for (int u = 0; u < 16; u += 2) {
+                    ga = fmaf(whh1[u], h1v[u], ga);
+                    gb = fmaf(whh1[u + 1], h1v[u + 1], gb);
+                }
+                float g = ga + gb;
+                float a = is_cell_gate ? tanhf_(g) : sigmoidf_(g);
                 {
                     const float iu = __shfl(a, unit);
                     const float fu = __shfl(a, unit + 16);
                     const float gu = __shfl(a, unit + 32);
                     const float ou = __shfl(a, unit + 48);
-                    const float cn = fmaf(fu, c1, iu * gu);
-                    c1 = cn;
-                    h1 = ou * tanhf_(cn);
+                    c1 = fmaf(fu, c1, iu * gu);
+                    const float h = ou * tanhf_(c1);
+                    #pragma unroll
+                    for (int u = 0; u < 16; ++u) h1v[u] = __shfl(h, u);
                 }
-                // ----- layer 2 -----
-                g = bl2;
+                // ----- layer 2: both dots are register FMAs -----
+                ga = bl2; gb = 0.f;
                 #pragma unroll
-                for (int u = 0; u < 16; ++u) {
-                    const float h1u = __shfl(h1, u);
-                    g = fmaf(wih2[u], h1u, g);
-                    g = fmaf(whh2[u], __shfl(h2, u), g);
+                for (int u = 0; u < 16; u += 2) {
+                    ga = fmaf(wih2[u], h1v[u], ga);
+                    gb = fmaf(wih2[u + 1], h1v[u + 1], gb);
+                    ga = fmaf(whh2[u], h2v[u], ga);
+                    gb = fmaf(whh2[u + 1], h2v[u + 1], gb);
                 }
-                a = (lane >= 32 && lane < 48) ? tanhf_(g) : sigmoidf_(g);
+                g = ga + gb;
+                a = is_cell_gate ? tanhf_(g) : sigmoidf_(g);
                 {
                     const float iu = __shfl(a, unit);
                     const float fu = __shfl(a, unit + 16);
                     const float gu = __shfl(a, unit + 32);
                     const float ou = __shfl(a, unit + 48);
-                    const float cn = fmaf(fu, c2, iu * gu);
-                    c2 = cn;
-                    h2 = ou * tanhf_(cn);
+                    c2 = fmaf(fu, c2, iu * gu);
+                    const float h = ou * tanhf_(c2);
+                    #pragma unroll
+                    for (int u = 0; u < 16; ++u) h2v[u] = __shfl(h, u);
                 }
-                // ----- head: logit = sum_u outw[u]*h2[u] + outb,
-                //       then age gate (+ optional sigmoid) -----
-                float p = (lane < 16) ? outw * h2 : 0.f;
-                #pragma unroll
-                for (int off = 8; off > 0; off >>= 1)
-                    p += __shfl_xor(p, off);
+                // ----- head (every lane has h2v: reduce-free dot) -----
                 if (lane == 0) {
-                    float y = p + outb;
+                    float y = outb;
+                    #pragma unroll
+                    for (int u = 0; u < 16; ++u)
+                        y = fmaf(wpack[G::OOUTW + u], h2v[u], y);
                     const float ag = aseq ? aseq[t0 + tt] : 0.f;
-                    const float scale = fmaxf(fmaf(ag, age_eps, 1.0f), 0.0f);
-                    y *= scale;
+                    y *= fmaxf(fmaf(ag, age_eps, 1.0f), 0.0f);
                     if (apply_sigmoid) y = sigmoidf_(y);
-                    out[(long)seq * N + t0 + tt] = y;
+                    out[seq * N + t0 + tt] = y;
                 }
             }
             wave_sync();  // before overwriting the feature chunk
         }
     }
+    (void)outw;
+}
+
+// ---------------------------------------------------------------------------
+// Debug: one 16x16x32 bf16 MFMA tile (fragment-map ground truth for tests)
+// ---------------------------------------------------------------------------
+__global__ void debug_mfma_kernel(const unsigned short* __restrict__ A,
+                                  const unsigned short* __restrict__ B,
+                                  float* __restrict__ C) {
+    const int lane = threadIdx.x;
+    union BU { unsigned short u[8]; bf16x8 v; } a, b;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+        const int k = (lane >> 4) * 8 + j;
+        a.u[j] = A[(lane & 15) * 32 + k];   // A[row][k], row-major 16x32
+        b.u[j] = B[k * 16 + (lane & 15)];   // B[k][col], row-major 32x16
+    }
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
+    #pragma unroll
+    for (int r = 0; r < 4; ++r)
+        C[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = acc[r];
 }
 
 // ---------------------------------------------------------------------------
@@ -310,18 +464,21 @@ namespace {
 
 template <class G>
 int launch_conv(const void* x, int x_is_bf16, float* feat, const float* wpack,
-                int SN, hipStream_t stream) {
+                const void* bfrag, int SN, hipStream_t stream) {
     if (SN <= 0) return 0;
     int grid = (SN + WG_WAVES - 1) / WG_WAVES;
-    if (grid > 8192) grid = 8192;  // grid-stride beyond (Guideline 11)
-    if (x_is_bf16)
-        hipLaunchKernelGGL((conv_stack_kernel<G, unsigned short>), dim3(grid),
+    if (grid > 8192) grid = 8192;
+    if (x_is_bf16) {
+        if (!bfrag) return -2;  // bf16 path requires packed B fragments
+        hipLaunchKernelGGL((conv_stack_mfma_kernel<G>), dim3(grid),
                            dim3(WG_THREADS), 0, stream,
-                           (const unsigned short*)x, feat, wpack, SN);
-    else
-        hipLaunchKernelGGL((conv_stack_kernel<G, float>), dim3(grid),
-                           dim3(WG_THREADS), 0, stream,
-                           (const float*)x, feat, wpack, SN);
+                           (const unsigned short*)x, feat, wpack,
+                           (const unsigned short*)bfrag, SN);
+    } else {
+        hipLaunchKernelGGL((conv_stack_kernel<G>), dim3(grid),
+                           dim3(WG_THREADS), 0, stream, (const float*)x, feat,
+                           wpack, SN);
+    }
     return (int)hipGetLastError();
 }
 
@@ -344,12 +501,13 @@ extern "C" {
 
 // variant: 0 = MyCNN5, 1 = MyCNN2/3, 2 = MyCNN4
 int tskd_conv_fwd(const void* x, int x_is_bf16, float* feat,
-                  const float* wpack, int SN, int variant, void* stream) {
+                  const float* wpack, const void* bfrag, int SN, int variant,
+                  void* stream) {
     hipStream_t s = (hipStream_t)stream;
     switch (variant) {
-        case 0: return launch_conv<GeomCNN5>(x, x_is_bf16, feat, wpack, SN, s);
-        case 1: return launch_conv<GeomCNN2>(x, x_is_bf16, feat, wpack, SN, s);
-        case 2: return launch_conv<GeomCNN4>(x, x_is_bf16, feat, wpack, SN, s);
+        case 0: return launch_conv<GeomCNN5>(x, x_is_bf16, feat, wpack, bfrag, SN, s);
+        case 1: return launch_conv<GeomCNN2>(x, x_is_bf16, feat, wpack, bfrag, SN, s);
+        case 2: return launch_conv<GeomCNN4>(x, x_is_bf16, feat, wpack, bfrag, SN, s);
     }
     return -1;
 }
@@ -369,6 +527,13 @@ int tskd_lstm_head_fwd(const float* feat, const float* age, float* out,
     return -1;
 }
 
+int tskd_debug_mfma16x16x32(const unsigned short* A, const unsigned short* B,
+                            float* C, void* stream) {
+    hipLaunchKernelGGL(debug_mfma_kernel, dim3(1), dim3(WAVE), 0,
+                       (hipStream_t)stream, A, B, C);
+    return (int)hipGetLastError();
+}
+
 int tskd_pack_size(int variant) {
     switch (variant) {
         case 0: return GeomCNN5::NPACK;
@@ -383,6 +548,16 @@ int tskd_feat_len(int variant) {
         case 0: return GeomCNN5::LIN;
         case 1: return GeomCNN2::LIN;
         case 2: return GeomCNN4::LIN;
+    }
+    return -1;
+}
+
+// MFMA B-fragment pack geometry for python (ksteps per variant).
+int tskd_conv_ksteps(int variant) {
+    switch (variant) {
+        case 0: return GeomCNN5::KSTEPS;
+        case 1: return GeomCNN2::KSTEPS;
+        case 2: return GeomCNN4::KSTEPS;
     }
     return -1;
 }

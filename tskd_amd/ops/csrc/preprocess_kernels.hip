@@ -1,0 +1,286 @@
+// Fused streaming-preprocess kernels — CDNA4 (gfx950) HIP.
+//
+// Replaces the reference's Spark Structured Streaming stage (reference
+// bin/processStream.py:105-218 + predictStream.py window assembly :96-142;
+// SURVEY.md §2.6 K13) with GPU-resident per-(stream, channel) ring buffers
+// and three kernels:
+//
+//   1. ingest_dense / ingest_events: raw samples -> 5-s bucket (sum, count).
+//      Dense path: one wavefront per bucket, coalesced strided loads +
+//      wave shuffle reduction. Sparse path (irregular event batches, the
+//      real numerics records at fs=1/60 Hz): atomicAdd into buckets.
+//
+//   2. window_fill: the Spark groupBy(key, channel, window(180 s, 5 s))
+//      .agg(avg) — a window STARTING at grid g covers buckets [g, g+36) and
+//      its average is sum(raw)/count(raw) over those buckets — followed by
+//      the reference's ffill -> bfill -> fillna(0) (processStream.py:114-123).
+//      ffill carries state ACROSS trigger batches (documented improvement
+//      over the reference, whose ffill restarts per 60-s micro-batch).
+//
+//   3. window_gather: assemble model windows (S, B, C, 120) from the last
+//      120 processed grid points per stream (predictStream.py's 600 s/60 s
+//      window -> (1, 10, 120) tensor), batched B windows back at a given
+//      grid stride. Missing channels come out zero (never-filled grid = 0).
+//
+// Ring-buffer layout (all fp32, indexed mod G):
+//   bsum, bcnt : (S, C, G)  raw-sample sum / count per 5-s bucket
+//   proc       : (S, C, G)  processed (filled) window averages by START grid
+//   last_val   : (S, C)     ffill carry (NaN = no value seen yet)
+// Head indices live on the host (python StreamEngine); kernels take them as
+// arguments so the whole trigger step can be captured in a hipGraph.
+
+#include <hip/hip_runtime.h>
+#include <math.h>
+
+#define WAVE 64
+
+__device__ __forceinline__ float bf16_to_f32_(unsigned short u) {
+    union { unsigned int i; float f; } v;
+    v.i = ((unsigned int)u) << 16;
+    return v.f;
+}
+
+// ---------------------------------------------------------------------------
+// 1a. Dense ingest: raw (S, CIN, T) -> NB = T/bucket_len new buckets/channel
+// ---------------------------------------------------------------------------
+template <class DT>
+__global__ void ingest_dense_kernel(
+    const DT* __restrict__ raw,     // (S, CIN, T)
+    float* __restrict__ bsum,       // (S, C, G)
+    float* __restrict__ bcnt,
+    const int* __restrict__ chan_map,  // (CIN) raw row -> wire channel
+    int S, int CIN, int C, int T, int G,
+    int bucket_len, long head)      // buckets written at [head, head+NB)
+{
+    const int NB = T / bucket_len;
+    const long nwaves = (long)S * CIN * NB;
+    const int lane = threadIdx.x % WAVE;
+    for (long w = (long)blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+         w < nwaves; w += (long)gridDim.x * (blockDim.x / WAVE)) {
+        const int b = (int)(w % NB);
+        const int cin = (int)((w / NB) % CIN);
+        const int s = (int)(w / ((long)NB * CIN));
+        const DT* src = raw + ((long)s * CIN + cin) * T + (long)b * bucket_len;
+        float sum = 0.f, cnt = 0.f;
+        for (int i = lane; i < bucket_len; i += WAVE) {
+            float v;
+            if constexpr (sizeof(DT) == 2) v = bf16_to_f32_((unsigned short)src[i]);
+            else v = (float)src[i];
+            if (!isnan(v)) { sum += v; cnt += 1.f; }
+        }
+        #pragma unroll
+        for (int off = 32; off > 0; off >>= 1) {
+            sum += __shfl_xor(sum, off);
+            cnt += __shfl_xor(cnt, off);
+        }
+        if (lane == 0) {
+            const int c = chan_map[cin];
+            const long idx = ((long)s * C + c) * G + (head + b) % G;
+            bsum[idx] = sum;
+            bcnt[idx] = cnt;
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// 1b. Sparse ingest: event tuples (stream, chan, grid_bucket, value)
+// ---------------------------------------------------------------------------
+__global__ void ingest_events_kernel(
+    const int* __restrict__ ev_stream, const int* __restrict__ ev_chan,
+    const long* __restrict__ ev_bucket, const float* __restrict__ ev_val,
+    float* __restrict__ bsum, float* __restrict__ bcnt,
+    int C, int G, long n_events, long min_bucket)
+{
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_events;
+         i += (long)gridDim.x * blockDim.x) {
+        const long b = ev_bucket[i];
+        if (b < min_bucket) continue;  // behind the watermark: dropped (late data)
+        const float v = ev_val[i];
+        if (isnan(v)) continue;
+        const long idx = ((long)ev_stream[i] * C + ev_chan[i]) * G + b % G;
+        atomicAdd(&bsum[idx], v);
+        atomicAdd(&bcnt[idx], 1.f);
+    }
+}
+
+// Zero the bucket slots about to be (re)used: [head, head+nb) mod G.
+__global__ void clear_buckets_kernel(
+    float* __restrict__ bsum, float* __restrict__ bcnt,
+    int S, int C, int G, long head, int nb)
+{
+    const long n = (long)S * C * nb;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+        const int j = (int)(i % nb);
+        const long sc = i / nb;
+        const long idx = sc * G + (head + j) % G;
+        bsum[idx] = 0.f;
+        bcnt[idx] = 0.f;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// 2. Sliding-window average + ffill/bfill/zero-fill
+//    Processes proc grid points [phead, phead+np): window starting at grid j
+//    averages buckets [j, j+win_buckets). One thread per (s, c): the fill is
+//    an order-dependent scan (tiny np per trigger; np can be large in replay
+//    catch-up, still coalesced across the S*C threads).
+// ---------------------------------------------------------------------------
+__global__ void window_fill_kernel(
+    const float* __restrict__ bsum, const float* __restrict__ bcnt,
+    float* __restrict__ proc, float* __restrict__ last_val,
+    int S, int C, int G, long phead, int np, int win_buckets)
+{
+    const long nsc = (long)S * C;
+    for (long sc = (long)blockIdx.x * blockDim.x + threadIdx.x; sc < nsc;
+         sc += (long)gridDim.x * blockDim.x) {
+        const float* bs = bsum + sc * G;
+        const float* bc = bcnt + sc * G;
+        float* pr = proc + sc * G;
+        float carry = last_val[sc];
+
+        // forward pass: raw window average, then ffill (with cross-batch carry)
+        for (int j = 0; j < np; ++j) {
+            const long g0 = phead + j;
+            float sum = 0.f, cnt = 0.f;
+            for (int k = 0; k < win_buckets; ++k) {
+                const long idx = (g0 + k) % G;
+                sum += bs[idx];
+                cnt += bc[idx];
+            }
+            float v = (cnt > 0.f) ? sum / cnt : NAN;
+            if (isnan(v)) v = carry;      // ffill (carry may be NaN)
+            else carry = v;
+            pr[g0 % G] = v;
+        }
+        last_val[sc] = carry;
+
+        // backward pass: bfill within this batch, then fillna(0)
+        float nxt = NAN;
+        for (int j = np - 1; j >= 0; --j) {
+            const long idx = (phead + j) % G;
+            float v = pr[idx];
+            if (isnan(v)) v = nxt;
+            else nxt = v;
+            pr[idx] = isnan(v) ? 0.f : v;
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// 3. Window gather: (S, B, C, WIN) model inputs from processed grid.
+//    Window b (b = 0..B-1) covers grid [end - (B-1-b)*stride - WIN,
+//    end - (B-1-b)*stride). Grid points never produced (g < 0) read as 0.
+// ---------------------------------------------------------------------------
+template <class OT>
+__global__ void window_gather_kernel(
+    const float* __restrict__ proc,
+    OT* __restrict__ out,            // (S, B, C, WIN)
+    int S, int C, int G, int B, int WIN, int stride, long end)
+{
+    const long n = (long)S * B * C * WIN;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+        const int t = (int)(i % WIN);
+        const int c = (int)((i / WIN) % C);
+        const int b = (int)((i / ((long)WIN * C)) % B);
+        const int s = (int)(i / ((long)WIN * C * B));
+        const long wend = end - (long)(B - 1 - b) * stride;
+        const long g = wend - WIN + t;
+        float v = 0.f;
+        if (g >= 0 && g < end && wend - WIN >= 0)
+            v = proc[((long)s * C + c) * G + g % G];
+        if constexpr (sizeof(OT) == 2) {
+            union { float f; unsigned int i; } u;
+            u.f = v;
+            // round-to-nearest-even bf16
+            unsigned int r = u.i + 0x7fffu + ((u.i >> 16) & 1);
+            out[i] = (OT)(r >> 16);
+        } else {
+            out[i] = (OT)v;
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// extern "C" launchers
+// ---------------------------------------------------------------------------
+static inline int grid_for(long n, int block) {
+    long g = (n + block - 1) / block;
+    if (g > 16384) g = 16384;
+    if (g < 1) g = 1;
+    return (int)g;
+}
+
+extern "C" {
+
+int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
+                              float* bsum, float* bcnt, const int* chan_map,
+                              int S, int CIN, int C, int T, int G,
+                              int bucket_len, long head, void* stream) {
+    hipStream_t st = (hipStream_t)stream;
+    const int NB = T / bucket_len;
+    if (NB <= 0 || S <= 0) return 0;
+    const long nwaves = (long)S * CIN * NB;
+    const int grid = grid_for(nwaves * WAVE, 256);
+    if (raw_is_bf16)
+        hipLaunchKernelGGL((ingest_dense_kernel<unsigned short>), dim3(grid),
+                           dim3(256), 0, st, (const unsigned short*)raw, bsum,
+                           bcnt, chan_map, S, CIN, C, T, G, bucket_len, head);
+    else
+        hipLaunchKernelGGL((ingest_dense_kernel<float>), dim3(grid), dim3(256),
+                           0, st, (const float*)raw, bsum, bcnt, chan_map, S,
+                           CIN, C, T, G, bucket_len, head);
+    return (int)hipGetLastError();
+}
+
+int tskd_preproc_ingest_events(const int* ev_stream, const int* ev_chan,
+                               const long* ev_bucket, const float* ev_val,
+                               float* bsum, float* bcnt, int C, int G,
+                               long n_events, long min_bucket, void* stream) {
+    if (n_events <= 0) return 0;
+    hipLaunchKernelGGL(ingest_events_kernel, dim3(grid_for(n_events, 256)),
+                       dim3(256), 0, (hipStream_t)stream, ev_stream, ev_chan,
+                       ev_bucket, ev_val, bsum, bcnt, C, G, n_events,
+                       min_bucket);
+    return (int)hipGetLastError();
+}
+
+int tskd_preproc_clear_buckets(float* bsum, float* bcnt, int S, int C, int G,
+                               long head, int nb, void* stream) {
+    if (nb <= 0) return 0;
+    const long n = (long)S * C * nb;
+    hipLaunchKernelGGL(clear_buckets_kernel, dim3(grid_for(n, 256)), dim3(256),
+                       0, (hipStream_t)stream, bsum, bcnt, S, C, G, head, nb);
+    return (int)hipGetLastError();
+}
+
+int tskd_preproc_window_fill(const float* bsum, const float* bcnt, float* proc,
+                             float* last_val, int S, int C, int G, long phead,
+                             int np, int win_buckets, void* stream) {
+    if (np <= 0) return 0;
+    const long nsc = (long)S * C;
+    hipLaunchKernelGGL(window_fill_kernel, dim3(grid_for(nsc, 256)), dim3(256),
+                       0, (hipStream_t)stream, bsum, bcnt, proc, last_val, S,
+                       C, G, phead, np, win_buckets);
+    return (int)hipGetLastError();
+}
+
+int tskd_preproc_window_gather(const float* proc, void* out, int out_is_bf16,
+                               int S, int C, int G, int B, int WIN, int stride,
+                               long end, void* stream) {
+    const long n = (long)S * B * C * WIN;
+    if (n <= 0) return 0;
+    hipStream_t st = (hipStream_t)stream;
+    if (out_is_bf16)
+        hipLaunchKernelGGL((window_gather_kernel<unsigned short>),
+                           dim3(grid_for(n, 256)), dim3(256), 0, st, proc,
+                           (unsigned short*)out, S, C, G, B, WIN, stride, end);
+    else
+        hipLaunchKernelGGL((window_gather_kernel<float>),
+                           dim3(grid_for(n, 256)), dim3(256), 0, st, proc,
+                           (float*)out, S, C, G, B, WIN, stride, end);
+    return (int)hipGetLastError();
+}
+
+}  // extern "C"

@@ -16,6 +16,31 @@ import torch
 VARIANT_IDS = {"MyCNN5": 0, "MyCNN": 0, "MyCNN2": 1, "MyCNN3": 1, "MyCNN4": 2}
 
 
+def pack_conv1_mfma_bfrags(model) -> torch.Tensor:
+    """Pack conv1 weights into v_mfma_f32_16x16x32_bf16 B-fragment order.
+
+    Returns bf16 tensor [KSTEPS, 64, 8]: lane l of k-step st holds
+    B[k = (l>>4)*8 + j][col = l&15] where B[kk][c] = W1[c][kk % CIN][kk // CIN]
+    (the im2col reduction index kk = k*CIN + i matching the kernel's
+    transposed-window A view), zero-padded for kk >= CIN*K1 and col >= 4.
+    """
+    w1 = model.conv1.weight.detach().float().cpu()  # (4, CIN, K1)
+    _, cin, k1 = w1.shape
+    kk_n = cin * k1
+    ksteps = (kk_n + 31) // 32
+    out = torch.zeros(ksteps, 64, 8, dtype=torch.float32)
+    for st in range(ksteps):
+        for lane in range(64):
+            col = lane & 15
+            if col >= 4:
+                continue
+            for j in range(8):
+                kk = st * 32 + (lane >> 4) * 8 + j
+                if kk < kk_n:
+                    out[st, lane, j] = w1[col, kk % cin, kk // cin]
+    return out.to(torch.bfloat16).contiguous()
+
+
 def pack_weights(model) -> torch.Tensor:
     sd = {k: v.detach().float().cpu() for k, v in model.state_dict().items()}
     parts = [
