@@ -1,0 +1,318 @@
+"""StreamEngine — GPU-resident streaming window engine.
+
+The MI355X-native replacement for the reference's Spark Structured Streaming
+stage: per-(stream, channel) ring buffers in HBM3E hold 5-s
+bucket aggregates and processed 180 s/5 s window averages; the fused HIP
+kernels in csrc/preprocess_kernels.hip do ingest, sliding-mean + gap-fill and
+model-window gather entirely on-GPU, so raw waveforms never round-trip to the
+host between ingestion and inference.
+
+Memory plan (SURVEY.md §5 long-context): state per (stream, channel) is
+O(ring), never O(stream): 3 fp32 rings of G grid points. At G=4096 (5.7 h of
+5-s history), 64k streams x 10 ch cost 64k*10*4096*12 B = 31 GB of the
+288 GB HBM3E — BASELINE.json config 4's 64k-stream scenario fits one GPU.
+
+CPU mode runs the same semantics in numpy (the tests' property oracle is
+pandas-free windowing.py, itself tested against the kernels).
+"""
+
+from __future__ import annotations
+
+import ctypes
+from typing import Optional, Sequence
+
+import numpy as np
+import torch
+
+from tskd_amd.engine import windowing as W
+
+_plib: Optional[ctypes.CDLL] = None
+
+
+def _load_preproc_lib() -> ctypes.CDLL:
+    global _plib
+    if _plib is not None:
+        return _plib
+    import os
+
+    from tskd_amd.ops import build as _build
+    path = _build.lib_path("_tskd_preprocess")
+    if not os.path.exists(path):
+        _build.build("_tskd_preprocess")
+    lib = ctypes.CDLL(path)
+    lib.tskd_preproc_ingest_dense.restype = ctypes.c_int
+    lib.tskd_preproc_ingest_dense.argtypes = [
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+        ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_long,
+        ctypes.c_void_p,
+    ]
+    lib.tskd_preproc_ingest_events.restype = ctypes.c_int
+    lib.tskd_preproc_ingest_events.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
+        ctypes.c_long, ctypes.c_long, ctypes.c_void_p,
+    ]
+    lib.tskd_preproc_clear_buckets.restype = ctypes.c_int
+    lib.tskd_preproc_clear_buckets.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
+        ctypes.c_int, ctypes.c_long, ctypes.c_int, ctypes.c_void_p,
+    ]
+    lib.tskd_preproc_window_fill.restype = ctypes.c_int
+    lib.tskd_preproc_window_fill.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_long,
+        ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
+    ]
+    lib.tskd_preproc_window_gather.restype = ctypes.c_int
+    lib.tskd_preproc_window_gather.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
+        ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+        ctypes.c_long, ctypes.c_void_p,
+    ]
+    _plib = lib
+    return lib
+
+
+def _sptr() -> ctypes.c_void_p:
+    return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+class StreamEngine:
+    """Event-time sliding-window engine over S streams x C channels.
+
+    Durations are expressed in GRID units internally; the reference constants
+    (180 s window / 5 s slide / 600 s model window) map to win_buckets=36 and
+    model_win=120. ``speed`` compresses wall-clock like the reference's
+    --speed flag but does NOT change grid semantics.
+    """
+
+    def __init__(self, n_streams: int, n_channels: int = 10,
+                 ring_grid: int = 4096, fs: float = 125.0,
+                 bucket_s: float = W.BUCKET_S, win_buckets: int = W.WIN_BUCKETS,
+                 model_win: int = W.MODEL_WIN, device: str = "cpu"):
+        self.S, self.C, self.G = n_streams, n_channels, ring_grid
+        self.fs = fs
+        self.bucket_s = bucket_s
+        self.bucket_len = int(round(fs * bucket_s))  # dense samples per bucket
+        self.win_buckets = win_buckets
+        self.model_win = model_win
+        self.device = torch.device(device)
+        self.head = 0    # buckets ingested (dense path)
+        self.nproc = 0   # processed grid points produced
+        self._cleared = 0  # ring slots zeroed up to this bucket (event path)
+        dev = self.device
+        self.bsum = torch.zeros(self.S, self.C, self.G, device=dev)
+        self.bcnt = torch.zeros(self.S, self.C, self.G, device=dev)
+        self.proc = torch.zeros(self.S, self.C, self.G, device=dev)
+        self.last_val = torch.full((self.S, self.C), float("nan"), device=dev)
+        self._gpu = dev.type == "cuda"
+        if self._gpu:
+            _load_preproc_lib()
+
+    # ------------------------------------------------------------------ ingest
+    def ingest_dense(self, raw: torch.Tensor,
+                     chan_map: Optional[Sequence[int]] = None) -> int:
+        """raw (S, CIN, T) contiguous samples at self.fs starting at the
+        current head; returns number of new buckets per channel."""
+        assert raw.shape[0] == self.S
+        cin, t = raw.shape[1], raw.shape[2]
+        nb = t // self.bucket_len
+        if nb > self.G - self.win_buckets:
+            raise ValueError(
+                f"batch spans {nb} buckets > ring capacity "
+                f"{self.G - self.win_buckets}; ingest in smaller chunks")
+        if chan_map is None:
+            chan_map = list(range(cin))
+        if self._gpu:
+            lib = _load_preproc_lib()
+            raw = raw.contiguous()
+            is_bf16 = 1 if raw.dtype == torch.bfloat16 else 0
+            if not is_bf16 and raw.dtype != torch.float32:
+                raw = raw.float()
+            cm = torch.tensor(list(chan_map), dtype=torch.int32,
+                              device=self.device)
+            rc = lib.tskd_preproc_ingest_dense(
+                ctypes.c_void_p(raw.data_ptr()), is_bf16,
+                ctypes.c_void_p(self.bsum.data_ptr()),
+                ctypes.c_void_p(self.bcnt.data_ptr()),
+                ctypes.c_void_p(cm.data_ptr()),
+                self.S, cin, self.C, t, self.G, self.bucket_len,
+                ctypes.c_long(self.head), _sptr())
+            if rc != 0:
+                raise RuntimeError(f"ingest_dense failed: hipError {rc}")
+            self._clear_stale_channels(chan_map, nb)
+        else:
+            rawf = raw.float().cpu().numpy()
+            for j in range(nb):
+                seg = rawf[:, :, j * self.bucket_len:(j + 1) * self.bucket_len]
+                ok = ~np.isnan(seg)
+                g = (self.head + j) % self.G
+                for ci, c in enumerate(chan_map):
+                    self.bsum[:, c, g] = torch.from_numpy(
+                        np.where(ok[:, ci], seg[:, ci], 0.0).sum(-1)).float()
+                    self.bcnt[:, c, g] = torch.from_numpy(
+                        ok[:, ci].sum(-1).astype(np.float64)).float()
+                # channels not in chan_map get empty buckets at this grid
+                other = [c for c in range(self.C) if c not in chan_map]
+                if other:
+                    self.bsum[:, other, g] = 0
+                    self.bcnt[:, other, g] = 0
+        self.head += nb
+        self._cleared = max(self._cleared, self.head)
+        self._refill()
+        return nb
+
+    def _clear_stale_channels(self, chan_map: Sequence[int], nb: int) -> None:
+        """Channels with no data this trigger still need their (re-used) ring
+        slots zeroed — GPU path (CPU path does it inline)."""
+        other = [c for c in range(self.C) if c not in set(chan_map)]
+        if not other or nb <= 0:
+            return
+        idx = torch.tensor([(self.head + j) % self.G for j in range(nb)],
+                           dtype=torch.long, device=self.device)
+        oth = torch.tensor(other, dtype=torch.long, device=self.device)
+        self.bsum[:, oth[:, None], idx[None, :]] = 0
+        self.bcnt[:, oth[:, None], idx[None, :]] = 0
+
+    def ingest_events(self, stream_idx: torch.Tensor, chan_idx: torch.Tensor,
+                      ts: torch.Tensor, vals: torch.Tensor,
+                      advance_to: Optional[float] = None) -> None:
+        """Irregular event batch (the real numerics-record path, fs=1/60 Hz).
+
+        Event time ``ts`` is in seconds; buckets behind the already-processed
+        grid (the watermark) are dropped, like Spark's withWatermark
+        (processStream.py:196). ``advance_to``: event-time high-water mark in
+        seconds (defaults to max ts) — buckets strictly before it become
+        eligible for processing.
+        """
+        bucket = torch.floor(ts.double() / self.bucket_s).long()
+        min_bucket = self.nproc  # processed grid is immutable
+        if len(bucket) and int(bucket.max()) - max(int(bucket.min()), min_bucket) \
+                >= self.G - self.win_buckets:
+            raise ValueError("event batch spans more buckets than the ring "
+                             "holds; ingest in smaller chunks")
+        self._clear_ahead(int(bucket.max().item()) + 1 if len(bucket) else 0)
+        if self._gpu:
+            lib = _load_preproc_lib()
+            si = stream_idx.to(torch.int32).contiguous().to(self.device)
+            ci = chan_idx.to(torch.int32).contiguous().to(self.device)
+            bi = bucket.contiguous().to(self.device)
+            vi = vals.float().contiguous().to(self.device)
+            rc = lib.tskd_preproc_ingest_events(
+                ctypes.c_void_p(si.data_ptr()), ctypes.c_void_p(ci.data_ptr()),
+                ctypes.c_void_p(bi.data_ptr()), ctypes.c_void_p(vi.data_ptr()),
+                ctypes.c_void_p(self.bsum.data_ptr()),
+                ctypes.c_void_p(self.bcnt.data_ptr()),
+                self.C, self.G, ctypes.c_long(len(vi)),
+                ctypes.c_long(min_bucket), _sptr())
+            if rc != 0:
+                raise RuntimeError(f"ingest_events failed: hipError {rc}")
+        else:
+            ok = ~torch.isnan(vals)
+            for s, c, b, v in zip(stream_idx[ok].tolist(), chan_idx[ok].tolist(),
+                                  bucket[ok].tolist(), vals[ok].tolist()):
+                if b < min_bucket:
+                    continue
+                self.bsum[s, c, b % self.G] += v
+                self.bcnt[s, c, b % self.G] += 1
+        hwm = float(advance_to) if advance_to is not None else float(ts.max())
+        new_head = int(np.floor(hwm / self.bucket_s))
+        if new_head > self.head:
+            self.head = new_head
+        self._refill()
+
+    def _clear_ahead(self, upto_bucket: int) -> None:
+        """Zero ring slots about to be re-used (stale data from G buckets
+        ago). No-op until the ring wraps."""
+        if upto_bucket <= self._cleared:
+            return
+        start = max(self._cleared, upto_bucket - self.G)
+        nb = upto_bucket - start
+        if self._gpu:
+            lib = _load_preproc_lib()
+            rc = lib.tskd_preproc_clear_buckets(
+                ctypes.c_void_p(self.bsum.data_ptr()),
+                ctypes.c_void_p(self.bcnt.data_ptr()),
+                self.S, self.C, self.G, ctypes.c_long(start), nb, _sptr())
+            if rc != 0:
+                raise RuntimeError(f"clear_buckets failed: hipError {rc}")
+        else:
+            idx = np.arange(start, start + nb) % self.G
+            self.bsum[:, :, idx] = 0
+            self.bcnt[:, :, idx] = 0
+        self._cleared = upto_bucket
+
+    # ----------------------------------------------------------------- process
+    def _refill(self) -> None:
+        """Produce newly-complete processed grid points (window starts)."""
+        navail = max(0, self.head - self.win_buckets + 1)
+        np_new = navail - self.nproc
+        if np_new <= 0:
+            return
+        if self._gpu:
+            lib = _load_preproc_lib()
+            rc = lib.tskd_preproc_window_fill(
+                ctypes.c_void_p(self.bsum.data_ptr()),
+                ctypes.c_void_p(self.bcnt.data_ptr()),
+                ctypes.c_void_p(self.proc.data_ptr()),
+                ctypes.c_void_p(self.last_val.data_ptr()),
+                self.S, self.C, self.G, ctypes.c_long(self.nproc), np_new,
+                self.win_buckets, _sptr())
+            if rc != 0:
+                raise RuntimeError(f"window_fill failed: hipError {rc}")
+        else:
+            bs = self.bsum.numpy()
+            bc = self.bcnt.numpy()
+            pr = self.proc.numpy()
+            lv = self.last_val.numpy()
+            for s in range(self.S):
+                for c in range(self.C):
+                    vals = np.empty(np_new)
+                    for j in range(np_new):
+                        g0 = self.nproc + j
+                        idx = (g0 + np.arange(self.win_buckets)) % self.G
+                        cnt = bc[s, c, idx].sum()
+                        vals[j] = (bs[s, c, idx].sum() / cnt if cnt > 0
+                                   else np.nan)
+                    filled, carry = W.fill_series(vals, lv[s, c])
+                    for j in range(np_new):
+                        pr[s, c, (self.nproc + j) % self.G] = filled[j]
+                    lv[s, c] = carry
+        self.nproc = navail
+
+    # ------------------------------------------------------------------ gather
+    def windows(self, batch: int = 1, stride: int = 12,
+                dtype: torch.dtype = torch.float32) -> torch.Tensor:
+        """Assemble (S, batch, C, model_win) model inputs ending at the latest
+        processed grid point; window b ends at nproc - (batch-1-b)*stride.
+        Early windows that would reach before grid 0 are all-zero."""
+        B, WIN = batch, self.model_win
+        out = torch.zeros(self.S, B, self.C, WIN, dtype=dtype,
+                          device=self.device)
+        if self._gpu:
+            lib = _load_preproc_lib()
+            is_bf16 = 1 if dtype == torch.bfloat16 else 0
+            assert dtype in (torch.bfloat16, torch.float32)
+            rc = lib.tskd_preproc_window_gather(
+                ctypes.c_void_p(self.proc.data_ptr()),
+                ctypes.c_void_p(out.data_ptr()), is_bf16,
+                self.S, self.C, self.G, B, WIN, stride,
+                ctypes.c_long(self.nproc), _sptr())
+            if rc != 0:
+                raise RuntimeError(f"window_gather failed: hipError {rc}")
+        else:
+            pr = self.proc.numpy()
+            for b in range(B):
+                wend = self.nproc - (B - 1 - b) * stride
+                if wend - WIN < 0:
+                    continue
+                idx = (np.arange(wend - WIN, wend)) % self.G
+                out[:, b] = torch.from_numpy(pr[:, :, idx]).to(dtype)
+        return out
+
+    @property
+    def ready(self) -> bool:
+        """A full model window exists (>= 600 s + 180 s of data, matching the
+        reference's ~10-minutes-to-first-prediction behavior)."""
+        return self.nproc >= self.model_win
