@@ -1,19 +1,25 @@
 #!/usr/bin/env python3
-"""Flagship serving benchmark: waveform windows/sec, MyCNN5, MI355X.
+"""Flagship benchmark: waveform windows/sec (whole node) + p50 step latency,
+MyCNN5 8-ch serving on MI355X.
 
-Measures the BASELINE.json headline metric — waveform windows/sec (whole
-node) + p50 step latency for the MyCNN5 8-channel serving path — on
-synthetic data (no network: synthetic 8-active-channel windows, random-init
-weights; BASELINE.md: the reference publishes no quantitative numbers, so
-vs_baseline is null).
+Two modes (BASELINE.json configs):
+  --mode pipeline (default; config 4): full fused end-to-end serving step —
+      synthetic raw 8-channel 125 Hz waveform chunks (one 60-s trigger per
+      step) -> GPU ring-buffer ingest -> fused 180 s/5 s sliding-mean +
+      gap-fill -> model-window gather -> fused MFMA conv + LSTM + head +
+      sigmoid -> prediction all-gather over RCCL/xGMI (world > 1).
+      One model window per stream per trigger (reference serving semantics:
+      600 s window / 60 s slide, predictStream.py:248-263).
+  --mode infer (config 2): model-path throughput — S sequences x B=1024-window
+      batches through the fused conv+LSTM path (reference batch semantics
+      incl. the LSTM batch-axis-as-time quirk).
 
-One step = one serving macro-batch per GPU: S sequences x B=1024-window
-batches through the fused HIP conv+LSTM+head+sigmoid path (reference
-semantics per batch, incl. the LSTM batch-axis-as-time quirk).
+value = whole-job windows/s across all ranks. Data is synthetic (no network)
+with random-init weights; the reference publishes no quantitative numbers
+ so vs_baseline is null.
 
-Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
-  For N>1 the driver launches this under torch.distributed.run with one rank
-  per GPU (RCCL over xGMI); we read RANK/LOCAL_RANK/WORLD_SIZE from the env.
+The driver launches N>1 via torch.distributed.run, one rank per GPU; we read
+RANK/LOCAL_RANK/WORLD_SIZE from the env (RCCL over xGMI).
 """
 
 from __future__ import annotations
@@ -30,15 +36,42 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 import torch  # noqa: E402
 
 
+def run_steps(step_fn, steps, warmup, dist, device):
+    for _ in range(warmup):
+        step_fn()
+    torch.cuda.synchronize()
+    if dist:
+        dist.barrier()
+    torch.cuda.synchronize()
+    lat = []
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        s0 = time.perf_counter()
+        step_fn()
+        torch.cuda.synchronize()
+        lat.append(time.perf_counter() - s0)
+    if dist:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    elapsed = torch.tensor([t1 - t0], device=device, dtype=torch.float64)
+    if dist:
+        dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
+    return float(elapsed.item()), lat
+
+
 def main() -> None:
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=50)
-    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=15)
+    p.add_argument("--mode", default="pipeline", choices=["pipeline", "infer"])
+    p.add_argument("--streams", type=int, default=16384,
+                   help="[pipeline] concurrent patient streams per GPU")
     p.add_argument("--seqs", type=int, default=2048,
-                   help="concurrent patient sequences per GPU")
+                   help="[infer] concurrent sequences per GPU")
     p.add_argument("--batch", type=int, default=1024,
-                   help="windows per sequence batch (reference batch semantics)")
+                   help="[infer] windows per sequence batch")
     p.add_argument("--variant", default="MyCNN5")
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     args = p.parse_args()
@@ -62,44 +95,51 @@ def main() -> None:
     torch.manual_seed(1234 + rank)
     model = build_model(args.variant).eval()
     eng = MyCNNEngine(model, device=device)
-
-    S, B = args.seqs, args.batch
     dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
-    # Synthetic 8-active-channel 125 Hz-derived windows (2 of the 10 wire
-    # channels absent -> zero, as in the MIMIC numerics records).
-    x = torch.randn(S, B, 10, 120, device=device, dtype=dtype)
-    x[:, :, 8:, :] = 0
-    age = torch.full((S, B), 65.0, device=device)
 
-    def step() -> None:
-        eng.forward(x, age, apply_sigmoid=True)
+    if args.mode == "infer":
+        S, B = args.seqs, args.batch
+        x = torch.randn(S, B, 10, 120, device=device, dtype=dtype)
+        x[:, :, 8:, :] = 0  # 8 active of 10 wire channels
+        age = torch.full((S, B), 65.0, device=device)
 
-    for _ in range(args.warmup):
-        step()
-    torch.cuda.synchronize()
-    if dist:
-        dist.barrier()
-    torch.cuda.synchronize()
+        def step():
+            eng.forward(x, age, apply_sigmoid=True)
 
-    lat = []
-    t0 = time.perf_counter()
-    for _ in range(args.steps):
-        s0 = time.perf_counter()
-        step()
-        torch.cuda.synchronize()
-        lat.append(time.perf_counter() - s0)
-    if dist:
-        dist.barrier()
-    torch.cuda.synchronize()
-    t1 = time.perf_counter()
+        windows_per_step = S * B
+        cfg = {"model": args.variant, "global_batch": windows_per_step * world,
+               "seq_len": 120, "batch_per_seq": B, "seqs_per_gpu": S,
+               "parallelism": f"dp{world}", "mode": "infer"}
+    else:
+        from tskd_amd.engine import StreamEngine
+        S = args.streams
+        fs = 125.0
+        se = StreamEngine(S, 10, ring_grid=2048, fs=fs, device=device)
+        chan_map = list(range(8))  # 8 active channels
+        trigger_samples = int(fs * 60)  # one 60-s trigger per step
+        # Pre-generate one trigger's worth of raw data (synthetic; in
+        # production this arrives from the bus).
+        raw = torch.randn(S, 8, trigger_samples, device=device, dtype=dtype)
+        age = torch.full((S, 1), 65.0, device=device)
+        gathered = [torch.empty(S, device=device) for _ in range(world)] \
+            if dist else None
 
-    elapsed = torch.tensor([t1 - t0], device=device, dtype=torch.float64)
-    if dist:
-        dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
-    elapsed_s = float(elapsed.item())
+        def step():
+            se.ingest_dense(raw, chan_map=chan_map)
+            w = se.windows(batch=1, stride=12, dtype=dtype)
+            probs = eng.forward(w, age, apply_sigmoid=True)
+            if dist:
+                # predictions to every rank (RCCL all-gather over xGMI)
+                dist.all_gather(gathered, probs.reshape(S).contiguous())
 
-    windows_per_step_per_gpu = S * B
-    total_windows = windows_per_step_per_gpu * args.steps * world
+        windows_per_step = S
+        cfg = {"model": args.variant, "global_batch": S * world,
+               "seq_len": 120, "streams_per_gpu": S,
+               "trigger_s": 60, "fs_hz": 125,
+               "parallelism": f"dp{world}", "mode": "pipeline"}
+
+    elapsed_s, lat = run_steps(step, args.steps, args.warmup, dist, device)
+    total_windows = windows_per_step * args.steps * world
     value = total_windows / elapsed_s
     ms_per_step = elapsed_s / args.steps * 1000.0
 
@@ -117,15 +157,8 @@ def main() -> None:
             "scaling": "weak",
             "vs_baseline": None,
             "dtype": args.dtype,
-            "data": "synthetic 8-active-ch windows, random-init weights",
-            "config": {
-                "model": args.variant,
-                "global_batch": windows_per_step_per_gpu * world,
-                "seq_len": 120,
-                "batch_per_seq": B,
-                "seqs_per_gpu": S,
-                "parallelism": f"dp{world}",
-            },
+            "data": "synthetic 8-active-ch 125 Hz waveforms, random-init weights",
+            "config": cfg,
         }), flush=True)
 
     if dist:
