@@ -40,6 +40,12 @@ __device__ __forceinline__ float bf16_to_f32_(unsigned short u) {
     return v.f;
 }
 
+// Same-wave LDS RAW fence (see mycnn_kernels.hip wave_sync).
+__device__ __forceinline__ void wsync_() {
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_wave_barrier();
+}
+
 // ---------------------------------------------------------------------------
 // 1a. Dense ingest: raw (S, CIN, T) -> NB = T/bucket_len new buckets/channel
 // ---------------------------------------------------------------------------
@@ -62,11 +68,50 @@ __global__ void ingest_dense_kernel(
         const int s = (int)(w / ((long)NB * CIN));
         const DT* src = raw + ((long)s * CIN + cin) * T + (long)b * bucket_len;
         float sum = 0.f, cnt = 0.f;
-        for (int i = lane; i < bucket_len; i += WAVE) {
-            float v;
-            if constexpr (sizeof(DT) == 2) v = bf16_to_f32_((unsigned short)src[i]);
-            else v = (float)src[i];
-            if (!isnan(v)) { sum += v; cnt += 1.f; }
+        // 16B-aligned vector body with scalar head/tail peel (bucket offsets
+        // like 625 samples are not 16B-aligned).
+        if constexpr (sizeof(DT) == 2) {
+            int pre = (int)(((16 - ((size_t)src & 15)) & 15) / 2);
+            if (pre > bucket_len) pre = bucket_len;
+            for (int i = lane; i < pre; i += WAVE) {
+                const float f = bf16_to_f32_((unsigned short)src[i]);
+                if (!isnan(f)) { sum += f; cnt += 1.f; }
+            }
+            const int oct = (bucket_len - pre) / 8;
+            const uint4* vp = (const uint4*)((const unsigned short*)src + pre);
+            for (int p = lane; p < oct; p += WAVE) {
+                union { uint4 q; unsigned short h[8]; } v;
+                v.q = vp[p];
+                #pragma unroll
+                for (int j = 0; j < 8; ++j) {
+                    const float f = bf16_to_f32_(v.h[j]);
+                    if (!isnan(f)) { sum += f; cnt += 1.f; }
+                }
+            }
+            for (int i = pre + oct * 8 + lane; i < bucket_len; i += WAVE) {
+                const float f = bf16_to_f32_((unsigned short)src[i]);
+                if (!isnan(f)) { sum += f; cnt += 1.f; }
+            }
+        } else {
+            int pre = (int)(((16 - ((size_t)src & 15)) & 15) / 4);
+            if (pre > bucket_len) pre = bucket_len;
+            for (int i = lane; i < pre; i += WAVE) {
+                const float f = (float)src[i];
+                if (!isnan(f)) { sum += f; cnt += 1.f; }
+            }
+            const int quad = (bucket_len - pre) / 4;
+            const float4* vp = (const float4*)((const float*)src + pre);
+            for (int p = lane; p < quad; p += WAVE) {
+                const float4 v = vp[p];
+                if (!isnan(v.x)) { sum += v.x; cnt += 1.f; }
+                if (!isnan(v.y)) { sum += v.y; cnt += 1.f; }
+                if (!isnan(v.z)) { sum += v.z; cnt += 1.f; }
+                if (!isnan(v.w)) { sum += v.w; cnt += 1.f; }
+            }
+            for (int i = pre + quad * 4 + lane; i < bucket_len; i += WAVE) {
+                const float f = (float)src[i];
+                if (!isnan(f)) { sum += f; cnt += 1.f; }
+            }
         }
         #pragma unroll
         for (int off = 32; off > 0; off >>= 1) {
@@ -126,44 +171,86 @@ __global__ void clear_buckets_kernel(
 //    an order-dependent scan (tiny np per trigger; np can be large in replay
 //    catch-up, still coalesced across the S*C threads).
 // ---------------------------------------------------------------------------
-__global__ void window_fill_kernel(
+// One WAVE per (stream, channel): coalesced bucket loads, LDS-staged sliding
+// sums (each lane owns one output point per 64-point chunk), the
+// order-dependent ffill scan runs on lane 0 over LDS (np is small per
+// trigger; large only in replay catch-up, still chunked).
+__global__ __launch_bounds__(256) void window_fill_kernel(
     const float* __restrict__ bsum, const float* __restrict__ bcnt,
     float* __restrict__ proc, float* __restrict__ last_val,
     int S, int C, int G, long phead, int np, int win_buckets)
 {
+    constexpr int CHUNK = 64;             // output points per iteration
+    const int MAXW = 64;                  // win_buckets <= 64 (default 36)
+    __shared__ float lds_s[4][CHUNK + 64];  // bucket sums (CHUNK+MAXW-1 used)
+    __shared__ float lds_c[4][CHUNK + 64];
+    __shared__ float lds_v[4][CHUNK];       // window values / filled output
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    float* ls = lds_s[wave];
+    float* lc = lds_c[wave];
+    float* lv = lds_v[wave];
+    (void)MAXW;
+
     const long nsc = (long)S * C;
-    for (long sc = (long)blockIdx.x * blockDim.x + threadIdx.x; sc < nsc;
-         sc += (long)gridDim.x * blockDim.x) {
+    for (long sc = (long)blockIdx.x * 4 + wave; sc < nsc;
+         sc += (long)gridDim.x * 4) {
         const float* bs = bsum + sc * G;
         const float* bc = bcnt + sc * G;
         float* pr = proc + sc * G;
         float carry = last_val[sc];
+        // After ffill-with-carry, NaNs can only be a PREFIX of the batch
+        // (once any value is seen, carry is set forever). bfill therefore
+        // reduces to: fill the prefix with the first valid value (or 0).
+        int nan_prefix = 0;
+        float first_val = NAN;
 
-        // forward pass: raw window average, then ffill (with cross-batch carry)
-        for (int j = 0; j < np; ++j) {
-            const long g0 = phead + j;
-            float sum = 0.f, cnt = 0.f;
-            for (int k = 0; k < win_buckets; ++k) {
-                const long idx = (g0 + k) % G;
-                sum += bs[idx];
-                cnt += bc[idx];
+        for (int j0 = 0; j0 < np; j0 += CHUNK) {
+            const int jn = min(CHUNK, np - j0);
+            const int nload = jn + win_buckets - 1;
+            // coalesced bucket loads into LDS
+            for (int i = lane; i < nload; i += WAVE) {
+                const long idx = (phead + j0 + i) % G;
+                ls[i] = bs[idx];
+                lc[i] = bc[idx];
             }
-            float v = (cnt > 0.f) ? sum / cnt : NAN;
-            if (isnan(v)) v = carry;      // ffill (carry may be NaN)
-            else carry = v;
-            pr[g0 % G] = v;
+            wsync_();
+            // lane j: sliding raw-sample mean of window starting at j0+j
+            if (lane < jn) {
+                float sum = 0.f, cnt = 0.f;
+                for (int k = 0; k < win_buckets; ++k) {
+                    sum += ls[lane + k];
+                    cnt += lc[lane + k];
+                }
+                lv[lane] = (cnt > 0.f) ? sum / cnt : NAN;
+            }
+            wsync_();
+            // lane 0: ffill scan (order-dependent; jn <= 64 steps)
+            if (lane == 0) {
+                for (int j = 0; j < jn; ++j) {
+                    const float v = lv[j];
+                    if (isnan(v)) {
+                        if (isnan(carry)) ++nan_prefix;  // still before 1st value
+                        else lv[j] = carry;
+                    } else {
+                        if (isnan(carry)) first_val = v;
+                        carry = v;
+                    }
+                }
+            }
+            wsync_();
+            if (lane < jn) pr[(phead + j0 + lane) % G] = lv[lane];
+            wsync_();
+            carry = __shfl(carry, 0);
         }
-        last_val[sc] = carry;
-
-        // backward pass: bfill within this batch, then fillna(0)
-        float nxt = NAN;
-        for (int j = np - 1; j >= 0; --j) {
-            const long idx = (phead + j) % G;
-            float v = pr[idx];
-            if (isnan(v)) v = nxt;
-            else nxt = v;
-            pr[idx] = isnan(v) ? 0.f : v;
+        nan_prefix = __shfl(nan_prefix, 0);
+        first_val = __shfl(first_val, 0);
+        if (nan_prefix > 0) {
+            const float fill = isnan(first_val) ? 0.f : first_val;  // bfill|0
+            for (int i = lane; i < nan_prefix; i += WAVE)
+                pr[(phead + i) % G] = fill;
         }
+        if (lane == 0) last_val[sc] = carry;
     }
 }
 
@@ -259,7 +346,7 @@ int tskd_preproc_window_fill(const float* bsum, const float* bcnt, float* proc,
                              float* last_val, int S, int C, int G, long phead,
                              int np, int win_buckets, void* stream) {
     if (np <= 0) return 0;
-    const long nsc = (long)S * C;
+    const long nsc = (long)S * C * WAVE;  // one wave per (stream, channel)
     hipLaunchKernelGGL(window_fill_kernel, dim3(grid_for(nsc, 256)), dim3(256),
                        0, (hipStream_t)stream, bsum, bcnt, proc, last_val, S,
                        C, G, phead, np, win_buckets);
