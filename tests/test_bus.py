@@ -1,0 +1,135 @@
+"""Bus tests: keyed topics, offsets, replay, cross-process delivery."""
+
+import json
+import multiprocessing as mp
+import os
+
+import pytest
+
+from tskd_amd.bus import Bus, Consumer, Producer
+
+
+@pytest.fixture
+def bus(tmp_path):
+    return Bus(str(tmp_path / "bus"))
+
+
+class TestBasic:
+    def test_produce_consume(self, bus):
+        bus.create_topic("HR")
+        c = Consumer(bus, starting="earliest")
+        c.subscribe(["HR"])
+        p = Producer(bus)
+        p.produce_sample("HR", "p000194", 0, 71.5)
+        p.produce_sample("HR", "p000194", 0, 72.0)
+        msgs = c.poll(timeout_ms=1000)
+        assert len(msgs) == 2
+        assert msgs[0].key == b"p000194"
+        ch, val = json.loads(msgs[0].value)
+        assert ch == 0 and val == 71.5
+        assert msgs[0].seq == 0 and msgs[1].seq == 1
+
+    def test_starting_latest_skips_history(self, bus):
+        bus.create_topic("t")
+        p = Producer(bus)
+        p.produce("t", "k", "old")
+        c = Consumer(bus, starting="latest")
+        c.subscribe(["t"])
+        assert c.poll(timeout_ms=10) == []
+        p.produce("t", "k", "new")
+        msgs = c.poll(timeout_ms=1000)
+        assert [m.value for m in msgs] == [b"new"]
+
+    def test_replay_from_offset(self, bus):
+        bus.create_topic("t")
+        p = Producer(bus)
+        for i in range(5):
+            p.produce("t", "k", f"v{i}")
+        c = Consumer(bus, starting="earliest")
+        c.subscribe(["t"])
+        msgs = c.poll()
+        assert len(msgs) == 5
+        # replay from message 2's byte offset
+        c2 = Consumer(bus)
+        c2.seek("t", 0, msgs[2].offset)
+        replay = c2.poll(timeout_ms=500)
+        assert [m.value for m in replay] == [b"v2", b"v3", b"v4"]
+
+    def test_keyed_partitions(self, bus):
+        bus.create_topic("multi", nparts=4)
+        p = Producer(bus)
+        keys = [f"p{i:06d}" for i in range(20)]
+        for k in keys:
+            for j in range(3):
+                p.produce("multi", k, f"{k}:{j}")
+        c = Consumer(bus, starting="earliest")
+        c.subscribe(["multi"])
+        msgs = c.poll(max_msgs=100)
+        assert len(msgs) == 60
+        # same key always lands in the same partition, order preserved
+        by_key = {}
+        for m in msgs:
+            by_key.setdefault(m.key, []).append((m.partition, m.value))
+        for k, lst in by_key.items():
+            parts = {p_ for p_, _ in lst}
+            assert len(parts) == 1
+            assert [v.decode().split(":")[1] for _, v in lst] == ["0", "1", "2"]
+        assert len({lst[0][0] for lst in by_key.values()}) > 1  # spread out
+
+    def test_grow_beyond_initial_capacity(self, bus):
+        bus.create_topic("big")
+        p = Producer(bus)
+        blob = "x" * 10000
+        for i in range(300):  # ~3 MB > 1 MB initial capacity
+            p.produce("big", f"k{i}", blob)
+        c = Consumer(bus, starting="earliest")
+        c.subscribe(["big"])
+        n = 0
+        while True:
+            msgs = c.poll(max_msgs=128, timeout_ms=100)
+            if not msgs:
+                break
+            n += len(msgs)
+        assert n == 300
+
+    def test_producer_callback(self, bus):
+        bus.create_topic("cb")
+        p = Producer(bus)
+        acked = []
+        p.produce("cb", "k", "v", callback=lambda err, info: acked.append((err, info)))
+        assert acked == [(None, ("cb", "k"))]
+
+
+def _producer_proc(bus_dir, topic, n, tag):
+    b = Bus(bus_dir)
+    p = Producer(b)
+    for i in range(n):
+        p.produce(topic, f"key{tag}", f"{tag}:{i}")
+
+
+class TestCrossProcess:
+    def test_two_producer_processes(self, tmp_path):
+        bus_dir = str(tmp_path / "xbus")
+        b = Bus(bus_dir)
+        b.create_topic("xp")
+        c = Consumer(b, starting="earliest")
+        c.subscribe(["xp"])
+        ctx = mp.get_context("spawn")
+        ps = [ctx.Process(target=_producer_proc, args=(bus_dir, "xp", 50, t))
+              for t in ("A", "B")]
+        for p in ps:
+            p.start()
+        for p in ps:
+            p.join(30)
+            assert p.exitcode == 0
+        got = []
+        for _ in range(50):
+            msgs = c.poll(max_msgs=256, timeout_ms=200)
+            got.extend(msgs)
+            if len(got) >= 100:
+                break
+        assert len(got) == 100
+        # per-key FIFO across processes
+        for tag in ("A", "B"):
+            vals = [m.value.decode() for m in got if m.key.decode() == f"key{tag}"]
+            assert vals == [f"{tag}:{i}" for i in range(50)]

@@ -1,0 +1,178 @@
+"""Prediction store / age table / WFDB reader tests."""
+
+import datetime as dt
+import multiprocessing as mp
+import os
+
+import numpy as np
+import pytest
+
+from tskd_amd.store import AgeTable, PredictionStore, patient_str, subject_id
+
+
+class TestPredictionStore:
+    def test_insert_and_latest(self, tmp_path):
+        st = PredictionStore(str(tmp_path / "pred.log"))
+        t0 = dt.datetime(2026, 9, 13, 12, 0, 0)
+        st.insert("p000194", t0, 0.25)
+        st.insert("p000194", t0 + dt.timedelta(minutes=1), 0.5)
+        st.insert("p044083", t0, 0.75)
+        assert st.count() == 3
+        t, score = st.latest("p000194")
+        assert score == 0.5 and t == t0 + dt.timedelta(minutes=1)
+        assert st.latest("p999999") is None
+
+    def test_since_query(self, tmp_path):
+        st = PredictionStore(str(tmp_path / "pred.log"))
+        t0 = dt.datetime(2026, 9, 13, 0, 0, 0)
+        for i in range(10):
+            st.insert(194, t0 + dt.timedelta(hours=i), i / 10)
+        rows = st.since(t0 + dt.timedelta(hours=5))
+        assert len(rows) == 5
+        assert rows[0][0] == "p000194"
+        assert all(r[1] >= t0 + dt.timedelta(hours=5) for r in rows)
+
+    def test_batch_and_growth(self, tmp_path):
+        st = PredictionStore(str(tmp_path / "pred.log"))
+        n = 100_000  # > initial 1 MiB capacity (24 B/record)
+        t0 = dt.datetime(2026, 1, 1)
+        st.insert_batch([194] * n, [t0] * n, np.linspace(0, 1, n).tolist())
+        assert st.count() == n
+        _, last = st.latest(194)
+        assert last == 1.0
+
+    def test_multiprocess_insert(self, tmp_path):
+        path = str(tmp_path / "pred.log")
+        PredictionStore(path)  # init
+        ctx = mp.get_context("spawn")
+        ps = [ctx.Process(target=_store_worker, args=(path, b))
+              for b in (100, 200)]
+        for p in ps:
+            p.start()
+        for p in ps:
+            p.join(30)
+            assert p.exitcode == 0
+        assert PredictionStore(path).count() == 400
+
+    def test_ids(self):
+        assert subject_id("p000194") == 194
+        assert subject_id("p044083-2112-05-04-19-50n") == 44083
+        assert patient_str(194) == "p000194"
+
+
+def _store_worker(path, base):
+    s = PredictionStore(path)
+    for i in range(200):
+        s.insert(base + i % 3, dt.datetime(2026, 1, 1), 0.1)
+
+
+class TestAgeTable:
+    def test_default_and_set(self):
+        at = AgeTable()
+        assert at.get("p000194") == 65.0  # reference default
+        at.set("p000194", 47.5)
+        assert at.get("p000194") == 47.5
+        assert "p000194" in at and "p000001" not in at
+
+    def test_dob_convention(self):
+        at = AgeTable()
+        now = dt.date(2026, 9, 13)
+        at.set_dob(194, dt.date(1976, 9, 13), now=now)
+        # (now - dob).days / 365 — the reference's DATEDIFF/365
+        assert abs(at.get(194) - (now - dt.date(1976, 9, 13)).days / 365.0) < 1e-3
+
+    def test_save_load(self, tmp_path):
+        at = AgeTable()
+        at.set(1, 30.0)
+        at.set(2, 40.0)
+        p = str(tmp_path / "ages.csv")
+        at.save(p)
+        at2 = AgeTable()
+        at2.load(p)
+        assert len(at2) == 2 and at2.get(2) == 40.0
+
+    def test_load_cohort_csv(self, tmp_path, reference_dir):
+        src = os.path.join(reference_dir, "data", "patients_age.csv")
+        if not os.path.exists(src):
+            pytest.skip("cohort csv missing")
+        at = AgeTable()
+        n = at.load_cohort_csv(src, now=dt.date(2112, 5, 23))
+        assert n > 10000
+        assert at.get(194) != 65.0  # p000194 is in the cohort
+
+
+class TestWfdb:
+    def _ref_record(self, reference_dir):
+        p = os.path.join(
+            reference_dir, "data/waveform/physionet.org/files/"
+            "mimic3wdb-matched/1.0/p00/p000194/p000194-2112-05-23-14-34n")
+        if not os.path.exists(p + ".hea"):
+            pytest.skip("reference waveform not present")
+        return p
+
+    def test_header_fields(self, reference_dir):
+        from tskd_amd.io import rdrecord
+        rec = rdrecord(self._ref_record(reference_dir))
+        assert rec.n_sig == 7
+        assert abs(rec.fs - 1 / 60) < 1e-6
+        assert rec.sig_len == 1625
+        assert rec.sig_name == ["HR", "PULSE", "RESP", "SpO2", "NBPSys",
+                                "NBPDias", "NBPMean"]
+        assert rec.p_signal.shape == (1625, 7)
+        bdt = rec.base_datetime
+        assert bdt is not None and bdt.year == 2112 and bdt.hour == 14
+
+    def test_decode_matches_numpy_oracle(self, reference_dir):
+        """Independent numpy fmt-16 decode of the multiplexed .dat."""
+        from tskd_amd.io import rdrecord
+        path = self._ref_record(reference_dir)
+        rec = rdrecord(path)
+        dat = os.path.join(os.path.dirname(path), "3400942n.dat")
+        raw = np.fromfile(dat, dtype="<i2")
+        n = 1625 * 7
+        adc = raw[:n].reshape(1625, 7).astype(np.float64)
+        adc[adc == -32768] = np.nan
+        gains = np.array([10.0, 10.0, 10.0, 10.0, 1.0, 1.0, 1.0])
+        # header gains: HR/PULSE/RESP/SpO2 = 10, NBP* = 1 (baseline 0)
+        hea = open(path + ".hea").read().splitlines()[1:8]
+        gains = np.array([float(l.split()[2].split("/")[0].split("(")[0])
+                          for l in hea])
+        baselines = np.zeros(7)
+        want = (adc - baselines) / gains
+        np.testing.assert_allclose(rec.p_signal, want, rtol=1e-12,
+                                   equal_nan=True)
+
+    def test_channel_selection(self, reference_dir):
+        from tskd_amd.io import rdrecord
+        rec = rdrecord(self._ref_record(reference_dir),
+                       channel_names=["SpO2", "HR"])
+        assert rec.sig_name == ["SpO2", "HR"]
+        assert rec.p_signal.shape == (1625, 2)
+
+    def test_fmt80_and_212_roundtrip(self, tmp_path):
+        """Synthesize tiny fmt-80 and fmt-212 records; decode vs known adc."""
+        from tskd_amd.io import rdrecord
+        d = str(tmp_path)
+        # fmt 80: offset binary, 1 signal
+        adc = np.array([-100, -1, 0, 1, 100, 127], dtype=np.int16)
+        (adc.astype(np.int16) + 128).astype(np.uint8).tofile(f"{d}/r80.dat")
+        open(f"{d}/r80.hea", "w").write(
+            "r80 1 125 6\nr80.dat 80 100(0)/mV 8 0 0 0 0 SIG\n")
+        rec = rdrecord(f"{d}/r80")
+        np.testing.assert_allclose(rec.p_signal[:, 0], adc / 100.0)
+        # fmt 212: two 12-bit samples in 3 bytes, 2 signals interleaved
+        s = np.array([100, -200, 300, -400, 500, -600], dtype=np.int32)
+        b = bytearray()
+        for i in range(0, 6, 2):
+            a0, a1 = int(s[i]) & 0xFFF, int(s[i + 1]) & 0xFFF
+            b += bytes([a0 & 0xFF, ((a0 >> 8) & 0x0F) | (((a1 >> 8) & 0x0F) << 4),
+                        a1 & 0xFF])
+        open(f"{d}/r212.dat", "wb").write(bytes(b))
+        open(f"{d}/r212.hea", "w").write(
+            "r212 2 360 3\nr212.dat 212 200(0)/mV 12 0 0 0 0 A\n"
+            "r212.dat 212 100(0)/mV 12 0 0 0 0 B\n")
+        rec = rdrecord(f"{d}/r212")
+        np.testing.assert_allclose(rec.p_signal[:, 0],
+                                   np.array([100, 300, 500]) / 200.0)
+        np.testing.assert_allclose(rec.p_signal[:, 1],
+                                   np.array([-200, -400, -600]) / 100.0)

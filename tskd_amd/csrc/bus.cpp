@@ -1,0 +1,454 @@
+// tskd bus — first-party broker-less keyed topic bus (C++17, POSIX shm).
+//
+// Replaces the reference's Kafka broker + librdkafka client:
+// keyed topics with per-key partitions, at-least-once delivery, replay from
+// offset. Instead of a JVM broker over docker-bridge TCP, topics are
+// memory-mapped append-only partition logs in a shared directory (tmpfs for
+// node-local IPC, any filesystem for durability): producers append under a
+// process-shared mutex and publish via an atomic committed-bytes counter;
+// consumers poll the counter (acquire) and read — multiple producer and
+// consumer PROCESSES interoperate with no daemon.
+//
+// Reference wire semantics preserved (reference sendStream.py:59-64,
+// utils.py:417-428): topic per channel, messages keyed by patient id,
+// producer acks=all (here: append+commit is the ack; flush() msyncs),
+// consumer startingOffsets latest|earliest, seek to byte offset for replay.
+
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <atomic>
+#include <cstring>
+#include <map>
+#include <mutex>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include <dirent.h>
+#include <fcntl.h>
+#include <pthread.h>
+#include <sys/mman.h>
+#include <sys/stat.h>
+#include <unistd.h>
+
+namespace py = pybind11;
+
+namespace {
+
+constexpr uint64_t MAGIC = 0x54534b44425553ull;  // "TSKDBUS"
+constexpr size_t HDR_SIZE = 4096;
+constexpr size_t INITIAL_CAP = 1 << 20;  // 1 MiB data region, grows 2x
+
+struct PartHeader {
+    uint64_t magic;
+    uint32_t version;
+    uint32_t pad_;
+    pthread_mutex_t mtx;               // process-shared writer lock
+    std::atomic<uint64_t> committed;   // valid data bytes after header
+    std::atomic<uint64_t> next_seq;    // per-partition message sequence
+};
+static_assert(sizeof(PartHeader) <= HDR_SIZE, "header fits");
+
+struct MsgHeader {
+    uint32_t klen;
+    uint32_t vlen;
+    uint64_t seq;
+    int64_t ts_us;
+};
+
+inline size_t rec_size(size_t klen, size_t vlen) {
+    return (sizeof(MsgHeader) + klen + vlen + 7) / 8 * 8;
+}
+
+// A memory-mapped partition log (shared across processes via the file).
+class PartMap {
+public:
+    PartMap(const std::string& path, bool create) : path_(path) {
+        int flags = O_RDWR | (create ? O_CREAT : 0);
+        fd_ = ::open(path.c_str(), flags, 0666);
+        if (fd_ < 0) throw std::runtime_error("bus: cannot open " + path);
+        struct stat st{};
+        fstat(fd_, &st);
+        bool fresh = st.st_size == 0;
+        if (fresh) {
+            if (!create) throw std::runtime_error("bus: empty partition " + path);
+            if (ftruncate(fd_, HDR_SIZE + INITIAL_CAP) != 0)
+                throw std::runtime_error("bus: ftruncate failed");
+        }
+        remap();
+        if (fresh) {
+            // Init header exactly once (file creation is atomic via O_EXCL
+            // in ensure_partition; racing openers see fresh==false).
+            auto* h = hdr();
+            pthread_mutexattr_t at;
+            pthread_mutexattr_init(&at);
+            pthread_mutexattr_setpshared(&at, PTHREAD_PROCESS_SHARED);
+            pthread_mutexattr_setrobust(&at, PTHREAD_MUTEX_ROBUST);
+            pthread_mutex_init(&h->mtx, &at);
+            pthread_mutexattr_destroy(&at);
+            h->committed.store(0);
+            h->next_seq.store(0);
+            h->version = 1;
+            std::atomic_thread_fence(std::memory_order_release);
+            h->magic = MAGIC;
+        } else {
+            // wait for initializer to finish
+            for (int i = 0; i < 100000 && hdr()->magic != MAGIC; ++i) usleep(10);
+            if (hdr()->magic != MAGIC)
+                throw std::runtime_error("bus: bad magic in " + path);
+        }
+    }
+
+    ~PartMap() {
+        if (base_) munmap(base_, mapped_);
+        if (fd_ >= 0) ::close(fd_);
+    }
+
+    PartHeader* hdr() { return reinterpret_cast<PartHeader*>(base_); }
+    uint8_t* data() { return static_cast<uint8_t*>(base_) + HDR_SIZE; }
+
+    uint64_t committed() {
+        maybe_remap();
+        return hdr()->committed.load(std::memory_order_acquire);
+    }
+
+    void append(const std::string& key, const std::string& val, int64_t ts_us) {
+        auto* h = hdr();
+        int rc = pthread_mutex_lock(&h->mtx);
+        if (rc == EOWNERDEAD) pthread_mutex_consistent(&h->mtx);
+        const size_t need = rec_size(key.size(), val.size());
+        uint64_t off = h->committed.load(std::memory_order_relaxed);
+        while (HDR_SIZE + off + need > (uint64_t)file_size_) {
+            if (ftruncate(fd_, (file_size_ - HDR_SIZE) * 2 + HDR_SIZE) != 0) {
+                pthread_mutex_unlock(&h->mtx);
+                throw std::runtime_error("bus: grow failed");
+            }
+            remap();
+            h = hdr();
+        }
+        auto* m = reinterpret_cast<MsgHeader*>(data() + off);
+        m->klen = (uint32_t)key.size();
+        m->vlen = (uint32_t)val.size();
+        m->seq = h->next_seq.fetch_add(1, std::memory_order_relaxed);
+        m->ts_us = ts_us;
+        memcpy(data() + off + sizeof(MsgHeader), key.data(), key.size());
+        memcpy(data() + off + sizeof(MsgHeader) + key.size(), val.data(),
+               val.size());
+        h->committed.store(off + need, std::memory_order_release);
+        pthread_mutex_unlock(&h->mtx);
+    }
+
+    // Read one message at byte offset; returns next offset or 0 if none.
+    bool read(uint64_t off, std::string& key, std::string& val, uint64_t& seq,
+              int64_t& ts_us, uint64_t& next) {
+        const uint64_t lim = committed();
+        if (off + sizeof(MsgHeader) > lim) return false;
+        auto* m = reinterpret_cast<MsgHeader*>(data() + off);
+        const size_t need = rec_size(m->klen, m->vlen);
+        if (off + need > lim) return false;
+        key.assign((char*)(data() + off + sizeof(MsgHeader)), m->klen);
+        val.assign((char*)(data() + off + sizeof(MsgHeader) + m->klen), m->vlen);
+        seq = m->seq;
+        ts_us = m->ts_us;
+        next = off + need;
+        return true;
+    }
+
+    void sync() { msync(base_, mapped_, MS_SYNC); }
+
+private:
+    void remap() {
+        struct stat st{};
+        fstat(fd_, &st);
+        if (base_) munmap(base_, mapped_);
+        file_size_ = st.st_size;
+        mapped_ = st.st_size;
+        base_ = mmap(nullptr, mapped_, PROT_READ | PROT_WRITE, MAP_SHARED,
+                     fd_, 0);
+        if (base_ == MAP_FAILED) throw std::runtime_error("bus: mmap failed");
+    }
+    void maybe_remap() {
+        // another process may have grown the file
+        struct stat st{};
+        fstat(fd_, &st);
+        if (st.st_size != (off_t)file_size_) remap();
+    }
+
+    std::string path_;
+    int fd_ = -1;
+    void* base_ = nullptr;
+    size_t mapped_ = 0;
+    off_t file_size_ = 0;
+};
+
+uint64_t fnv1a(const std::string& s) {
+    uint64_t h = 1469598103934665603ull;
+    for (unsigned char c : s) { h ^= c; h *= 1099511628211ull; }
+    return h;
+}
+
+std::string sanitize(const std::string& t) {
+    std::string o;
+    for (char c : t) o += (isalnum((unsigned char)c) || c == '_' || c == '-')
+                              ? c : '_';
+    return o;
+}
+
+}  // namespace
+
+class Bus {
+public:
+    explicit Bus(const std::string& dir, int default_parts = 1)
+        : dir_(dir), default_parts_(default_parts) {
+        ::mkdir(dir.c_str(), 0777);
+    }
+
+    void create_topic(const std::string& topic, int nparts) {
+        const std::string td = topic_dir(topic);
+        ::mkdir(td.c_str(), 0777);
+        std::string meta = td + "/nparts";
+        int fd = ::open(meta.c_str(), O_WRONLY | O_CREAT | O_EXCL, 0666);
+        if (fd >= 0) {
+            std::string s = std::to_string(nparts);
+            (void)!write(fd, s.data(), s.size());
+            ::close(fd);
+        }
+        for (int p = 0; p < topic_nparts(topic); ++p) ensure_partition(topic, p);
+    }
+
+    int topic_nparts(const std::string& topic) {
+        const std::string meta = topic_dir(topic) + "/nparts";
+        FILE* f = fopen(meta.c_str(), "r");
+        if (!f) return default_parts_;
+        int n = default_parts_;
+        if (fscanf(f, "%d", &n) != 1) n = default_parts_;
+        fclose(f);
+        return n;
+    }
+
+    std::vector<std::string> list_topics() {
+        std::vector<std::string> out;
+        DIR* d = opendir(dir_.c_str());
+        if (!d) return out;
+        while (auto* e = readdir(d)) {
+            if (e->d_name[0] == '.') continue;
+            std::string p = dir_ + "/" + e->d_name;
+            struct stat st{};
+            if (stat(p.c_str(), &st) == 0 && S_ISDIR(st.st_mode))
+                out.push_back(e->d_name);
+        }
+        closedir(d);
+        return out;
+    }
+
+    PartMap* part(const std::string& topic, int p) {
+        const std::string key = topic + "/" + std::to_string(p);
+        std::lock_guard<std::mutex> g(maps_mtx_);
+        auto it = maps_.find(key);
+        if (it != maps_.end()) return it->second.get();
+        ensure_partition(topic, p);
+        auto pm = std::make_unique<PartMap>(part_path(topic, p), true);
+        auto* raw = pm.get();
+        maps_[key] = std::move(pm);
+        return raw;
+    }
+
+    uint64_t end_offset(const std::string& topic, int p) {
+        return part(topic, p)->committed();
+    }
+
+    int partition_for(const std::string& topic, const std::string& key) {
+        const int n = topic_nparts(topic);
+        return key.empty() ? 0 : (int)(fnv1a(key) % (uint64_t)n);
+    }
+
+    std::string topic_dir(const std::string& t) { return dir_ + "/" + sanitize(t); }
+    std::string part_path(const std::string& t, int p) {
+        return topic_dir(t) + "/p" + std::to_string(p) + ".log";
+    }
+    void ensure_partition(const std::string& topic, int p) {
+        ::mkdir(topic_dir(topic).c_str(), 0777);
+        const std::string path = part_path(topic, p);
+        // O_EXCL create makes exactly one process the initializer.
+        int fd = ::open(path.c_str(), O_RDWR | O_CREAT | O_EXCL, 0666);
+        if (fd >= 0) {
+            ::close(fd);
+            PartMap init(path, true);  // initializes header
+        }
+    }
+
+    const std::string& dir() const { return dir_; }
+
+private:
+    std::string dir_;
+    int default_parts_;
+    std::mutex maps_mtx_;
+    std::map<std::string, std::unique_ptr<PartMap>> maps_;
+};
+
+// Producer with reference ack/retry semantics (utils.py:417-422: acks=all,
+// retries=5 — locally an append is durably visible once committed; retries
+// cover transient grow/lock failures).
+class Producer {
+public:
+    Producer(std::shared_ptr<Bus> bus, int retries = 5)
+        : bus_(std::move(bus)), retries_(retries) {}
+
+    void produce(const std::string& topic, const std::string& key,
+                 const std::string& value, int partition = -1,
+                 int64_t ts_us = -1) {
+        if (ts_us < 0) {
+            struct timespec ts{};
+            clock_gettime(CLOCK_REALTIME, &ts);
+            ts_us = (int64_t)ts.tv_sec * 1000000 + ts.tv_nsec / 1000;
+        }
+        const int p = partition >= 0 ? partition
+                                     : bus_->partition_for(topic, key);
+        for (int attempt = 0;; ++attempt) {
+            try {
+                bus_->part(topic, p)->append(key, value, ts_us);
+                ++produced_;
+                return;
+            } catch (const std::exception&) {
+                if (attempt >= retries_) throw;
+                usleep(1000 << attempt);
+            }
+        }
+    }
+
+    void flush(const std::string& topic = "") {
+        // acks=all analog: force page-cache sync of produced partitions.
+        for (auto& t : topic.empty() ? bus_->list_topics()
+                                     : std::vector<std::string>{topic})
+            for (int p = 0; p < bus_->topic_nparts(t); ++p)
+                bus_->part(t, p)->sync();
+    }
+
+    uint64_t produced() const { return produced_; }
+
+private:
+    std::shared_ptr<Bus> bus_;
+    int retries_;
+    uint64_t produced_ = 0;
+};
+
+struct Message {
+    std::string topic;
+    int partition;
+    uint64_t offset;      // byte offset of this record (seek target)
+    uint64_t next_offset; // byte offset after this record
+    uint64_t seq;
+    int64_t ts_us;
+    py::bytes key() const { return py::bytes(key_); }
+    py::bytes value() const { return py::bytes(val_); }
+    std::string key_, val_;
+};
+
+class Consumer {
+public:
+    Consumer(std::shared_ptr<Bus> bus, std::string starting = "latest")
+        : bus_(std::move(bus)), starting_(std::move(starting)) {}
+
+    void subscribe(const std::vector<std::string>& topics) {
+        for (auto& t : topics) {
+            const int n = bus_->topic_nparts(t);
+            for (int p = 0; p < n; ++p) {
+                const std::string k = t + "/" + std::to_string(p);
+                if (pos_.count(k)) continue;
+                pos_[k] = starting_ == "earliest" ? 0
+                                                  : bus_->end_offset(t, p);
+                parts_.push_back({t, p});
+            }
+        }
+    }
+
+    void seek(const std::string& topic, int p, uint64_t offset) {
+        pos_[topic + "/" + std::to_string(p)] = offset;
+        for (auto& q : parts_)
+            if (q.first == topic && q.second == p) return;
+        parts_.push_back({topic, p});
+    }
+
+    std::vector<Message> poll(int max_msgs = 256, int timeout_ms = 0) {
+        std::vector<Message> out;
+        const int64_t deadline = now_us() + (int64_t)timeout_ms * 1000;
+        while (true) {
+            for (auto& [t, p] : parts_) {
+                const std::string k = t + "/" + std::to_string(p);
+                auto* pm = bus_->part(t, p);
+                uint64_t off = pos_[k];
+                Message m;
+                while ((int)out.size() < max_msgs &&
+                       pm->read(off, m.key_, m.val_, m.seq, m.ts_us,
+                                m.next_offset)) {
+                    m.topic = t;
+                    m.partition = p;
+                    m.offset = off;
+                    off = m.next_offset;
+                    out.push_back(m);
+                    m = Message();
+                }
+                pos_[k] = off;
+                if ((int)out.size() >= max_msgs) return out;
+            }
+            if (!out.empty() || now_us() >= deadline) return out;
+            {
+                py::gil_scoped_release rel;
+                usleep(1000);
+            }
+        }
+    }
+
+    std::map<std::string, uint64_t> positions() const { return pos_; }
+
+private:
+    static int64_t now_us() {
+        struct timespec ts{};
+        clock_gettime(CLOCK_MONOTONIC, &ts);
+        return (int64_t)ts.tv_sec * 1000000 + ts.tv_nsec / 1000;
+    }
+    std::shared_ptr<Bus> bus_;
+    std::string starting_;
+    std::map<std::string, uint64_t> pos_;
+    std::vector<std::pair<std::string, int>> parts_;
+};
+
+PYBIND11_MODULE(_tskd_bus, m) {
+    m.doc() = "tskd broker-less keyed topic bus (Kafka replacement)";
+    py::class_<Bus, std::shared_ptr<Bus>>(m, "Bus")
+        .def(py::init<const std::string&, int>(), py::arg("dir"),
+             py::arg("default_parts") = 1)
+        .def("create_topic", &Bus::create_topic, py::arg("topic"),
+             py::arg("nparts") = 1)
+        .def("topic_nparts", &Bus::topic_nparts)
+        .def("list_topics", &Bus::list_topics)
+        .def("end_offset", &Bus::end_offset)
+        .def("partition_for", &Bus::partition_for)
+        .def_property_readonly("dir", &Bus::dir);
+    py::class_<Producer>(m, "Producer")
+        .def(py::init<std::shared_ptr<Bus>, int>(), py::arg("bus"),
+             py::arg("retries") = 5)
+        .def("produce", &Producer::produce, py::arg("topic"), py::arg("key"),
+             py::arg("value"), py::arg("partition") = -1,
+             py::arg("ts_us") = -1)
+        .def("flush", &Producer::flush, py::arg("topic") = "")
+        .def_property_readonly("produced", &Producer::produced);
+    py::class_<Message>(m, "Message")
+        .def_readonly("topic", &Message::topic)
+        .def_readonly("partition", &Message::partition)
+        .def_readonly("offset", &Message::offset)
+        .def_readonly("next_offset", &Message::next_offset)
+        .def_readonly("seq", &Message::seq)
+        .def_readonly("ts_us", &Message::ts_us)
+        .def_property_readonly("key", &Message::key)
+        .def_property_readonly("value", &Message::value);
+    py::class_<Consumer>(m, "Consumer")
+        .def(py::init<std::shared_ptr<Bus>, std::string>(), py::arg("bus"),
+             py::arg("starting") = "latest")
+        .def("subscribe", &Consumer::subscribe)
+        .def("seek", &Consumer::seek)
+        .def("poll", &Consumer::poll, py::arg("max_msgs") = 256,
+             py::arg("timeout_ms") = 0)
+        .def("positions", &Consumer::positions);
+}

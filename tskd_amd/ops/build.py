@@ -54,11 +54,53 @@ def build(name: str, force: bool = False, verbose: bool = True) -> str:
     return out
 
 
+# Host-side C++ pybind11 extensions (bus, store, wfdb — no GPU code).
+PKG_DIR = os.path.dirname(OPS_DIR)
+HOST_CSRC = os.path.join(PKG_DIR, "csrc")
+HOST_MODULES = {
+    "_tskd_bus": "bus.cpp",
+    "_tskd_store": "store.cpp",
+    "_tskd_wfdb": "wfdb.cpp",
+}
+
+
+def _ext_suffix() -> str:
+    import sysconfig
+    return sysconfig.get_config_var("EXT_SUFFIX") or ".so"
+
+
+def host_lib_path(name: str) -> str:
+    return os.path.join(PKG_DIR, name + _ext_suffix())
+
+
+def build_host(name: str, force: bool = False, verbose: bool = True) -> str:
+    import pybind11
+    import sysconfig
+    src = os.path.join(HOST_CSRC, HOST_MODULES[name])
+    out = host_lib_path(name)
+    if not force and os.path.exists(out) and \
+            os.path.getmtime(out) >= os.path.getmtime(src):
+        return out
+    cmd = [
+        "g++", "-O2", "-std=c++17", "-shared", "-fPIC",
+        f"-I{pybind11.get_include()}",
+        f"-I{sysconfig.get_paths()['include']}",
+        src, "-o", out, "-lpthread",
+    ]
+    if verbose:
+        print("[tskd build]", " ".join(cmd), file=sys.stderr)
+    subprocess.run(cmd, check=True)
+    return out
+
+
 def build_all(force: bool = False) -> list:
     outs = []
     for name, srcs in SOURCES.items():
         if all(os.path.exists(os.path.join(CSRC, s)) for s in srcs):
             outs.append(build(name, force=force))
+    for name, src in HOST_MODULES.items():
+        if os.path.exists(os.path.join(HOST_CSRC, src)):
+            outs.append(build_host(name, force=force))
     return outs
 
 
