@@ -1,0 +1,160 @@
+"""End-to-end pipeline integration: sendStream -> bus -> processStream ->
+call-stream -> predictStream -> prediction store -> dashboard API.
+
+Exercises the reference's full dataflow in one process with a
+synthetic WFDB numerics record at the real MIMIC rate (fs = 1/60 Hz).
+"""
+
+import json
+import os
+import threading
+import time
+
+import numpy as np
+import pytest
+import torch
+
+from tskd_amd.bus import Bus, Consumer
+from tskd_amd.cli.predictstream import PredictStream
+from tskd_amd.cli.processstream import ProcessStream
+from tskd_amd.cli.sendstream import send_csv_data, send_record_data
+from tskd_amd.config import GlobalConfig
+from tskd_amd.store import AgeTable, PredictionStore
+
+
+def _write_wfdb_record(root: str, record: str, sig_names, fs: float,
+                       nsamp: int, seed=0):
+    """Synthesize a multiplexed fmt-16 record under the reference's
+    pXX/pXXXXXX directory layout."""
+    pid = record[0:7]
+    d = os.path.join(root, pid[0:3], pid)
+    os.makedirs(d, exist_ok=True)
+    rng = np.random.default_rng(seed)
+    nsig = len(sig_names)
+    adc = (rng.normal(700, 80, size=(nsamp, nsig))).astype(np.int16)
+    adc.tofile(os.path.join(d, f"{record}.dat"))
+    with open(os.path.join(d, f"{record}.hea"), "w") as f:
+        f.write(f"{record} {nsig} {fs:.13f} {nsamp} 14:34:23.221 23/05/2112\n")
+        for name in sig_names:
+            f.write(f"{record}.dat 16 10/bpm 16 0 736 0 0 {name}\n")
+    return adc / 10.0
+
+
+@pytest.fixture
+def cfg(tmp_path):
+    c = GlobalConfig()
+    c.wavef_path = str(tmp_path / "wavef")
+    c.channel_names = ["HR", "RESP", "PULSE", "SpO2"]
+    c.patient_records = ["p000194-test"]
+    return c
+
+
+class TestEndToEnd:
+    def test_full_pipeline(self, tmp_path, cfg):
+        bus = Bus(str(tmp_path / "bus"))
+        nsamp = 32  # 32 min of stream time at fs=1/60
+        phys = _write_wfdb_record(cfg.wavef_path, "p000194-test",
+                                  cfg.channel_names, 1 / 60, nsamp)
+
+        # stage 1: replay producer (speed so high sleeps are ~0)
+        sent = send_record_data(bus, "p000194-test", None, speed=1e6,
+                                frequency=1.0, cfg=cfg)
+        assert sent == nsamp * len(cfg.channel_names)
+
+        # stage 2: preprocessor (from earliest, one micro-batch)
+        ps = ProcessStream(bus, cfg, max_streams=4, device="cpu",
+                           starting="earliest")
+        emitted = ps.trigger()
+        assert emitted > 0
+        assert ps.engine.nproc >= 120  # enough grid for a model window
+
+        # the call-stream wire format: key pid_chan, value JSON float array
+        cc = Consumer(bus, starting="earliest")
+        cc.subscribe(["call-stream"])
+        msgs = cc.poll(max_msgs=1024, timeout_ms=500)
+        keys = {m.key.decode() for m in msgs}
+        assert "p000194_0" in keys
+        pts = json.loads(msgs[0].value)
+        assert isinstance(pts, list) and len(pts) == ps.engine.nproc
+
+        # stage 3: inference -> store
+        store = PredictionStore(str(tmp_path / "pred.log"))
+        ages = AgeTable()
+        ages.set("p000194", 47.0)
+        pr = PredictStream(bus, cfg, store, ages, device="cpu",
+                           starting="earliest")
+        n = pr.trigger()
+        assert n == 1
+        assert store.count() == 1
+        t, risk = store.latest("p000194")
+        assert 0.0 <= risk <= 1.0
+
+        # grid values should reflect the synthetic signal's scale
+        hr = np.array(json.loads(
+            [m for m in msgs if m.key == b"p000194_0"][0].value))
+        assert abs(hr[hr != 0].mean() - phys[:, 0].mean()) < 5.0
+
+    def test_incremental_triggers_produce_more_predictions(self, tmp_path, cfg):
+        bus = Bus(str(tmp_path / "bus"))
+        _write_wfdb_record(cfg.wavef_path, "p000194-test", cfg.channel_names,
+                           1 / 60, 60)
+        store = PredictionStore(str(tmp_path / "pred.log"))
+        ps = ProcessStream(bus, cfg, max_streams=4, starting="earliest")
+        pr = PredictStream(bus, cfg, store, device="cpu", starting="earliest")
+        send_record_data(bus, "p000194-test", None, 1e6, 1.0, cfg)
+        total = 0
+        for _ in range(3):
+            ps.trigger()
+            total += pr.trigger()
+        assert total >= 1
+        assert store.count() == total
+
+    def test_csv_replay_mode(self, tmp_path, cfg):
+        from tskd_amd.cli.makedata import make_data
+        p = str(tmp_path / "data.csv")
+        n = make_data(p, seed=42, hours=0.25)
+        bus = Bus(str(tmp_path / "bus"))
+        bus.create_topic("data")
+        c = Consumer(bus, starting="earliest")
+        c.subscribe(["data"])
+        sent = send_csv_data(bus, p, "data", "csv", speed=1e9, cfg=cfg)
+        assert sent == n
+        msgs = c.poll(max_msgs=10, timeout_ms=500)
+        assert len(msgs) == 10
+        ch, val = json.loads(msgs[0].value)
+        assert ch == 0 and 0 <= val <= 100
+
+    def test_makedata_deterministic(self, tmp_path):
+        from tskd_amd.cli.makedata import make_data
+        p1, p2 = str(tmp_path / "a.csv"), str(tmp_path / "b.csv")
+        make_data(p1, seed=42, hours=0.1)
+        make_data(p2, seed=42, hours=0.1)
+        assert open(p1).read() == open(p2).read()
+
+
+class TestDashboard:
+    def test_api_endpoints(self, tmp_path, cfg):
+        fastapi = pytest.importorskip("fastapi")  # noqa: F841
+        from fastapi.testclient import TestClient
+
+        from tskd_amd.cli.plotdata import DashState, build_app
+        bus = Bus(str(tmp_path / "bus"))
+        _write_wfdb_record(cfg.wavef_path, "p000194-test", cfg.channel_names,
+                           1 / 60, 20)
+        store = PredictionStore(str(tmp_path / "pred.log"))
+        store.insert("p000194", time.time(), 0.42)
+        state = DashState(bus, cfg, store, starting="earliest")
+        send_record_data(bus, "p000194-test", None, 1e6, 1.0, cfg)
+        t = threading.Thread(target=state.pump, daemon=True)
+        t.start()
+        time.sleep(0.6)
+        state._stop = True
+        client = TestClient(build_app(state))
+        assert "tskd" in client.get("/").text
+        pats = client.get("/api/patients").json()
+        assert "p000194" in pats
+        raw = client.get("/api/raw/p000194").json()
+        assert "0" in raw and len(raw["0"]) == 20
+        preds = client.get("/api/predictions").json()
+        assert preds and preds[0]["patient"] == "p000194"
+        assert abs(preds[0]["risk"] - 0.42) < 1e-6
