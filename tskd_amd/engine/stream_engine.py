@@ -130,8 +130,13 @@ class StreamEngine:
             is_bf16 = 1 if raw.dtype == torch.bfloat16 else 0
             if not is_bf16 and raw.dtype != torch.float32:
                 raw = raw.float()
-            cm = torch.tensor(list(chan_map), dtype=torch.int32,
-                              device=self.device)
+            key = tuple(chan_map)
+            cm = getattr(self, "_cm_cache", {}).get(key)
+            if cm is None:
+                cm = torch.tensor(list(chan_map), dtype=torch.int32,
+                                  device=self.device)
+                self._cm_cache = getattr(self, "_cm_cache", {})
+                self._cm_cache[key] = cm
             rc = lib.tskd_preproc_ingest_dense(
                 ctypes.c_void_p(raw.data_ptr()), is_bf16,
                 ctypes.c_void_p(self.bsum.data_ptr()),

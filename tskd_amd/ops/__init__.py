@@ -135,7 +135,9 @@ class MyCNNEngine:
         out = torch.empty(s, n, dtype=torch.float32, device=feat.device)
         age_ptr = ctypes.c_void_p(0)
         if age is not None:
-            age = age.to(torch.float32).expand(s, n).contiguous()
+            if age.dtype != torch.float32 or age.shape != (s, n) \
+                    or not age.is_contiguous():
+                age = age.to(torch.float32).expand(s, n).contiguous()
             age_ptr = ctypes.c_void_p(age.data_ptr())
         rc = lib.tskd_lstm_head_fwd(
             ctypes.c_void_p(feat.data_ptr()), age_ptr,
