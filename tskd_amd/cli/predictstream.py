@@ -145,6 +145,8 @@ def main(argv=None) -> None:
     ap.add_argument("--starting", default="latest",
                     choices=["latest", "earliest"])
     ap.add_argument("--max-triggers", type=int, default=0)
+    ap.add_argument("--offsets-file", default=None,
+                    help="persist/restore consumer offsets (resume-on-restart)")
     args = ap.parse_args(argv)
 
     bus = Bus(args.bus_dir)
@@ -164,10 +166,18 @@ def main(argv=None) -> None:
     trigger_period = cfg.predict_slide_s / args.speed
     stop = []
     signal.signal(signal.SIGTERM, lambda *a: stop.append(1))
+    if args.offsets_file:
+        from tskd_amd.parallel.supervisor import restore_consumer, save_offsets
+        restored = restore_consumer(ps.consumer, args.offsets_file)
+        if restored:
+            log.info("resumed %d partition offsets from %s", restored,
+                     args.offsets_file)
     n = 0
     while not stop:
         t0 = time.time()
         ps.trigger()
+        if args.offsets_file:
+            save_offsets(args.offsets_file, ps.consumer.positions())
         n += 1
         if args.max_triggers and n >= args.max_triggers:
             break
