@@ -1,0 +1,149 @@
+"""HIP training-path numerics vs torch autograd (the fp32 oracle).
+
+Gradient parity covers the whole K10-K12 chain: BCE-with-logits(pos_weight)
+-> head + age gate -> reverse BPTT through the 2-layer batch-as-time LSTM ->
+maxpool/tanh/conv backward. The oracle maps torch autograd grads onto the
+packed layout (d bias_ih == d bias_hh == d combined bias).
+"""
+
+import numpy as np
+import pytest
+import torch
+
+from tskd_amd.models import build_model
+from tskd_amd.ops.pack import pack_offsets, pack_weights
+
+
+def _torch_loss_and_grads(model, x, age, y, pos_weight):
+    """Oracle: autograd over the per-sequence forward; packed-layout grads."""
+    model.zero_grad()
+    logits = []
+    for s in range(x.shape[0]):
+        logits.append(model(x[s], age[s]))
+    logits = torch.cat(logits)
+    crit = torch.nn.BCEWithLogitsLoss(pos_weight=torch.tensor(pos_weight))
+    loss = crit(logits, y.reshape(-1))
+    loss.backward()
+    g = {n: p.grad.detach().clone() for n, p in model.named_parameters()
+         if p.grad is not None}
+    o = pack_offsets(model)
+    packed = torch.zeros(o["npack"])
+
+    def put(name, t):
+        a, b = o[name]
+        packed[a:b] = t.reshape(-1)
+
+    put("w1", g["conv1.weight"]); put("b1", g["conv1.bias"])
+    put("w2", g["conv2.weight"]); put("b2", g["conv2.bias"])
+    put("wih1", g["lstm.weight_ih_l0"]); put("whh1", g["lstm.weight_hh_l0"])
+    put("bl1", g["lstm.bias_ih_l0"])  # == bias_hh grad
+    put("wih2", g["lstm.weight_ih_l1"]); put("whh2", g["lstm.weight_hh_l1"])
+    put("bl2", g["lstm.bias_ih_l1"])
+    put("outw", g["out.weight"]); put("outb", g["out.bias"])
+    assert torch.allclose(g["lstm.bias_ih_l0"], g["lstm.bias_hh_l0"])
+    return float(loss.item()), packed
+
+
+@pytest.mark.gpu
+class TestHipTraining:
+    @pytest.mark.parametrize("variant,pos_weight", [("MyCNN5", 1.0),
+                                                    ("MyCNN5", 3.5),
+                                                    ("MyCNN2", 1.0),
+                                                    ("MyCNN4", 2.0)])
+    def test_grad_parity(self, variant, pos_weight):
+        from tskd_amd.train.hip_trainer import MyCNNHipTrainer
+        torch.manual_seed(7)
+        model = build_model(variant).eval()
+        S, B = 3, 24
+        x = torch.randn(S, B, model.IN_CHANNELS, 120)
+        age = torch.full((S, B), 55.0)
+        y = (torch.rand(S, B) < 0.3).float()
+        ref_loss, ref_g = _torch_loss_and_grads(model, x, age, y, pos_weight)
+
+        tr = MyCNNHipTrainer(model, device="cuda", pos_weight=pos_weight)
+        loss = tr.forward_backward(x.cuda(), age.cuda(), y.cuda())
+        torch.cuda.synchronize()
+        assert abs(loss - ref_loss) / max(abs(ref_loss), 1e-6) < 1e-3
+        got = tr.grads.cpu()
+        # compare per-slice for diagnosable failures
+        o = tr.offsets
+        for name in ("outw", "outb", "wih2", "whh2", "bl2", "wih1", "whh1",
+                     "bl1", "w2", "b2", "w1", "b1"):
+            a, b = o[name]
+            torch.testing.assert_close(
+                got[a:b], ref_g[a:b], rtol=2e-3, atol=2e-4,
+                msg=lambda m, n=name: f"slice {n}: {m}")
+
+    def test_grad_accumulation_is_additive(self):
+        from tskd_amd.train.hip_trainer import MyCNNHipTrainer
+        torch.manual_seed(8)
+        model = build_model("MyCNN5").eval()
+        tr = MyCNNHipTrainer(model, device="cuda")
+        x = torch.randn(2, 16, 10, 120).cuda()
+        age = torch.full((2, 16), 65.0).cuda()
+        y = (torch.rand(2, 16) < 0.5).float().cuda()
+        tr.forward_backward(x, age, y)
+        torch.cuda.synchronize()
+        g1 = tr.grads.clone()
+        tr.forward_backward(x, age, y)
+        torch.cuda.synchronize()
+        torch.testing.assert_close(tr.grads, 2 * g1, rtol=1e-4, atol=1e-6)
+
+    def test_adam_matches_torch(self):
+        from tskd_amd.train.hip_trainer import _load_train_lib, _p, _sp
+        import ctypes
+        lib = _load_train_lib()
+        torch.manual_seed(9)
+        n = 5242
+        p0 = torch.randn(n)
+        g0 = torch.randn(n)
+        # torch oracle
+        p_t = p0.clone().requires_grad_(True)
+        opt = torch.optim.Adam([p_t], lr=1e-3)
+        for step in range(3):
+            p_t.grad = g0 * (step + 1)
+            opt.step()
+        # hip
+        p_h = p0.clone().cuda()
+        m = torch.zeros(n).cuda()
+        v = torch.zeros(n).cuda()
+        for step in range(3):
+            g = (g0 * (step + 1)).cuda()
+            rc = lib.tskd_train_adam(_p(p_h), _p(g), _p(m), _p(v), n,
+                                     ctypes.c_float(1e-3),
+                                     ctypes.c_float(0.9),
+                                     ctypes.c_float(0.999),
+                                     ctypes.c_float(1e-8), step + 1, _sp())
+            assert rc == 0
+            torch.cuda.synchronize()
+            assert (g == 0).all()  # grads zeroed by the fused kernel
+        torch.testing.assert_close(p_h.cpu(), p_t.detach(), rtol=1e-5,
+                                   atol=1e-6)
+
+    def test_training_reduces_loss(self):
+        from tskd_amd.train.data import make_synthetic_labeled_windows
+        from tskd_amd.train.hip_trainer import MyCNNHipTrainer
+        torch.manual_seed(10)
+        model = build_model("MyCNN5").eval()
+        tr = MyCNNHipTrainer(model, device="cuda", lr=3e-3)
+        x, age, y = make_synthetic_labeled_windows(512, seed=4)
+        x = torch.from_numpy(x).reshape(8, 64, 10, 120).cuda()
+        age = torch.from_numpy(age).reshape(8, 64).cuda()
+        y = torch.from_numpy(y).reshape(8, 64).cuda()
+        losses = [tr.step(x, age, y) for _ in range(15)]
+        assert losses[-1] < losses[0] * 0.9, losses
+
+    def test_export_model_roundtrip(self):
+        from tskd_amd.train.hip_trainer import MyCNNHipTrainer
+        torch.manual_seed(11)
+        model = build_model("MyCNN5").eval()
+        tr = MyCNNHipTrainer(model, device="cuda", lr=1e-2)
+        x = torch.randn(2, 32, 10, 120).cuda()
+        age = torch.full((2, 32), 65.0).cuda()
+        y = (torch.rand(2, 32) < 0.5).float().cuda()
+        tr.step(x, age, y)
+        m2 = tr.export_model()
+        # exported torch model reproduces the packed forward
+        wpack2 = pack_weights(m2)
+        torch.testing.assert_close(wpack2, tr.wpack.cpu(), rtol=1e-6,
+                                   atol=1e-7)

@@ -58,3 +58,57 @@ def pack_weights(model) -> torch.Tensor:
         sd["out.bias"].reshape(-1),
     ]
     return torch.cat(parts).contiguous()
+
+
+def pack_offsets(model) -> dict:
+    """Python mirror of the Geom<> pack offsets (see mycnn_kernels.hip)."""
+    cin = int(model.IN_CHANNELS)
+    k1 = int(model.CONV1_K)
+    lin = int(model.LSTM_IN)
+    o = {}
+    o["w1"] = (0, 4 * cin * k1)
+    o["b1"] = (o["w1"][1], o["w1"][1] + 4)
+    o["w2"] = (o["b1"][1], o["b1"][1] + 20)
+    o["b2"] = (o["w2"][1], o["w2"][1] + 1)
+    o["wih1"] = (o["b2"][1], o["b2"][1] + 64 * lin)
+    o["whh1"] = (o["wih1"][1], o["wih1"][1] + 64 * 16)
+    o["bl1"] = (o["whh1"][1], o["whh1"][1] + 64)
+    o["wih2"] = (o["bl1"][1], o["bl1"][1] + 64 * 16)
+    o["whh2"] = (o["wih2"][1], o["wih2"][1] + 64 * 16)
+    o["bl2"] = (o["whh2"][1], o["whh2"][1] + 64)
+    o["outw"] = (o["bl2"][1], o["bl2"][1] + 16)
+    o["outb"] = (o["outw"][1], o["outw"][1] + 1)
+    o["npack"] = o["outb"][1]
+    return o
+
+
+def unpack_weights_into(model, wpack) -> None:
+    """Write a packed parameter vector back into a MyCNN module.
+
+    The combined LSTM bias is split half/half between bias_ih and bias_hh
+    (only the sum enters the forward math — documented parameterization
+    choice of the packed trainer)."""
+    import torch as _t
+    o = pack_offsets(model)
+    w = wpack.detach().float().cpu()
+
+    def sl(name):
+        a, b = o[name]
+        return w[a:b]
+
+    cin, k1, lin = int(model.IN_CHANNELS), int(model.CONV1_K), int(model.LSTM_IN)
+    with _t.no_grad():
+        model.conv1.weight.copy_(sl("w1").view(4, cin, k1))
+        model.conv1.bias.copy_(sl("b1"))
+        model.conv2.weight.copy_(sl("w2").view(1, 4, 5))
+        model.conv2.bias.copy_(sl("b2"))
+        model.lstm.weight_ih_l0.copy_(sl("wih1").view(64, lin))
+        model.lstm.weight_hh_l0.copy_(sl("whh1").view(64, 16))
+        model.lstm.bias_ih_l0.copy_(sl("bl1") * 0.5)
+        model.lstm.bias_hh_l0.copy_(sl("bl1") * 0.5)
+        model.lstm.weight_ih_l1.copy_(sl("wih2").view(64, 16))
+        model.lstm.weight_hh_l1.copy_(sl("whh2").view(64, 16))
+        model.lstm.bias_ih_l1.copy_(sl("bl2") * 0.5)
+        model.lstm.bias_hh_l1.copy_(sl("bl2") * 0.5)
+        model.out.weight.copy_(sl("outw").view(1, 16))
+        model.out.bias.copy_(sl("outb"))
