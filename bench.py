@@ -13,6 +13,9 @@ Two modes (BASELINE.json configs):
   --mode infer (config 2): model-path throughput — S sequences x B=1024-window
       batches through the fused conv+LSTM path (reference batch semantics
       incl. the LSTM batch-axis-as-time quirk).
+  --mode train (config 5): training step — fused conv/LSTM forward-with-stash,
+      BCEWithLogits(pos_weight), BPTT backward, DP gradient all-reduce over
+      RCCL, fused Adam (fp32, reference recipe).
 
 value = whole-job windows/s across all ranks. Data is synthetic (no network)
 with random-init weights; the reference publishes no quantitative numbers
@@ -65,7 +68,8 @@ def main() -> None:
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=15)
-    p.add_argument("--mode", default="pipeline", choices=["pipeline", "infer"])
+    p.add_argument("--mode", default="pipeline",
+                   choices=["pipeline", "infer", "train"])
     p.add_argument("--streams", type=int, default=16384,
                    help="[pipeline] concurrent patient streams per GPU")
     p.add_argument("--seqs", type=int, default=2048,
@@ -97,7 +101,25 @@ def main() -> None:
     eng = MyCNNEngine(model, device=device)
     dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
 
-    if args.mode == "infer":
+    if args.mode == "train":
+        from tskd_amd.train.hip_trainer import MyCNNHipTrainer
+        S, B = max(args.seqs // 8, 1), 64  # reference batch size 64
+        tr = MyCNNHipTrainer(model, device=device, lr=1e-5, pos_weight=3.0)
+        x = torch.randn(S, B, 10, 120, device=device)
+        x[:, :, 8:, :] = 0
+        age = torch.full((S, B), 65.0, device=device)
+        y = (torch.rand(S, B, device=device) < 0.3).float()
+
+        def step():
+            tr.step(x, age, y)
+
+        windows_per_step = S * B
+        cfg = {"model": args.variant, "global_batch": windows_per_step * world,
+               "seq_len": 120, "batch_per_seq": B, "seqs_per_gpu": S,
+               "parallelism": f"dp{world}", "mode": "train",
+               "optimizer": "fused_adam", "loss": "bce_pos_weight"}
+        args.dtype = "fp32"  # training runs fp32 like the reference
+    elif args.mode == "infer":
         S, B = args.seqs, args.batch
         x = torch.randn(S, B, 10, 120, device=device, dtype=dtype)
         x[:, :, 8:, :] = 0  # 8 active of 10 wire channels
