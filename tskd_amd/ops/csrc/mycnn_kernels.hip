@@ -112,9 +112,14 @@ __device__ __forceinline__ float tanhf_(float x) {
 
 // ---------------------------------------------------------------------------
 // Shared epilogue: pool1 -> conv2 -> tanh -> pool2 -> feature store.
-// Reads lds_c1 (4 x C1 tanh'd conv1 rows), uses lds_p1/lds_c2 scratch.
+// Reads lds_c1 (4 x C1 conv1 rows), uses lds_p1/lds_c2 scratch.
+//
+// TANH_AT_POOL: tanh is monotone, so tanh(max(a,b)) == max(tanh(a), tanh(b))
+// — the MFMA path stores RAW conv1 pre-activations (bias folded) and applies
+// tanh AFTER each pool, cutting transcendental count ~2x (444->220, 51->25)
+// and keeping every lane active.
 // ---------------------------------------------------------------------------
-template <class G>
+template <class G, bool TANH_AT_POOL>
 __device__ __forceinline__ void conv_tail(
     int lane, const float* __restrict__ lds_w, float* c1w, float* p1w,
     float* c2w, float* __restrict__ feat, long win)
@@ -128,10 +133,11 @@ __device__ __forceinline__ void conv_tail(
         float m = src[0];
         #pragma unroll
         for (int k = 1; k < G::PK; ++k) m = fmaxf(m, src[k]);
-        p1w[o] = m;
+        p1w[o] = TANH_AT_POOL ? tanhf_(m) : m;
     }
     wave_sync();
-    // conv2 + tanh (k=5, 4 in-ch)
+    // conv2 (+ tanh here only when pool1 already produced tanh'd inputs and
+    // pool2 will apply the second tanh)
     for (int s = lane; s < G::C2; s += WAVE) {
         float acc = b2;
         #pragma unroll
@@ -141,7 +147,7 @@ __device__ __forceinline__ void conv_tail(
             for (int k = 0; k < 5; ++k)
                 acc = fmaf(w2[c * 5 + k], pr[k], acc);
         }
-        c2w[s] = tanhf_(acc);
+        c2w[s] = TANH_AT_POOL ? acc : tanhf_(acc);
     }
     wave_sync();
     // pool2 -> feature vector
@@ -150,7 +156,7 @@ __device__ __forceinline__ void conv_tail(
         float m = src[0];
         #pragma unroll
         for (int k = 1; k < G::PK; ++k) m = fmaxf(m, src[k]);
-        feat[win * G::LIN + q] = m;
+        feat[win * G::LIN + q] = TANH_AT_POOL ? tanhf_(m) : m;
     }
     wave_sync();  // before the next window overwrites scratch
 }
@@ -200,8 +206,8 @@ __global__ __launch_bounds__(WG_THREADS) void conv_stack_kernel(
             lds_c1[wave][o] = tanhf_(acc);
         }
         wave_sync();
-        conv_tail<G>(lane, lds_w, lds_c1[wave], lds_p1[wave], lds_c2[wave],
-                     feat, win);
+        conv_tail<G, false>(lane, lds_w, lds_c1[wave], lds_p1[wave],
+                            lds_c2[wave], feat, win);
     }
 }
 
@@ -237,14 +243,16 @@ __global__ __launch_bounds__(WG_THREADS) void conv_stack_mfma_kernel(
     const float* b1 = lds_w + G::OB1;
     unsigned short* xt = lds_xt[wave];
 
-    // Preload B fragments (per-lane, all k-steps) — 16B-aligned global loads.
+    // Preload B fragments (per-lane, all k-steps): one dwordx4 per step.
     bf16x8 bfr[G::KSTEPS];
-    union BU { unsigned short u[8]; bf16x8 v; };
+    union BU { unsigned int d[4]; unsigned short u[8]; bf16x8 v; };
     #pragma unroll
     for (int st = 0; st < G::KSTEPS; ++st) {
         BU bu;
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) bu.u[j] = bfrag[(st * WAVE + lane) * 8 + j];
+        bu.d[0] = ((const unsigned int*)bfrag)[(st * WAVE + lane) * 4 + 0];
+        bu.d[1] = ((const unsigned int*)bfrag)[(st * WAVE + lane) * 4 + 1];
+        bu.d[2] = ((const unsigned int*)bfrag)[(st * WAVE + lane) * 4 + 2];
+        bu.d[3] = ((const unsigned int*)bfrag)[(st * WAVE + lane) * 4 + 3];
         bfr[st] = bu.v;
     }
 
@@ -277,14 +285,13 @@ __global__ __launch_bounds__(WG_THREADS) void conv_stack_mfma_kernel(
                 const int base = s * G::CIN + st * 32 + (lane >> 4) * 8;
                 BU au;
                 if constexpr (G::CIN % 2 == 0) {
-                    // base is even: 4-byte-aligned LDS reads
+                    // base is even: 4-byte-aligned LDS reads land directly
+                    // in the fragment dwords (little-endian bf16 pairs).
                     const unsigned int* xtu = (const unsigned int*)xt;
-                    unsigned int w0 = xtu[base / 2], w1v = xtu[base / 2 + 1];
-                    unsigned int w2v = xtu[base / 2 + 2], w3 = xtu[base / 2 + 3];
-                    au.u[0] = (unsigned short)w0;  au.u[1] = (unsigned short)(w0 >> 16);
-                    au.u[2] = (unsigned short)w1v; au.u[3] = (unsigned short)(w1v >> 16);
-                    au.u[4] = (unsigned short)w2v; au.u[5] = (unsigned short)(w2v >> 16);
-                    au.u[6] = (unsigned short)w3;  au.u[7] = (unsigned short)(w3 >> 16);
+                    au.d[0] = xtu[base / 2];
+                    au.d[1] = xtu[base / 2 + 1];
+                    au.d[2] = xtu[base / 2 + 2];
+                    au.d[3] = xtu[base / 2 + 3];
                 } else {
                     #pragma unroll
                     for (int j = 0; j < 8; ++j) au.u[j] = xt[base + j];
@@ -292,20 +299,21 @@ __global__ __launch_bounds__(WG_THREADS) void conv_stack_mfma_kernel(
                 acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(au.v, bfr[st],
                                                               acc, 0, 0, 0);
             }
-            // epilogue: bias + tanh; only the 4 real output channels land
+            // epilogue: bias-folded RAW pre-activations; tanh is applied
+            // after pool1 (monotonicity) in conv_tail<., true>
             const int c = lane & 15;
             if (c < 4) {
                 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     const int row = mt * 16 + (lane >> 4) * 4 + r;
                     if (row < G::C1)
-                        lds_c1[wave][c * G::C1 + row] = tanhf_(acc[r] + b1[c]);
+                        lds_c1[wave][c * G::C1 + row] = acc[r] + b1[c];
                 }
             }
         }
         wave_sync();
-        conv_tail<G>(lane, lds_w, lds_c1[wave], lds_p1[wave], lds_c2[wave],
-                     feat, win);
+        conv_tail<G, true>(lane, lds_w, lds_c1[wave], lds_p1[wave],
+                           lds_c2[wave], feat, win);
     }
 }
 
