@@ -138,3 +138,44 @@ class TestHipForward:
         ref = _oracle(m, x)
         got = eng.forward(x.cuda()).cpu()
         torch.testing.assert_close(got, ref, rtol=2e-4, atol=2e-5)
+
+
+@pytest.mark.gpu
+class TestGraphedForward:
+    def test_graph_matches_eager(self):
+        from tskd_amd.ops import GraphedForward
+        m = build_model("MyCNN5").eval()
+        eng = MyCNNEngine(m, device="cuda")
+        g = GraphedForward(eng, s=8, n=16, dtype=torch.bfloat16)
+        x = _x(8, 16, 10, seed=21, dtype=torch.bfloat16).cuda()
+        age = torch.full((8, 16), 65.0, device="cuda")
+        g.x.copy_(x)
+        g.age.copy_(age)
+        out = g.replay().clone()
+        torch.cuda.synchronize()
+        ref = eng.forward(x, age, apply_sigmoid=True)
+        torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-6)
+        # replay with NEW data must track the new input
+        x2 = _x(8, 16, 10, seed=22, dtype=torch.bfloat16).cuda()
+        g.x.copy_(x2)
+        out2 = g.replay().clone()
+        torch.cuda.synchronize()
+        ref2 = eng.forward(x2, age, apply_sigmoid=True)
+        torch.testing.assert_close(out2, ref2, rtol=1e-5, atol=1e-6)
+        assert not torch.allclose(out, out2)
+
+    def test_gather_into_graph_buffer(self):
+        from tskd_amd.engine import StreamEngine
+        from tskd_amd.ops import GraphedForward
+        m = build_model("MyCNN5").eval()
+        eng = MyCNNEngine(m, device="cuda")
+        se = StreamEngine(4, 10, ring_grid=1024, fs=125.0, device="cuda")
+        raw = torch.randn(4, 8, int(125 * 60 * 16), device="cuda")
+        se.ingest_dense(raw, chan_map=list(range(8)))
+        g = GraphedForward(eng, s=4, n=1, dtype=torch.bfloat16)
+        w = se.windows(batch=1, stride=12, dtype=torch.bfloat16, out=g.x)
+        assert w.data_ptr() == g.x.data_ptr()
+        out = g.replay()
+        torch.cuda.synchronize()
+        ref = eng.forward(g.x, g.age, apply_sigmoid=True)
+        torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-6)

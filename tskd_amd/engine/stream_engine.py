@@ -288,13 +288,19 @@ class StreamEngine:
 
     # ------------------------------------------------------------------ gather
     def windows(self, batch: int = 1, stride: int = 12,
-                dtype: torch.dtype = torch.float32) -> torch.Tensor:
+                dtype: torch.dtype = torch.float32,
+                out: Optional[torch.Tensor] = None) -> torch.Tensor:
         """Assemble (S, batch, C, model_win) model inputs ending at the latest
         processed grid point; window b ends at nproc - (batch-1-b)*stride.
-        Early windows that would reach before grid 0 are all-zero."""
+        Early windows that would reach before grid 0 are all-zero.
+        ``out``: optional pre-allocated destination (e.g. a GraphedForward's
+        static input buffer) — every element is overwritten."""
         B, WIN = batch, self.model_win
-        out = torch.zeros(self.S, B, self.C, WIN, dtype=dtype,
-                          device=self.device)
+        if out is not None:
+            assert out.shape == (self.S, B, self.C, WIN) and out.dtype == dtype
+            assert out.is_contiguous()
+        out = out if out is not None else torch.zeros(
+            self.S, B, self.C, WIN, dtype=dtype, device=self.device)
         if self._gpu:
             lib = _load_preproc_lib()
             is_bf16 = 1 if dtype == torch.bfloat16 else 0
