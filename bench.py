@@ -77,6 +77,9 @@ def main() -> None:
     p.add_argument("--batch", type=int, default=1024,
                    help="[infer] windows per sequence batch")
     p.add_argument("--variant", default="MyCNN5")
+    p.add_argument("--graph", action="store_true", default=True,
+                   help="hipGraph-capture the model forward (pipeline mode)")
+    p.add_argument("--no-graph", dest="graph", action="store_false")
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     args = p.parse_args()
 
@@ -145,11 +148,21 @@ def main() -> None:
         age = torch.full((S, 1), 65.0, device=device)
         gathered = [torch.empty(S, device=device) for _ in range(world)] \
             if dist else None
+        graphed = None
+        if args.graph and dtype == torch.bfloat16:
+            from tskd_amd.ops import GraphedForward
+            graphed = GraphedForward(eng, s=S, n=1, dtype=dtype)
 
         def step():
             se.ingest_dense(raw, chan_map=chan_map)
-            w = se.windows(batch=1, stride=12, dtype=dtype)
-            probs = eng.forward(w, age, apply_sigmoid=True)
+            if graphed is not None:
+                # gather writes the hipGraph's static input; replay the
+                # captured conv+LSTM+head+sigmoid sequence
+                se.windows(batch=1, stride=12, dtype=dtype, out=graphed.x)
+                probs = graphed.replay()
+            else:
+                w = se.windows(batch=1, stride=12, dtype=dtype)
+                probs = eng.forward(w, age, apply_sigmoid=True)
             if dist:
                 # predictions to every rank (RCCL all-gather over xGMI)
                 dist.all_gather(gathered, probs.reshape(S).contiguous())
@@ -157,7 +170,7 @@ def main() -> None:
         windows_per_step = S
         cfg = {"model": args.variant, "global_batch": S * world,
                "seq_len": 120, "streams_per_gpu": S,
-               "trigger_s": 60, "fs_hz": 125,
+               "trigger_s": 60, "fs_hz": 125, "hipgraph": graphed is not None,
                "parallelism": f"dp{world}", "mode": "pipeline"}
 
     elapsed_s, lat = run_steps(step, args.steps, args.warmup, dist, device)
