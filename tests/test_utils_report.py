@@ -75,3 +75,20 @@ class TestReport:
         # the synthetic signal is linearly separable enough for logreg
         base = sklearn_baselines(x[:300], y[:300], x[300:], y[300:])
         assert base["logreg"]["test_auc"] > 0.8
+
+
+class TestMetrics:
+    def test_stage_timer(self):
+        import time as _time
+
+        from tskd_amd.metrics import PipelineMetrics
+        pm = PipelineMetrics()
+        t = pm.stage("preprocess")
+        for _ in range(5):
+            with t:
+                _time.sleep(0.001)
+            t.add_items(12)
+        snap = t.snapshot()
+        assert snap["calls"] == 5 and snap["items"] == 60
+        assert snap["p50_ms"] >= 1.0
+        assert "preprocess" in pm.report()

@@ -172,13 +172,19 @@ def main(argv=None) -> None:
         if restored:
             log.info("resumed %d partition offsets from %s", restored,
                      args.offsets_file)
+    from tskd_amd.metrics import StageTimer
+    timer = StageTimer(ap.prog or "stage")
     n = 0
     while not stop:
         t0 = time.time()
-        ps.trigger()
+        with timer:
+            out = ps.trigger()
+        timer.add_items(out)
         if args.offsets_file:
             save_offsets(args.offsets_file, ps.consumer.positions())
         n += 1
+        if n % 10 == 0:
+            log.info("metrics %s", timer.log_line())
         if args.max_triggers and n >= args.max_triggers:
             break
         time.sleep(max(0.0, trigger_period - (time.time() - t0)))
