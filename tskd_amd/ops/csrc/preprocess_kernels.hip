@@ -49,6 +49,12 @@ __device__ __forceinline__ void wsync_() {
 // ---------------------------------------------------------------------------
 // 1a. Dense ingest: raw (S, CIN, T) -> NB = T/bucket_len new buckets/channel
 // ---------------------------------------------------------------------------
+// One wave processes FOUR consecutive buckets of one (stream, channel) row
+// in 16-lane groups — a 625-sample bucket is 78 oct-loads, so a whole-wave
+// bucket wastes 39% of lanes on the ragged tail; 16-lane groups keep ~95%
+// busy (78/16 = 4.9 balanced iterations per group).
+#define ING_GRP 4   // buckets per wave
+#define ING_GL 16   // lanes per bucket group
 template <class DT>
 __global__ void ingest_dense_kernel(
     const DT* __restrict__ raw,     // (S, CIN, T)
@@ -59,66 +65,75 @@ __global__ void ingest_dense_kernel(
     int bucket_len, long head)      // buckets written at [head, head+NB)
 {
     const int NB = T / bucket_len;
-    const long nwaves = (long)S * CIN * NB;
-    const int lane = threadIdx.x % WAVE;
+    const int NBG = (NB + ING_GRP - 1) / ING_GRP;
+    const long nwaves = (long)S * CIN * NBG;
+    const int wlane = threadIdx.x % WAVE;
+    const int grp = wlane / ING_GL;     // which of the wave's 4 buckets
+    const int lane = wlane % ING_GL;    // lane within the bucket group
     for (long w = (long)blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
          w < nwaves; w += (long)gridDim.x * (blockDim.x / WAVE)) {
-        const int b = (int)(w % NB);
-        const int cin = (int)((w / NB) % CIN);
-        const int s = (int)(w / ((long)NB * CIN));
-        const DT* src = raw + ((long)s * CIN + cin) * T + (long)b * bucket_len;
+        const int bg = (int)(w % NBG);
+        const int cin = (int)((w / NBG) % CIN);
+        const int s = (int)(w / ((long)NBG * CIN));
+        const int b = bg * ING_GRP + grp;
+        const bool active = b < NB;
+        const DT* src = raw + ((long)s * CIN + cin) * T +
+                        (long)(active ? b : 0) * bucket_len;
         float sum = 0.f, cnt = 0.f;
         // 16B-aligned vector body with scalar head/tail peel (bucket offsets
         // like 625 samples are not 16B-aligned).
-        if constexpr (sizeof(DT) == 2) {
-            int pre = (int)(((16 - ((size_t)src & 15)) & 15) / 2);
-            if (pre > bucket_len) pre = bucket_len;
-            for (int i = lane; i < pre; i += WAVE) {
-                const float f = bf16_to_f32_((unsigned short)src[i]);
-                if (!isnan(f)) { sum += f; cnt += 1.f; }
-            }
-            const int oct = (bucket_len - pre) / 8;
-            const uint4* vp = (const uint4*)((const unsigned short*)src + pre);
-            for (int p = lane; p < oct; p += WAVE) {
-                union { uint4 q; unsigned short h[8]; } v;
-                v.q = vp[p];
-                #pragma unroll
-                for (int j = 0; j < 8; ++j) {
-                    const float f = bf16_to_f32_(v.h[j]);
+        if (active) {
+            if constexpr (sizeof(DT) == 2) {
+                int pre = (int)(((16 - ((size_t)src & 15)) & 15) / 2);
+                if (pre > bucket_len) pre = bucket_len;
+                for (int i = lane; i < pre; i += ING_GL) {
+                    const float f = bf16_to_f32_((unsigned short)src[i]);
+                    if (!isnan(f)) { sum += f; cnt += 1.f; }
+                }
+                const int oct = (bucket_len - pre) / 8;
+                const uint4* vp = (const uint4*)((const unsigned short*)src + pre);
+                for (int p = lane; p < oct; p += ING_GL) {
+                    union { uint4 q; unsigned short h[8]; } v;
+                    v.q = vp[p];
+                    #pragma unroll
+                    for (int j = 0; j < 8; ++j) {
+                        const float f = bf16_to_f32_(v.h[j]);
+                        if (!isnan(f)) { sum += f; cnt += 1.f; }
+                    }
+                }
+                for (int i = pre + oct * 8 + lane; i < bucket_len; i += ING_GL) {
+                    const float f = bf16_to_f32_((unsigned short)src[i]);
+                    if (!isnan(f)) { sum += f; cnt += 1.f; }
+                }
+            } else {
+                int pre = (int)(((16 - ((size_t)src & 15)) & 15) / 4);
+                if (pre > bucket_len) pre = bucket_len;
+                for (int i = lane; i < pre; i += ING_GL) {
+                    const float f = (float)src[i];
+                    if (!isnan(f)) { sum += f; cnt += 1.f; }
+                }
+                const int quad = (bucket_len - pre) / 4;
+                const float4* vp = (const float4*)((const float*)src + pre);
+                for (int p = lane; p < quad; p += ING_GL) {
+                    const float4 v = vp[p];
+                    if (!isnan(v.x)) { sum += v.x; cnt += 1.f; }
+                    if (!isnan(v.y)) { sum += v.y; cnt += 1.f; }
+                    if (!isnan(v.z)) { sum += v.z; cnt += 1.f; }
+                    if (!isnan(v.w)) { sum += v.w; cnt += 1.f; }
+                }
+                for (int i = pre + quad * 4 + lane; i < bucket_len; i += ING_GL) {
+                    const float f = (float)src[i];
                     if (!isnan(f)) { sum += f; cnt += 1.f; }
                 }
             }
-            for (int i = pre + oct * 8 + lane; i < bucket_len; i += WAVE) {
-                const float f = bf16_to_f32_((unsigned short)src[i]);
-                if (!isnan(f)) { sum += f; cnt += 1.f; }
-            }
-        } else {
-            int pre = (int)(((16 - ((size_t)src & 15)) & 15) / 4);
-            if (pre > bucket_len) pre = bucket_len;
-            for (int i = lane; i < pre; i += WAVE) {
-                const float f = (float)src[i];
-                if (!isnan(f)) { sum += f; cnt += 1.f; }
-            }
-            const int quad = (bucket_len - pre) / 4;
-            const float4* vp = (const float4*)((const float*)src + pre);
-            for (int p = lane; p < quad; p += WAVE) {
-                const float4 v = vp[p];
-                if (!isnan(v.x)) { sum += v.x; cnt += 1.f; }
-                if (!isnan(v.y)) { sum += v.y; cnt += 1.f; }
-                if (!isnan(v.z)) { sum += v.z; cnt += 1.f; }
-                if (!isnan(v.w)) { sum += v.w; cnt += 1.f; }
-            }
-            for (int i = pre + quad * 4 + lane; i < bucket_len; i += WAVE) {
-                const float f = (float)src[i];
-                if (!isnan(f)) { sum += f; cnt += 1.f; }
-            }
         }
+        // reduce within each 16-lane bucket group
         #pragma unroll
-        for (int off = 32; off > 0; off >>= 1) {
+        for (int off = 8; off > 0; off >>= 1) {
             sum += __shfl_xor(sum, off);
             cnt += __shfl_xor(cnt, off);
         }
-        if (lane == 0) {
+        if (lane == 0 && active) {
             const int c = chan_map[cin];
             const long idx = ((long)s * C + c) * G + (head + b) % G;
             bsum[idx] = sum;
@@ -259,32 +274,48 @@ __global__ __launch_bounds__(256) void window_fill_kernel(
 //    Window b (b = 0..B-1) covers grid [end - (B-1-b)*stride - WIN,
 //    end - (B-1-b)*stride). Grid points never produced (g < 0) read as 0.
 // ---------------------------------------------------------------------------
+__device__ __forceinline__ unsigned short f32_to_bf16_(float f) {
+    union { float f; unsigned int i; } u;
+    u.f = f;
+    const unsigned int r = u.i + 0x7fffu + ((u.i >> 16) & 1);  // RNE
+    return (unsigned short)(r >> 16);
+}
+
+// Vectorized: each thread emits 4 consecutive time points (WIN % 4 == 0),
+// reading contiguous proc and writing one 8 B (bf16) / 16 B (f32) store.
 template <class OT>
 __global__ void window_gather_kernel(
     const float* __restrict__ proc,
     OT* __restrict__ out,            // (S, B, C, WIN)
     int S, int C, int G, int B, int WIN, int stride, long end)
 {
-    const long n = (long)S * B * C * WIN;
+    const int WQ = WIN / 4;
+    const long n = (long)S * B * C * WQ;
     for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
          i += (long)gridDim.x * blockDim.x) {
-        const int t = (int)(i % WIN);
-        const int c = (int)((i / WIN) % C);
-        const int b = (int)((i / ((long)WIN * C)) % B);
-        const int s = (int)(i / ((long)WIN * C * B));
+        const int tq = (int)(i % WQ);
+        const int c = (int)((i / WQ) % C);
+        const int b = (int)((i / ((long)WQ * C)) % B);
+        const int s = (int)(i / ((long)WQ * C * B));
         const long wend = end - (long)(B - 1 - b) * stride;
-        const long g = wend - WIN + t;
-        float v = 0.f;
-        if (g >= 0 && g < end && wend - WIN >= 0)
-            v = proc[((long)s * C + c) * G + g % G];
+        const long g0 = wend - WIN + tq * 4;
+        const float* pr = proc + ((long)s * C + c) * G;
+        float v[4];
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+            const long g = g0 + j;
+            v[j] = (g >= 0 && g < end && wend - WIN >= 0 && wend <= end)
+                       ? pr[g % G] : 0.f;
+        }
+        const long o = (((long)s * B + b) * C + c) * WIN + (long)tq * 4;
         if constexpr (sizeof(OT) == 2) {
-            union { float f; unsigned int i; } u;
-            u.f = v;
-            // round-to-nearest-even bf16
-            unsigned int r = u.i + 0x7fffu + ((u.i >> 16) & 1);
-            out[i] = (OT)(r >> 16);
+            union { unsigned short h[4]; unsigned long long u; } pk;
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) pk.h[j] = f32_to_bf16_(v[j]);
+            *(unsigned long long*)(out + o) = pk.u;
         } else {
-            out[i] = (OT)v;
+            float4 pk = {v[0], v[1], v[2], v[3]};
+            *(float4*)((float*)out + o) = pk;
         }
     }
 }
@@ -308,7 +339,7 @@ int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
     hipStream_t st = (hipStream_t)stream;
     const int NB = T / bucket_len;
     if (NB <= 0 || S <= 0) return 0;
-    const long nwaves = (long)S * CIN * NB;
+    const long nwaves = (long)S * CIN * ((NB + ING_GRP - 1) / ING_GRP);
     const int grid = grid_for(nwaves * WAVE, 256);
     if (raw_is_bf16)
         hipLaunchKernelGGL((ingest_dense_kernel<unsigned short>), dim3(grid),
@@ -356,7 +387,8 @@ int tskd_preproc_window_fill(const float* bsum, const float* bcnt, float* proc,
 int tskd_preproc_window_gather(const float* proc, void* out, int out_is_bf16,
                                int S, int C, int G, int B, int WIN, int stride,
                                long end, void* stream) {
-    const long n = (long)S * B * C * WIN;
+    if (WIN % 4 != 0) return -3;  // vectorized gather needs WIN % 4 == 0
+    const long n = (long)S * B * C * (WIN / 4);
     if (n <= 0) return 0;
     hipStream_t st = (hipStream_t)stream;
     if (out_is_bf16)
