@@ -60,7 +60,8 @@ class TestHipTraining:
         y = (torch.rand(S, B) < 0.3).float()
         ref_loss, ref_g = _torch_loss_and_grads(model, x, age, y, pos_weight)
 
-        tr = MyCNNHipTrainer(model, device="cuda", pos_weight=pos_weight)
+        tr = MyCNNHipTrainer(model, device="cuda", pos_weight=pos_weight,
+                             train_dropout=False)
         loss = tr.forward_backward(x.cuda(), age.cuda(), y.cuda())
         torch.cuda.synchronize()
         assert abs(loss - ref_loss) / max(abs(ref_loss), 1e-6) < 1e-3
@@ -78,7 +79,7 @@ class TestHipTraining:
         from tskd_amd.train.hip_trainer import MyCNNHipTrainer
         torch.manual_seed(8)
         model = build_model("MyCNN5").eval()
-        tr = MyCNNHipTrainer(model, device="cuda")
+        tr = MyCNNHipTrainer(model, device="cuda", train_dropout=False)
         x = torch.randn(2, 16, 10, 120).cuda()
         age = torch.full((2, 16), 65.0).cuda()
         y = (torch.rand(2, 16) < 0.5).float().cuda()
@@ -137,7 +138,8 @@ class TestHipTraining:
         from tskd_amd.train.hip_trainer import MyCNNHipTrainer
         torch.manual_seed(11)
         model = build_model("MyCNN5").eval()
-        tr = MyCNNHipTrainer(model, device="cuda", lr=1e-2)
+        tr = MyCNNHipTrainer(model, device="cuda", lr=1e-2,
+                             train_dropout=False)
         x = torch.randn(2, 32, 10, 120).cuda()
         age = torch.full((2, 32), 65.0).cuda()
         y = (torch.rand(2, 32) < 0.5).float().cuda()
@@ -147,3 +149,76 @@ class TestHipTraining:
         wpack2 = pack_weights(m2)
         torch.testing.assert_close(wpack2, tr.wpack.cpu(), rtol=1e-6,
                                    atol=1e-7)
+
+
+@pytest.mark.gpu
+class TestDropout:
+    def test_mask_exact_grad_parity(self):
+        """Train-mode dropout (K5): extract the kernel's Bernoulli masks from
+        the stash and replay them through a torch functional forward — loss
+        and gradients must match exactly."""
+        import torch.nn.functional as F
+
+        from tskd_amd.train.hip_trainer import MyCNNHipTrainer
+        torch.manual_seed(12)
+        model = build_model("MyCNN5").eval()
+        S, B = 2, 16
+        x = torch.randn(S, B, 10, 120)
+        age = torch.full((S, B), 65.0)
+        y = (torch.rand(S, B) < 0.4).float()
+        tr = MyCNNHipTrainer(model, device="cuda", pos_weight=1.0,
+                             train_dropout=True, seed=99)
+        loss = tr.forward_backward(x.cuda(), age.cuda(), y.cuda())
+        torch.cuda.synchronize()
+        stash = tr._last_stash_conv.cpu()  # (S*B, SC_SIZE)
+        # stash offsets for MyCNN5: C1=111, C2=51, P1=55, LIN=25
+        off_m1 = 4 * 111 + 51 + 4 * 55 + 25
+        off_m2 = off_m1 + 4 * 55
+        m1 = stash[:, off_m1:off_m1 + 4 * 55].reshape(S, B, 4, 55)
+        m2 = stash[:, off_m2:off_m2 + 25].reshape(S, B, 25)
+        drop_rate = float((m1 == 0).float().mean())
+        assert 0.05 < drop_rate < 0.16  # p=0.1 Bernoulli
+
+        # torch oracle with the SAME masks
+        model.zero_grad()
+        logits = []
+        for s_ in range(S):
+            h = torch.tanh(model.conv1(x[s_]))
+            h = model.pool(h) * m1[s_]
+            h = torch.tanh(model.conv2(h))
+            h = model.pool(h).squeeze(1) * m2[s_]
+            out, _ = model.lstm(h)
+            z = model.out(out)
+            scale = torch.relu(age[s_].unsqueeze(1) * 1e-8 + 1)
+            logits.append((z * scale).squeeze(1))
+        zz = torch.cat(logits)
+        ref_loss = F.binary_cross_entropy_with_logits(zz, y.reshape(-1))
+        ref_loss.backward()
+        assert abs(loss - float(ref_loss)) / max(float(ref_loss), 1e-6) < 1e-3
+        got = tr.grads.cpu()
+        o = tr.offsets
+        g = {n: p_.grad for n, p_ in model.named_parameters()
+             if p_.grad is not None}
+        for name, t in [("w1", g["conv1.weight"]), ("b1", g["conv1.bias"]),
+                        ("w2", g["conv2.weight"]),
+                        ("wih1", g["lstm.weight_ih_l0"]),
+                        ("outw", g["out.weight"])]:
+            a, b = o[name]
+            torch.testing.assert_close(got[a:b], t.reshape(-1), rtol=2e-3,
+                                       atol=2e-4,
+                                       msg=lambda m, n=name: f"{n}: {m}")
+
+    def test_masks_change_per_step(self):
+        from tskd_amd.train.hip_trainer import MyCNNHipTrainer
+        model = build_model("MyCNN5").eval()
+        tr = MyCNNHipTrainer(model, device="cuda", train_dropout=True)
+        x = torch.randn(1, 8, 10, 120).cuda()
+        age = torch.full((1, 8), 65.0).cuda()
+        y = torch.zeros(1, 8).cuda()
+        tr.forward_backward(x, age, y)
+        torch.cuda.synchronize()
+        m_a = tr._last_stash_conv[:, 740:].cpu().clone()
+        tr.forward_backward(x, age, y)
+        torch.cuda.synchronize()
+        m_b = tr._last_stash_conv[:, 740:].cpu()
+        assert not torch.equal(m_a, m_b)

@@ -60,7 +60,9 @@ struct TGeom {
     static constexpr int SC_C2T = SC_C1T + 4 * C1;   // [C2]  tanh(conv2)
     static constexpr int SC_I1 = SC_C2T + C2;        // [4*P1] pool1 argmax
     static constexpr int SC_I2 = SC_I1 + 4 * P1;     // [LIN] pool2 argmax
-    static constexpr int SC_SIZE = SC_I2 + LIN;
+    static constexpr int SC_M1 = SC_I2 + LIN;        // [4*P1] dropout1 mult
+    static constexpr int SC_M2 = SC_M1 + 4 * P1;     // [LIN] dropout2 mult
+    static constexpr int SC_SIZE = SC_M2 + LIN;
     // lstm stash layout (per step, fp32 words)
     static constexpr int SL_A1 = 0;    // [64] layer1 activated gates
     static constexpr int SL_C1 = 64;   // [16]
@@ -87,6 +89,25 @@ __device__ __forceinline__ float tanh_(float x) {
     return 2.0f / (1.0f + __expf(-2.0f * x)) - 1.0f;
 }
 
+// Counter-based RNG for dropout masks (K5): stateless splitmix64 hash of
+// (seed, element index) -> uniform [0,1). Not ATen's Philox stream — the
+// reference's train-mode Bernoulli semantics, reproducible per seed.
+__device__ __forceinline__ float rnd_uniform_(unsigned long long seed,
+                                              unsigned long long idx) {
+    unsigned long long z = seed + idx * 0x9E3779B97F4A7C15ull;
+    z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+    z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+    z ^= z >> 31;
+    return (float)(unsigned int)(z >> 33) * (1.0f / 2147483648.0f);
+}
+
+// inverted-dropout multiplier: 0 with prob p, else 1/(1-p)
+__device__ __forceinline__ float drop_mult_(float p, unsigned long long seed,
+                                            unsigned long long idx) {
+    if (p <= 0.f) return 1.f;
+    return (rnd_uniform_(seed, idx) < p) ? 0.f : 1.f / (1.f - p);
+}
+
 // ---------------------------------------------------------------------------
 // conv forward with stash (fp32; one wave per window, 4 waves per block)
 // ---------------------------------------------------------------------------
@@ -95,7 +116,8 @@ __global__ __launch_bounds__(256) void train_conv_fwd_kernel(
     const float* __restrict__ x,      // (SN, CIN, L)
     float* __restrict__ feat,         // (SN, LIN)
     float* __restrict__ stash,        // (SN, SC_SIZE) (int words for argmax)
-    const float* __restrict__ wpack, int SN)
+    const float* __restrict__ wpack, int SN,
+    float drop1_p, float drop2_p, unsigned long long seed)
 {
     constexpr int NW = 4 * G::CIN * G::K1 + 4 + 20 + 1;
     __shared__ float lw[NW];
@@ -117,6 +139,8 @@ __global__ __launch_bounds__(256) void train_conv_fwd_kernel(
         float* c2t = st + G::SC_C2T;
         int* i1 = (int*)(st + G::SC_I1);
         int* i2 = (int*)(st + G::SC_I2);
+        float* m1 = st + G::SC_M1;
+        float* m2 = st + G::SC_M2;
         float* xw = lx[wave];
         for (int i = lane; i < G::CIN * G::L; i += WAVE) xw[i] = xin[i];
         twsync();
@@ -140,7 +164,11 @@ __global__ __launch_bounds__(256) void train_conv_fwd_kernel(
             #pragma unroll
             for (int k = 1; k < G::PK; ++k)
                 if (src[k] > m) { m = src[k]; am = k; }
-            lp1[wave][o] = m;
+            // dropout site 1 (after pool1; MyCNN5's first dropout)
+            const float dm = drop_mult_(drop1_p, seed,
+                                        (unsigned long long)win * 1024 + o);
+            m1[o] = dm;
+            lp1[wave][o] = m * dm;
             i1[o] = am;
         }
         twsync();
@@ -162,7 +190,11 @@ __global__ __launch_bounds__(256) void train_conv_fwd_kernel(
             #pragma unroll
             for (int k = 1; k < G::PK; ++k)
                 if (src[k] > m) { m = src[k]; am = k; }
-            feat[win * G::LIN + q] = m;
+            // dropout site 2 (after pool2)
+            const float dm = drop_mult_(drop2_p, seed,
+                                        (unsigned long long)win * 1024 + 512 + q);
+            m2[q] = dm;
+            feat[win * G::LIN + q] = m * dm;
             i2[q] = am;
         }
         twsync();
@@ -495,14 +527,16 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
         const float* c2t = st + G::SC_C2T;
         const int* i1 = (const int*)(st + G::SC_I1);
         const int* i2 = (const int*)(st + G::SC_I2);
+        const float* m1s = st + G::SC_M1;
+        const float* m2s = st + G::SC_M2;
         const float* dfw = dfeat + win * G::LIN;
         const float* xin = x + win * (G::CIN * G::L);
         float* xw = lx[wave];
         for (int i = lane; i < G::CIN * G::L; i += WAVE) xw[i] = xin[i];
-        // recompute pool1 output (cheap; avoids stashing it)
+        // recompute the (dropout-masked) pool1 output = conv2's input
         for (int o = lane; o < 4 * G::P1; o += WAVE) {
             const int c = o / G::P1, q = o % G::P1;
-            lp1[wave][o] = c1t[c * G::C1 + q * G::PS + i1[o]];
+            lp1[wave][o] = c1t[c * G::C1 + q * G::PS + i1[o]] * m1s[o];
             ldp1[wave][o] = 0.f;
         }
         for (int sidx = lane; sidx < G::C2; sidx += WAVE) lda2[wave][sidx] = 0.f;
@@ -511,7 +545,7 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
         if (lane == 0) {
             for (int q = 0; q < G::LIN; ++q) {
                 const int p = q * G::PS + i2[q];
-                const float d = dfw[q];
+                const float d = dfw[q] * m2s[q];       // through dropout2
                 const float ct = c2t[p];
                 lda2[wave][p] += d * (1.f - ct * ct);  // through tanh'
             }
@@ -551,6 +585,7 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
                 const int p = c * G::C1 + q * G::PS + i1[c * G::P1 + q];
                 const float ct = c1t[p];
                 lda1[wave][p] += ldp1[wave][c * G::P1 + q] *
+                                 m1s[c * G::P1 + q] *   // through dropout1
                                  (1.f - ct * ct);
             }
         }
@@ -618,11 +653,12 @@ __global__ void train_adam_kernel(
 namespace {
 template <class G>
 int conv_fwd(const float* x, float* feat, float* stash, const float* wpack,
-             int SN, hipStream_t s) {
+             int SN, float d1, float d2, unsigned long long seed,
+             hipStream_t s) {
     if (SN <= 0) return 0;
     int grid = min((SN + 3) / 4, 8192);
     hipLaunchKernelGGL((train_conv_fwd_kernel<G>), dim3(grid), dim3(256), 0,
-                       s, x, feat, stash, wpack, SN);
+                       s, x, feat, stash, wpack, SN, d1, d2, seed);
     return (int)hipGetLastError();
 }
 template <class G>
@@ -659,11 +695,16 @@ int conv_bwd(const float* x, const float* stash, const float* dfeat,
 extern "C" {
 
 int tskd_train_conv_fwd(const float* x, float* feat, float* stash,
-                        const float* wpack, int SN, int variant, void* s) {
+                        const float* wpack, int SN, float drop1_p,
+                        float drop2_p, unsigned long long seed, int variant,
+                        void* s) {
     switch (variant) {
-        case 0: return conv_fwd<TG5>(x, feat, stash, wpack, SN, (hipStream_t)s);
-        case 1: return conv_fwd<TG2>(x, feat, stash, wpack, SN, (hipStream_t)s);
-        case 2: return conv_fwd<TG4>(x, feat, stash, wpack, SN, (hipStream_t)s);
+        case 0: return conv_fwd<TG5>(x, feat, stash, wpack, SN, drop1_p,
+                                     drop2_p, seed, (hipStream_t)s);
+        case 1: return conv_fwd<TG2>(x, feat, stash, wpack, SN, drop1_p,
+                                     drop2_p, seed, (hipStream_t)s);
+        case 2: return conv_fwd<TG4>(x, feat, stash, wpack, SN, drop1_p,
+                                     drop2_p, seed, (hipStream_t)s);
     }
     return -1;
 }

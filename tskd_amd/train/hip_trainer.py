@@ -37,7 +37,8 @@ def _load_train_lib() -> ctypes.CDLL:
     lib = ctypes.CDLL(path)
     P, I, L, F = ctypes.c_void_p, ctypes.c_int, ctypes.c_long, ctypes.c_float
     lib.tskd_train_conv_fwd.restype = I
-    lib.tskd_train_conv_fwd.argtypes = [P, P, P, P, I, I, P]
+    lib.tskd_train_conv_fwd.argtypes = [P, P, P, P, I, F, F,
+                                        ctypes.c_ulonglong, I, P]
     lib.tskd_train_lstm_fwd.restype = I
     lib.tskd_train_lstm_fwd.argtypes = [P, P, P, P, P, I, I, F, I, P]
     lib.tskd_train_loss.restype = I
@@ -65,8 +66,16 @@ def _p(t: torch.Tensor) -> ctypes.c_void_p:
 class MyCNNHipTrainer:
     def __init__(self, model, device: str = "cuda", lr: float = 1e-5,
                  betas: Tuple[float, float] = (0.9, 0.999), eps: float = 1e-8,
-                 pos_weight: float = 1.0):
+                 pos_weight: float = 1.0, train_dropout: bool = True,
+                 seed: int = 0):
         self.model = model
+        # reference train-mode dropout sites: MyCNN5 drops
+        # after BOTH pools; MyCNN2/3/4 only after pool2
+        p_drop = float(model.DROPOUT_P) if train_dropout else 0.0
+        self.drop1_p = p_drop if getattr(model, "TWO_DROPOUT_SITES", True) \
+            else 0.0
+        self.drop2_p = p_drop
+        self.seed = int(seed)
         self.variant = VARIANT_IDS[type(model).__name__]
         self.age_eps = float(model.AGE_EPS)
         self.cin = int(model.IN_CHANNELS)
@@ -106,6 +115,7 @@ class MyCNNHipTrainer:
             age = age.to(dev, torch.float32).expand(S, B).contiguous()
         feat = torch.empty(SN, self.lin, device=dev)
         stash_c = torch.empty(SN, self._stash_conv_words, device=dev)
+        self._last_stash_conv = stash_c  # exposed for mask-exact testing
         stash_l = torch.empty(S, B, self._stash_lstm_words, device=dev)
         logits = torch.empty(S, B, device=dev)
         dlogit = torch.empty(S, B, device=dev)
@@ -113,8 +123,11 @@ class MyCNNHipTrainer:
         loss = torch.zeros(1, device=dev)
         st = _sp()
         age_p = _p(age) if age is not None else ctypes.c_void_p(0)
-        rc = lib.tskd_train_conv_fwd(_p(x), _p(feat), _p(stash_c),
-                                     _p(self.wpack), SN, self.variant, st)
+        self.seed += 1  # fresh dropout masks every call
+        rc = lib.tskd_train_conv_fwd(
+            _p(x), _p(feat), _p(stash_c), _p(self.wpack), SN,
+            ctypes.c_float(self.drop1_p), ctypes.c_float(self.drop2_p),
+            ctypes.c_ulonglong(self.seed), self.variant, st)
         assert rc == 0, f"conv_fwd {rc}"
         rc = lib.tskd_train_lstm_fwd(_p(feat), age_p, _p(self.wpack),
                                      _p(stash_l), _p(logits), S, B,
