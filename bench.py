@@ -123,9 +123,17 @@ def main() -> None:
                "optimizer": "fused_adam", "loss": "bce_pos_weight"}
         args.dtype = "fp32"  # training runs fp32 like the reference
     elif args.mode == "infer":
+        from tskd_amd.ops import alloc_windows
         S, B = args.seqs, args.batch
-        x = torch.randn(S, B, 10, 120, device=device, dtype=dtype)
-        x[:, :, 8:, :] = 0  # 8 active of 10 wire channels
+        if dtype == torch.bfloat16:
+            # timelast layout: the serving engine's internal window format
+            x = alloc_windows(S, B, 10, timelast=True, dtype=dtype,
+                              device=device)
+            x.copy_(torch.randn(S, B, 120, 10, device=device, dtype=dtype))
+            x[:, :, :, 8:] = 0  # 8 active of 10 wire channels
+        else:
+            x = torch.randn(S, B, 10, 120, device=device, dtype=dtype)
+            x[:, :, 8:, :] = 0
         age = torch.full((S, B), 65.0, device=device)
 
         def step():
@@ -151,14 +159,16 @@ def main() -> None:
         graphed = None
         if args.graph and dtype == torch.bfloat16:
             from tskd_amd.ops import GraphedForward
-            graphed = GraphedForward(eng, s=S, n=1, dtype=dtype)
+            graphed = GraphedForward(eng, s=S, n=1, dtype=dtype,
+                                     timelast=True)
 
         def step():
             se.ingest_dense(raw, chan_map=chan_map)
             if graphed is not None:
-                # gather writes the hipGraph's static input; replay the
-                # captured conv+LSTM+head+sigmoid sequence
-                se.windows(batch=1, stride=12, dtype=dtype, out=graphed.x)
+                # gather writes the hipGraph's static input (timelast layout
+                # feeds the LDS-free MFMA conv); replay the captured sequence
+                se.windows(batch=1, stride=12, dtype=dtype, out=graphed.x,
+                           timelast=True)
                 probs = graphed.replay()
             else:
                 w = se.windows(batch=1, stride=12, dtype=dtype)

@@ -179,3 +179,56 @@ class TestGraphedForward:
         torch.cuda.synchronize()
         ref = eng.forward(g.x, g.age, apply_sigmoid=True)
         torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-6)
+
+
+@pytest.mark.gpu
+class TestTimelastLayout:
+    def test_tlast_matches_std(self):
+        from tskd_amd.ops import alloc_windows
+        m = build_model("MyCNN5").eval()
+        eng = MyCNNEngine(m, device="cuda")
+        x = _x(2, 32, 10, seed=31, dtype=torch.bfloat16).cuda()
+        ref = eng.forward(x)  # staged-layout path
+        xt = alloc_windows(2, 32, 10, timelast=True, dtype=torch.bfloat16)
+        xt.copy_(x.transpose(-1, -2))
+        got = eng.forward(xt)  # LDS-free timelast path
+        torch.testing.assert_close(got, ref, rtol=1e-4, atol=1e-5)
+
+    def test_tlast_without_slack_is_copied_safely(self):
+        m = build_model("MyCNN5").eval()
+        eng = MyCNNEngine(m, device="cuda")
+        x = _x(1, 8, 10, seed=32, dtype=torch.bfloat16).cuda()
+        xt = x.transpose(-1, -2).contiguous()  # no slack tag -> wrapper copies
+        ref = eng.forward(x)
+        got = eng.forward(xt)
+        torch.testing.assert_close(got, ref, rtol=1e-4, atol=1e-5)
+
+    def test_gather_timelast(self):
+        from tskd_amd.engine import StreamEngine
+        se_c = StreamEngine(3, 10, ring_grid=1024, fs=25.0, device="cpu")
+        se_g = StreamEngine(3, 10, ring_grid=1024, fs=25.0, device="cuda")
+        raw = torch.randn(3, 8, int(25 * 60 * 16))
+        se_c.ingest_dense(raw, chan_map=list(range(8)))
+        se_g.ingest_dense(raw.cuda(), chan_map=list(range(8)))
+        wc = se_c.windows(batch=2, stride=12, timelast=True)
+        wg = se_g.windows(batch=2, stride=12, timelast=True)
+        torch.cuda.synchronize()
+        assert wc.shape == (3, 2, 120, 10)
+        torch.testing.assert_close(wg.cpu(), wc, rtol=1e-4, atol=1e-5)
+
+    def test_graphed_timelast_pipeline(self):
+        from tskd_amd.engine import StreamEngine
+        from tskd_amd.ops import GraphedForward
+        m = build_model("MyCNN5").eval()
+        eng = MyCNNEngine(m, device="cuda")
+        se = StreamEngine(4, 10, ring_grid=1024, fs=125.0, device="cuda")
+        se.ingest_dense(torch.randn(4, 8, int(125 * 60 * 16), device="cuda"),
+                        chan_map=list(range(8)))
+        g = GraphedForward(eng, s=4, n=1, dtype=torch.bfloat16, timelast=True)
+        se.windows(batch=1, stride=12, dtype=torch.bfloat16, out=g.x,
+                   timelast=True)
+        out = g.replay().clone()
+        torch.cuda.synchronize()
+        w_std = se.windows(batch=1, stride=12, dtype=torch.bfloat16)
+        ref = eng.forward(w_std, g.age, apply_sigmoid=True)
+        torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-6)

@@ -68,7 +68,7 @@ def _load_preproc_lib() -> ctypes.CDLL:
     lib.tskd_preproc_window_gather.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
         ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
-        ctypes.c_long, ctypes.c_void_p,
+        ctypes.c_int, ctypes.c_long, ctypes.c_void_p,
     ]
     _plib = lib
     return lib
@@ -289,25 +289,28 @@ class StreamEngine:
     # ------------------------------------------------------------------ gather
     def windows(self, batch: int = 1, stride: int = 12,
                 dtype: torch.dtype = torch.float32,
-                out: Optional[torch.Tensor] = None) -> torch.Tensor:
+                out: Optional[torch.Tensor] = None,
+                timelast: bool = False) -> torch.Tensor:
         """Assemble (S, batch, C, model_win) model inputs ending at the latest
         processed grid point; window b ends at nproc - (batch-1-b)*stride.
         Early windows that would reach before grid 0 are all-zero.
         ``out``: optional pre-allocated destination (e.g. a GraphedForward's
         static input buffer) — every element is overwritten."""
         B, WIN = batch, self.model_win
+        shape = (self.S, B, WIN, self.C) if timelast \
+            else (self.S, B, self.C, WIN)
         if out is not None:
-            assert out.shape == (self.S, B, self.C, WIN) and out.dtype == dtype
+            assert out.shape == shape and out.dtype == dtype
             assert out.is_contiguous()
         out = out if out is not None else torch.zeros(
-            self.S, B, self.C, WIN, dtype=dtype, device=self.device)
+            shape, dtype=dtype, device=self.device)
         if self._gpu:
             lib = _load_preproc_lib()
             is_bf16 = 1 if dtype == torch.bfloat16 else 0
             assert dtype in (torch.bfloat16, torch.float32)
             rc = lib.tskd_preproc_window_gather(
                 ctypes.c_void_p(self.proc.data_ptr()),
-                ctypes.c_void_p(out.data_ptr()), is_bf16,
+                ctypes.c_void_p(out.data_ptr()), is_bf16, int(timelast),
                 self.S, self.C, self.G, B, WIN, stride,
                 ctypes.c_long(self.nproc), _sptr())
             if rc != 0:
@@ -319,7 +322,8 @@ class StreamEngine:
                 if wend - WIN < 0:
                     continue
                 idx = (np.arange(wend - WIN, wend)) % self.G
-                out[:, b] = torch.from_numpy(pr[:, :, idx]).to(dtype)
+                w = torch.from_numpy(pr[:, :, idx]).to(dtype)
+                out[:, b] = w.transpose(-1, -2) if timelast else w
         return out
 
     @property

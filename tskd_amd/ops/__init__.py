@@ -36,8 +36,9 @@ def _load_lib() -> ctypes.CDLL:
     lib = ctypes.CDLL(path)
     lib.tskd_conv_fwd.restype = ctypes.c_int
     lib.tskd_conv_fwd.argtypes = [
-        ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
+        ctypes.c_void_p,
     ]
     lib.tskd_debug_mfma16x16x32.restype = ctypes.c_int
     lib.tskd_debug_mfma16x16x32.argtypes = [
@@ -67,6 +68,24 @@ def hip_available() -> bool:
 
 def _stream_ptr() -> ctypes.c_void_p:
     return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+TLAST_SLACK = 256  # trailing elements the timelast conv kernel may overread
+
+
+def alloc_windows(s: int, n: int, cin: int, win: int = 120,
+                  timelast: bool = False, dtype=torch.bfloat16,
+                  device="cuda") -> torch.Tensor:
+    """Allocate a window tensor with the trailing slack the timelast
+    (LDS-free) conv kernel requires; returns a (s, n, cin, win) or
+    (s, n, win, cin) view tagged slack-safe."""
+    numel = s * n * cin * win
+    buf = torch.zeros(numel + TLAST_SLACK, dtype=dtype, device=device)
+    shape = (s, n, win, cin) if timelast else (s, n, cin, win)
+    t = buf[:numel].view(*shape)
+    t._tskd_slack = True  # type: ignore[attr-defined]
+    t._tskd_buf = buf  # keep storage alive  # type: ignore[attr-defined]
+    return t
 
 
 class MyCNNEngine:
@@ -100,13 +119,25 @@ class MyCNNEngine:
             self.feat_len = int(model.MAGICNUM)
 
     def conv_features(self, x: torch.Tensor) -> torch.Tensor:
-        """(..., C, 120) -> (..., LIN) conv-stack features."""
+        """(..., C, 120) or timelast (..., 120, C) -> (..., LIN) features.
+
+        The timelast layout runs the LDS-free MFMA kernel (even channel
+        counts, bf16); its fragment reads may overrun the tensor by up to
+        TLAST_SLACK elements, so non-slack-tagged inputs are copied into a
+        padded buffer (allocate with :func:`alloc_windows` to avoid that).
+        """
+        timelast = (x.shape[-1] == self.cin and x.shape[-2] == 120
+                    and self.cin != 120)
         lead = x.shape[:-2]
         xf = x.reshape(-1, x.shape[-2], x.shape[-1]).contiguous()
         sn = xf.shape[0]
         if xf.device.type != "cuda":
             raise RuntimeError("conv_features is the GPU path; use the model on CPU")
         lib = _load_lib()
+        if timelast and (self.cin % 2 != 0 or xf.dtype != torch.bfloat16):
+            # odd-CIN / fp32: transpose back to the staged layout
+            xf = xf.transpose(-1, -2).contiguous()
+            timelast = False
         if xf.dtype == torch.bfloat16:
             is_bf16 = 1
         elif xf.dtype == torch.float32:
@@ -114,9 +145,14 @@ class MyCNNEngine:
         else:
             xf = xf.to(torch.float32)
             is_bf16 = 0
+        if timelast and not getattr(x, "_tskd_slack", False):
+            pad = alloc_windows(1, sn, self.cin, 120, timelast=True,
+                                dtype=xf.dtype, device=xf.device)
+            pad.reshape(-1).copy_(xf.reshape(-1))
+            xf = pad.reshape(sn, 120, self.cin)
         feat = torch.empty(sn, self.feat_len, dtype=torch.float32, device=xf.device)
         rc = lib.tskd_conv_fwd(
-            ctypes.c_void_p(xf.data_ptr()), is_bf16,
+            ctypes.c_void_p(xf.data_ptr()), is_bf16, int(timelast),
             ctypes.c_void_p(feat.data_ptr()),
             ctypes.c_void_p(self.wpack.data_ptr()),
             ctypes.c_void_p(self.bfrag.data_ptr()), sn, self.variant,
@@ -160,7 +196,9 @@ class MyCNNEngine:
             x = x.unsqueeze(0)
             if age is not None and age.dim() == 1:
                 age = age.unsqueeze(0)
-        assert x.dim() == 4 and x.shape[-1] == 120 and x.shape[-2] == self.cin
+        assert x.dim() == 4 and (
+            (x.shape[-1] == 120 and x.shape[-2] == self.cin)
+            or (x.shape[-2] == 120 and x.shape[-1] == self.cin))
         if x.device.type == "cuda":
             s, n = x.shape[0], x.shape[1]
             feat = self.conv_features(x).reshape(s, n, self.feat_len)
@@ -191,11 +229,13 @@ class GraphedForward:
 
     def __init__(self, engine: "MyCNNEngine", s: int, n: int,
                  dtype: torch.dtype = torch.bfloat16,
-                 apply_sigmoid: bool = True, warmup: int = 2):
+                 apply_sigmoid: bool = True, warmup: int = 2,
+                 timelast: bool = False):
         assert engine.device.type == "cuda"
         self.engine = engine
-        self.x = torch.zeros(s, n, engine.cin, 120, dtype=dtype,
-                             device=engine.device)
+        self.timelast = timelast
+        self.x = alloc_windows(s, n, engine.cin, 120, timelast=timelast,
+                               dtype=dtype, device=engine.device)
         self.age = torch.full((s, n), 65.0, device=engine.device)
         stream = torch.cuda.Stream()
         with torch.cuda.stream(stream):

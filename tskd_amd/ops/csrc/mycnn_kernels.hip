@@ -318,6 +318,92 @@ __global__ __launch_bounds__(WG_THREADS) void conv_stack_mfma_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Kernel 1c (bf16 TIME-LAST layout): conv1 MFMA im2col with A-fragments read
+// STRAIGHT FROM GLOBAL — no LDS staging, no transpose pass.
+//
+// Input layout (SN, L, CIN) ("timelast"): the im2col row A[s][kk]
+// (kk = k*CIN + i) is the contiguous global range x[win*L*CIN + s*CIN + kk],
+// so each lane's 8-element fragment is 4 dword loads through the vector-
+// memory path (L1-resident: a window is 2.4 KB). PMC showed the staged
+// variant spends 39% of cycles waiting on LDS (profiles/r01) — this variant
+// keeps LDS only for the small pooled tail.
+//
+// NOTE: fragment reads overrun a window by up to (KSTEPS*32 - KK + C1%16
+// rows) elements into the NEXT window (zero B-columns make the values
+// irrelevant); callers must allocate the x buffer with TLAST_SLACK trailing
+// elements (tskd_amd/ops wrapper does).
+// ---------------------------------------------------------------------------
+#define TLAST_SLACK 256  // trailing bf16 elements required after the tensor
+template <class G>
+__global__ __launch_bounds__(WG_THREADS) void conv_stack_mfma_tlast_kernel(
+    const unsigned short* __restrict__ x,   // (SN, L, CIN) bf16 bits
+    float* __restrict__ feat,               // (SN, LIN)
+    const float* __restrict__ wpack,
+    const unsigned short* __restrict__ bfrag,  // [KSTEPS][64][8] bf16 bits
+    int SN)
+{
+    constexpr int NW = 4 * G::CIN * G::K1 + 4 + 20 + 1;
+    __shared__ float lds_w[NW];
+    __shared__ float lds_c1[WG_WAVES][4 * G::C1];
+    __shared__ float lds_p1[WG_WAVES][4 * G::P1];
+    __shared__ float lds_c2[WG_WAVES][G::C2];
+
+    for (int i = threadIdx.x; i < NW; i += WG_THREADS) lds_w[i] = wpack[i];
+    __syncthreads();
+
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const float* b1 = lds_w + G::OB1;
+
+    bf16x8 bfr[G::KSTEPS];
+    union BU { unsigned int d[4]; unsigned short u[8]; bf16x8 v; };
+    #pragma unroll
+    for (int st = 0; st < G::KSTEPS; ++st) {
+        BU bu;
+        #pragma unroll
+        for (int q = 0; q < 4; ++q)
+            bu.d[q] = ((const unsigned int*)bfrag)[(st * WAVE + lane) * 4 + q];
+        bfr[st] = bu.v;
+    }
+    // per-lane fragment base within a window (dwords): rows l&15, k-group l>>4
+    const int lane_dw = ((lane & 15) * G::CIN + (lane >> 4) * 8) / 2;
+
+    for (long win = blockIdx.x * WG_WAVES + wave; win < SN;
+         win += (long)gridDim.x * WG_WAVES) {
+        const unsigned int* xw =
+            (const unsigned int*)(x + win * (long)(G::L * G::CIN));
+        #pragma unroll 1
+        for (int mt = 0; mt < G::MTILES; ++mt) {
+            f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+            const unsigned int* base = xw + (long)mt * (16 * G::CIN / 2)
+                                       + lane_dw;
+            #pragma unroll
+            for (int st = 0; st < G::KSTEPS; ++st) {
+                BU au;
+                au.d[0] = base[st * 16 + 0];
+                au.d[1] = base[st * 16 + 1];
+                au.d[2] = base[st * 16 + 2];
+                au.d[3] = base[st * 16 + 3];
+                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(au.v, bfr[st],
+                                                              acc, 0, 0, 0);
+            }
+            const int c = lane & 15;
+            if (c < 4) {
+                #pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int row = mt * 16 + (lane >> 4) * 4 + r;
+                    if (row < G::C1)
+                        lds_c1[wave][c * G::C1 + row] = acc[r] + b1[c];
+                }
+            }
+        }
+        wave_sync();
+        conv_tail<G, true>(lane, lds_w, lds_c1[wave], lds_p1[wave],
+                           lds_c2[wave], feat, win);
+    }
+}
+
+// ---------------------------------------------------------------------------
 // Kernel 2: fused LSTM (batch-as-time) + Linear head + age gate (+sigmoid)
 // (K6+K7+K8+K9 of SURVEY.md §2.6)
 // ---------------------------------------------------------------------------
@@ -471,18 +557,30 @@ __global__ void debug_mfma_kernel(const unsigned short* __restrict__ A,
 namespace {
 
 template <class G>
-int launch_conv(const void* x, int x_is_bf16, float* feat, const float* wpack,
-                const void* bfrag, int SN, hipStream_t stream) {
+int launch_conv(const void* x, int x_is_bf16, int x_timelast, float* feat,
+                const float* wpack, const void* bfrag, int SN,
+                hipStream_t stream) {
     if (SN <= 0) return 0;
     int grid = (SN + WG_WAVES - 1) / WG_WAVES;
     if (grid > 8192) grid = 8192;
     if (x_is_bf16) {
         if (!bfrag) return -2;  // bf16 path requires packed B fragments
+        if (x_timelast) {
+            if constexpr (G::CIN % 2 == 0) {
+                hipLaunchKernelGGL((conv_stack_mfma_tlast_kernel<G>),
+                                   dim3(grid), dim3(WG_THREADS), 0, stream,
+                                   (const unsigned short*)x, feat, wpack,
+                                   (const unsigned short*)bfrag, SN);
+                return (int)hipGetLastError();
+            }
+            return -4;  // odd-CIN variants: use the staged layout
+        }
         hipLaunchKernelGGL((conv_stack_mfma_kernel<G>), dim3(grid),
                            dim3(WG_THREADS), 0, stream,
                            (const unsigned short*)x, feat, wpack,
                            (const unsigned short*)bfrag, SN);
     } else {
+        if (x_timelast) return -4;  // fp32 path is (C, L) only
         hipLaunchKernelGGL((conv_stack_kernel<G>), dim3(grid),
                            dim3(WG_THREADS), 0, stream, (const float*)x, feat,
                            wpack, SN);
@@ -508,14 +606,17 @@ int launch_lstm(const float* feat, const float* age, float* out,
 extern "C" {
 
 // variant: 0 = MyCNN5, 1 = MyCNN2/3, 2 = MyCNN4
-int tskd_conv_fwd(const void* x, int x_is_bf16, float* feat,
+int tskd_conv_fwd(const void* x, int x_is_bf16, int x_timelast, float* feat,
                   const float* wpack, const void* bfrag, int SN, int variant,
                   void* stream) {
     hipStream_t s = (hipStream_t)stream;
     switch (variant) {
-        case 0: return launch_conv<GeomCNN5>(x, x_is_bf16, feat, wpack, bfrag, SN, s);
-        case 1: return launch_conv<GeomCNN2>(x, x_is_bf16, feat, wpack, bfrag, SN, s);
-        case 2: return launch_conv<GeomCNN4>(x, x_is_bf16, feat, wpack, bfrag, SN, s);
+        case 0: return launch_conv<GeomCNN5>(x, x_is_bf16, x_timelast, feat,
+                                             wpack, bfrag, SN, s);
+        case 1: return launch_conv<GeomCNN2>(x, x_is_bf16, x_timelast, feat,
+                                             wpack, bfrag, SN, s);
+        case 2: return launch_conv<GeomCNN4>(x, x_is_bf16, x_timelast, feat,
+                                             wpack, bfrag, SN, s);
     }
     return -1;
 }

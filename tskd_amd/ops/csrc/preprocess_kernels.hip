@@ -283,10 +283,12 @@ __device__ __forceinline__ unsigned short f32_to_bf16_(float f) {
 
 // Vectorized: each thread emits 4 consecutive time points (WIN % 4 == 0),
 // reading contiguous proc and writing one 8 B (bf16) / 16 B (f32) store.
-template <class OT>
+// TLAST: emit (S, B, WIN, C) instead — the layout the LDS-free MFMA conv
+// kernel consumes (scalar scattered stores; the read side stays coalesced).
+template <class OT, bool TLAST = false>
 __global__ void window_gather_kernel(
     const float* __restrict__ proc,
-    OT* __restrict__ out,            // (S, B, C, WIN)
+    OT* __restrict__ out,            // (S, B, C, WIN) or (S, B, WIN, C)
     int S, int C, int G, int B, int WIN, int stride, long end)
 {
     const int WQ = WIN / 4;
@@ -307,15 +309,25 @@ __global__ void window_gather_kernel(
             v[j] = (g >= 0 && g < end && wend - WIN >= 0 && wend <= end)
                        ? pr[g % G] : 0.f;
         }
-        const long o = (((long)s * B + b) * C + c) * WIN + (long)tq * 4;
-        if constexpr (sizeof(OT) == 2) {
-            union { unsigned short h[4]; unsigned long long u; } pk;
+        if constexpr (TLAST) {
+            const long ob = ((long)s * B + b) * C * WIN;
             #pragma unroll
-            for (int j = 0; j < 4; ++j) pk.h[j] = f32_to_bf16_(v[j]);
-            *(unsigned long long*)(out + o) = pk.u;
+            for (int j = 0; j < 4; ++j) {
+                const long o = ob + (long)(tq * 4 + j) * C + c;
+                if constexpr (sizeof(OT) == 2) out[o] = f32_to_bf16_(v[j]);
+                else out[o] = (OT)v[j];
+            }
         } else {
-            float4 pk = {v[0], v[1], v[2], v[3]};
-            *(float4*)((float*)out + o) = pk;
+            const long o = (((long)s * B + b) * C + c) * WIN + (long)tq * 4;
+            if constexpr (sizeof(OT) == 2) {
+                union { unsigned short h[4]; unsigned long long u; } pk;
+                #pragma unroll
+                for (int j = 0; j < 4; ++j) pk.h[j] = f32_to_bf16_(v[j]);
+                *(unsigned long long*)(out + o) = pk.u;
+            } else {
+                float4 pk = {v[0], v[1], v[2], v[3]};
+                *(float4*)((float*)out + o) = pk;
+            }
         }
     }
 }
@@ -385,20 +397,33 @@ int tskd_preproc_window_fill(const float* bsum, const float* bcnt, float* proc,
 }
 
 int tskd_preproc_window_gather(const float* proc, void* out, int out_is_bf16,
-                               int S, int C, int G, int B, int WIN, int stride,
-                               long end, void* stream) {
+                               int out_timelast, int S, int C, int G, int B,
+                               int WIN, int stride, long end, void* stream) {
     if (WIN % 4 != 0) return -3;  // vectorized gather needs WIN % 4 == 0
     const long n = (long)S * B * C * (WIN / 4);
     if (n <= 0) return 0;
     hipStream_t st = (hipStream_t)stream;
-    if (out_is_bf16)
-        hipLaunchKernelGGL((window_gather_kernel<unsigned short>),
-                           dim3(grid_for(n, 256)), dim3(256), 0, st, proc,
-                           (unsigned short*)out, S, C, G, B, WIN, stride, end);
-    else
-        hipLaunchKernelGGL((window_gather_kernel<float>),
-                           dim3(grid_for(n, 256)), dim3(256), 0, st, proc,
-                           (float*)out, S, C, G, B, WIN, stride, end);
+    if (out_is_bf16) {
+        if (out_timelast)
+            hipLaunchKernelGGL((window_gather_kernel<unsigned short, true>),
+                               dim3(grid_for(n, 256)), dim3(256), 0, st, proc,
+                               (unsigned short*)out, S, C, G, B, WIN, stride,
+                               end);
+        else
+            hipLaunchKernelGGL((window_gather_kernel<unsigned short>),
+                               dim3(grid_for(n, 256)), dim3(256), 0, st, proc,
+                               (unsigned short*)out, S, C, G, B, WIN, stride,
+                               end);
+    } else {
+        if (out_timelast)
+            hipLaunchKernelGGL((window_gather_kernel<float, true>),
+                               dim3(grid_for(n, 256)), dim3(256), 0, st, proc,
+                               (float*)out, S, C, G, B, WIN, stride, end);
+        else
+            hipLaunchKernelGGL((window_gather_kernel<float>),
+                               dim3(grid_for(n, 256)), dim3(256), 0, st, proc,
+                               (float*)out, S, C, G, B, WIN, stride, end);
+    }
     return (int)hipGetLastError();
 }
 
