@@ -70,7 +70,7 @@ def main() -> None:
     p.add_argument("--warmup", type=int, default=15)
     p.add_argument("--mode", default="pipeline",
                    choices=["pipeline", "infer", "train"])
-    p.add_argument("--streams", type=int, default=16384,
+    p.add_argument("--streams", type=int, default=65536,
                    help="[pipeline] concurrent patient streams per GPU")
     p.add_argument("--seqs", type=int, default=4096,
                    help="[infer] concurrent sequences per GPU")
