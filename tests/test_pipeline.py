@@ -158,3 +158,58 @@ class TestDashboard:
         preds = client.get("/api/predictions").json()
         assert preds and preds[0]["patient"] == "p000194"
         assert abs(preds[0]["risk"] - 0.42) < 1e-6
+
+
+@pytest.mark.gpu
+class TestEndToEndGPU:
+    def test_full_pipeline_gpu_engines(self, tmp_path, cfg):
+        """The CLI stages with their GPU engines: bus -> ProcessStream(cuda)
+        -> call-stream -> PredictStream(cuda) -> store."""
+        from tskd_amd.bus import Bus
+        from tskd_amd.cli.predictstream import PredictStream
+        from tskd_amd.cli.processstream import ProcessStream
+        from tskd_amd.cli.sendstream import send_record_data
+        from tskd_amd.store import PredictionStore
+        bus = Bus(str(tmp_path / "bus"))
+        _write_wfdb_record(cfg.wavef_path, "p000194-test", cfg.channel_names,
+                           1 / 60, 40)
+        send_record_data(bus, "p000194-test", None, 1e6, 1.0, cfg)
+        ps = ProcessStream(bus, cfg, max_streams=4, device="cuda",
+                           starting="earliest")
+        emitted = ps.trigger()
+        torch.cuda.synchronize()
+        assert emitted > 0
+        store = PredictionStore(str(tmp_path / "pred.log"))
+        pr = PredictStream(bus, cfg, store, device="cuda",
+                           starting="earliest")
+        n = pr.trigger()
+        torch.cuda.synchronize()
+        assert n == 1 and store.count() == 1
+        _, risk = store.latest("p000194")
+        assert 0.0 <= risk <= 1.0
+
+    def test_gpu_cpu_processstream_equivalence(self, tmp_path, cfg):
+        """ProcessStream on cuda and cpu produce the same call-stream points."""
+        import json as _json
+
+        from tskd_amd.bus import Bus, Consumer
+        from tskd_amd.cli.processstream import ProcessStream
+        from tskd_amd.cli.sendstream import send_record_data
+        out = {}
+        for dev in ("cpu", "cuda"):
+            bus = Bus(str(tmp_path / f"bus_{dev}"))
+            _write_wfdb_record(cfg.wavef_path, "p000194-test",
+                               cfg.channel_names, 1 / 60, 36, seed=3)
+            send_record_data(bus, "p000194-test", None, 1e6, 1.0, cfg)
+            ps = ProcessStream(bus, cfg, max_streams=4, device=dev,
+                               starting="earliest")
+            ps.trigger()
+            torch.cuda.synchronize()
+            c = Consumer(bus, starting="earliest")
+            c.subscribe(["call-stream"])
+            msgs = c.poll(max_msgs=1024, timeout_ms=500)
+            out[dev] = {m.key.decode(): _json.loads(m.value) for m in msgs}
+        assert out["cpu"].keys() == out["cuda"].keys()
+        for k in out["cpu"]:
+            np.testing.assert_allclose(out["cpu"][k], out["cuda"][k],
+                                       rtol=1e-4, atol=1e-5)
