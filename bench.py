@@ -157,20 +157,25 @@ def main() -> None:
         gathered = [torch.empty(S, device=device) for _ in range(world)] \
             if dist else None
         graphed = None
+        tg = None
         if args.graph and dtype == torch.bfloat16:
+            from tskd_amd.engine.stream_engine import TriggerGraph
             from tskd_amd.ops import GraphedForward
             graphed = GraphedForward(eng, s=S, n=1, dtype=dtype,
                                      timelast=True)
+            # warm the rings to steady state, then capture the WHOLE trigger
+            # (ingest -> fill -> gather -> conv -> LSTM -> advance) as ONE
+            # hipGraph (BASELINE config 4: fused preprocess+inference graph)
+            while se.nproc < se.head - se.win_buckets + 1 or se.nproc == 0:
+                se.ingest_dense(raw, chan_map=chan_map)
+            torch.cuda.synchronize()
+            tg = TriggerGraph(se, raw, chan_map, graphed, stride=12)
 
         def step():
-            se.ingest_dense(raw, chan_map=chan_map)
-            if graphed is not None:
-                # gather writes the hipGraph's static input (timelast layout
-                # feeds the LDS-free MFMA conv); replay the captured sequence
-                se.windows(batch=1, stride=12, dtype=dtype, out=graphed.x,
-                           timelast=True)
-                probs = graphed.replay()
+            if tg is not None:
+                probs = tg.replay()
             else:
+                se.ingest_dense(raw, chan_map=chan_map)
                 w = se.windows(batch=1, stride=12, dtype=dtype)
                 probs = eng.forward(w, age, apply_sigmoid=True)
             if dist:

@@ -250,3 +250,53 @@ class TestStreamEngineGPU:
         np.testing.assert_allclose(
             gpu.proc[:, :, :gpu.nproc].cpu().numpy(),
             cpu.proc[:, :, :cpu.nproc].numpy(), rtol=1e-3, atol=1e-4)
+
+
+@pytest.mark.gpu
+class TestTriggerGraph:
+    def test_graphed_trigger_matches_eager(self):
+        """Whole-trigger hipGraph (ingest+fill+gather+model) vs an eager twin
+        engine fed the same raw chunks."""
+        from tskd_amd.engine.stream_engine import TriggerGraph
+        from tskd_amd.models import build_model
+        from tskd_amd.ops import GraphedForward, MyCNNEngine
+        fs = 125.0
+        S = 8
+        torch.manual_seed(5)
+        model = build_model("MyCNN5").eval()
+        me = MyCNNEngine(model, device="cuda")
+        chunks = [torch.randn(S, 8, int(fs * 60), device="cuda",
+                              dtype=torch.bfloat16) for _ in range(20)]
+        cm = list(range(8))
+
+        # eager twin
+        e1 = StreamEngine(S, 10, ring_grid=512, fs=fs, device="cuda")
+        eager_probs = []
+        for ch in chunks:
+            e1.ingest_dense(ch, chan_map=cm)
+            w = e1.windows(batch=1, stride=12, dtype=torch.bfloat16)
+            eager_probs.append(me.forward(
+                w, torch.full((S, 1), 65.0, device="cuda"),
+                apply_sigmoid=True).clone())
+
+        # graphed engine: warm 14 triggers eagerly, capture on the 15th
+        e2 = StreamEngine(S, 10, ring_grid=512, fs=fs, device="cuda")
+        raw_buf = chunks[0].clone()
+        idx = 0
+        while e2.nproc == 0 or e2.nproc < e2.head - e2.win_buckets + 1:
+            raw_buf.copy_(chunks[idx])
+            e2.ingest_dense(raw_buf, chan_map=cm)
+            idx += 1
+        torch.cuda.synchronize()
+        gf = GraphedForward(me, s=S, n=1, dtype=torch.bfloat16, timelast=True)
+        raw_buf.copy_(chunks[idx])  # consumed by TriggerGraph's warm run
+        tg = TriggerGraph(e2, raw_buf, cm, gf, stride=12)
+        idx += 1
+        while idx < len(chunks):
+            raw_buf.copy_(chunks[idx])
+            out = tg.replay()
+            torch.cuda.synchronize()
+            torch.testing.assert_close(out, eager_probs[idx], rtol=2e-3,
+                                       atol=2e-3)
+            idx += 1
+        assert e2.head == e1.head and e2.nproc == e1.nproc

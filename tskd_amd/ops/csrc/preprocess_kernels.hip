@@ -40,6 +40,22 @@ __device__ __forceinline__ float bf16_to_f32_(unsigned short u) {
     return v.f;
 }
 
+// Device-resident ring indices (hipGraph mode): dstate[0]=head,
+// dstate[1]=nproc. A null dstate falls back to the host-passed values.
+__device__ __forceinline__ long ring_head(const long long* dstate, long h) {
+    return dstate ? (long)dstate[0] : h;
+}
+__device__ __forceinline__ long ring_nproc(const long long* dstate, long p) {
+    return dstate ? (long)dstate[1] : p;
+}
+
+__global__ void advance_state_kernel(long long* dstate, int nb, int np) {
+    if (threadIdx.x == 0 && blockIdx.x == 0) {
+        dstate[0] += nb;
+        dstate[1] += np;
+    }
+}
+
 // Same-wave LDS RAW fence (see mycnn_kernels.hip wave_sync).
 __device__ __forceinline__ void wsync_() {
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -62,8 +78,10 @@ __global__ void ingest_dense_kernel(
     float* __restrict__ bcnt,
     const int* __restrict__ chan_map,  // (CIN) raw row -> wire channel
     int S, int CIN, int C, int T, int G,
-    int bucket_len, long head)      // buckets written at [head, head+NB)
+    int bucket_len, long head_in,   // buckets written at [head, head+NB)
+    const long long* __restrict__ dstate)
 {
+    const long head = ring_head(dstate, head_in);
     const int NB = T / bucket_len;
     const int NBG = (NB + ING_GRP - 1) / ING_GRP;
     const long nwaves = (long)S * CIN * NBG;
@@ -193,8 +211,10 @@ __global__ void clear_buckets_kernel(
 __global__ __launch_bounds__(256) void window_fill_kernel(
     const float* __restrict__ bsum, const float* __restrict__ bcnt,
     float* __restrict__ proc, float* __restrict__ last_val,
-    int S, int C, int G, long phead, int np, int win_buckets)
+    int S, int C, int G, long phead_in, int np, int win_buckets,
+    const long long* __restrict__ dstate)
 {
+    const long phead = ring_nproc(dstate, phead_in);
     constexpr int CHUNK = 64;             // output points per iteration
     const int MAXW = 64;                  // win_buckets <= 64 (default 36)
     __shared__ float lds_s[4][CHUNK + 64];  // bucket sums (CHUNK+MAXW-1 used)
@@ -289,8 +309,10 @@ template <class OT, bool TLAST = false>
 __global__ void window_gather_kernel(
     const float* __restrict__ proc,
     OT* __restrict__ out,            // (S, B, C, WIN) or (S, B, WIN, C)
-    int S, int C, int G, int B, int WIN, int stride, long end)
+    int S, int C, int G, int B, int WIN, int stride, long end_in,
+    const long long* __restrict__ dstate, int end_extra)
 {
+    const long end = dstate ? (long)dstate[1] + end_extra : end_in;
     const int WQ = WIN / 4;
     const long n = (long)S * B * C * WQ;
     for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
@@ -347,7 +369,8 @@ extern "C" {
 int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
                               float* bsum, float* bcnt, const int* chan_map,
                               int S, int CIN, int C, int T, int G,
-                              int bucket_len, long head, void* stream) {
+                              int bucket_len, long head,
+                              const long long* dstate, void* stream) {
     hipStream_t st = (hipStream_t)stream;
     const int NB = T / bucket_len;
     if (NB <= 0 || S <= 0) return 0;
@@ -356,11 +379,12 @@ int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
     if (raw_is_bf16)
         hipLaunchKernelGGL((ingest_dense_kernel<unsigned short>), dim3(grid),
                            dim3(256), 0, st, (const unsigned short*)raw, bsum,
-                           bcnt, chan_map, S, CIN, C, T, G, bucket_len, head);
+                           bcnt, chan_map, S, CIN, C, T, G, bucket_len, head,
+                           dstate);
     else
         hipLaunchKernelGGL((ingest_dense_kernel<float>), dim3(grid), dim3(256),
                            0, st, (const float*)raw, bsum, bcnt, chan_map, S,
-                           CIN, C, T, G, bucket_len, head);
+                           CIN, C, T, G, bucket_len, head, dstate);
     return (int)hipGetLastError();
 }
 
@@ -387,18 +411,28 @@ int tskd_preproc_clear_buckets(float* bsum, float* bcnt, int S, int C, int G,
 
 int tskd_preproc_window_fill(const float* bsum, const float* bcnt, float* proc,
                              float* last_val, int S, int C, int G, long phead,
-                             int np, int win_buckets, void* stream) {
+                             int np, int win_buckets,
+                             const long long* dstate, void* stream) {
     if (np <= 0) return 0;
     const long nsc = (long)S * C * WAVE;  // one wave per (stream, channel)
     hipLaunchKernelGGL(window_fill_kernel, dim3(grid_for(nsc, 256)), dim3(256),
                        0, (hipStream_t)stream, bsum, bcnt, proc, last_val, S,
-                       C, G, phead, np, win_buckets);
+                       C, G, phead, np, win_buckets, dstate);
+    return (int)hipGetLastError();
+}
+
+int tskd_preproc_advance_state(long long* dstate, int nb, int np,
+                               void* stream) {
+    hipLaunchKernelGGL(advance_state_kernel, dim3(1), dim3(64), 0,
+                       (hipStream_t)stream, dstate, nb, np);
     return (int)hipGetLastError();
 }
 
 int tskd_preproc_window_gather(const float* proc, void* out, int out_is_bf16,
                                int out_timelast, int S, int C, int G, int B,
-                               int WIN, int stride, long end, void* stream) {
+                               int WIN, int stride, long end,
+                               const long long* dstate, int end_extra,
+                               void* stream) {
     if (WIN % 4 != 0) return -3;  // vectorized gather needs WIN % 4 == 0
     const long n = (long)S * B * C * (WIN / 4);
     if (n <= 0) return 0;
@@ -408,21 +442,23 @@ int tskd_preproc_window_gather(const float* proc, void* out, int out_is_bf16,
             hipLaunchKernelGGL((window_gather_kernel<unsigned short, true>),
                                dim3(grid_for(n, 256)), dim3(256), 0, st, proc,
                                (unsigned short*)out, S, C, G, B, WIN, stride,
-                               end);
+                               end, dstate, end_extra);
         else
             hipLaunchKernelGGL((window_gather_kernel<unsigned short>),
                                dim3(grid_for(n, 256)), dim3(256), 0, st, proc,
                                (unsigned short*)out, S, C, G, B, WIN, stride,
-                               end);
+                               end, dstate, end_extra);
     } else {
         if (out_timelast)
             hipLaunchKernelGGL((window_gather_kernel<float, true>),
                                dim3(grid_for(n, 256)), dim3(256), 0, st, proc,
-                               (float*)out, S, C, G, B, WIN, stride, end);
+                               (float*)out, S, C, G, B, WIN, stride, end,
+                               dstate, end_extra);
         else
             hipLaunchKernelGGL((window_gather_kernel<float>),
                                dim3(grid_for(n, 256)), dim3(256), 0, st, proc,
-                               (float*)out, S, C, G, B, WIN, stride, end);
+                               (float*)out, S, C, G, B, WIN, stride, end,
+                               dstate, end_extra);
     }
     return (int)hipGetLastError();
 }
