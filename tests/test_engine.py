@@ -300,3 +300,43 @@ class TestTriggerGraph:
                                        atol=2e-3)
             idx += 1
         assert e2.head == e1.head and e2.nproc == e1.nproc
+
+    def test_trigger_graph_survives_ring_wrap(self):
+        """Replay the captured trigger past several full ring turns: the
+        device-side mod-G indexing must keep matching the eager twin."""
+        from tskd_amd.engine.stream_engine import TriggerGraph
+        from tskd_amd.models import build_model
+        from tskd_amd.ops import GraphedForward, MyCNNEngine
+        fs, S, G = 25.0, 4, 192  # 12 buckets/trigger -> wraps every 16
+        torch.manual_seed(6)
+        me = MyCNNEngine(build_model("MyCNN5").eval(), device="cuda")
+        gen = torch.Generator(device="cuda").manual_seed(7)
+        n_triggers = 60  # ~3.75 ring turns
+        chunks = [torch.randn(S, 8, int(fs * 60), device="cuda",
+                              generator=gen, dtype=torch.float32)
+                  .to(torch.bfloat16) for _ in range(n_triggers)]
+        cm = list(range(8))
+        e1 = StreamEngine(S, 10, ring_grid=G, fs=fs, device="cuda")
+        eager = []
+        for ch in chunks:
+            e1.ingest_dense(ch, chan_map=cm)
+            w = e1.windows(batch=1, stride=12, dtype=torch.bfloat16)
+            eager.append(me.forward(w, None, apply_sigmoid=True).clone())
+        e2 = StreamEngine(S, 10, ring_grid=G, fs=fs, device="cuda")
+        raw_buf = chunks[0].clone()
+        idx = 0
+        while e2.nproc == 0 or e2.nproc < e2.head - e2.win_buckets + 1:
+            raw_buf.copy_(chunks[idx]); e2.ingest_dense(raw_buf, chan_map=cm)
+            idx += 1
+        torch.cuda.synchronize()
+        gf = GraphedForward(me, s=S, n=1, dtype=torch.bfloat16, timelast=True)
+        gf.age.zero_()
+        raw_buf.copy_(chunks[idx])
+        tg = TriggerGraph(e2, raw_buf, cm, gf, stride=12)
+        idx += 1
+        while idx < n_triggers:
+            raw_buf.copy_(chunks[idx])
+            out = tg.replay()
+            torch.cuda.synchronize()
+            torch.testing.assert_close(out, eager[idx], rtol=2e-3, atol=2e-3)
+            idx += 1
