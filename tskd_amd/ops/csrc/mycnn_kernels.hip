@@ -367,39 +367,62 @@ __global__ __launch_bounds__(WG_THREADS) void conv_stack_mfma_tlast_kernel(
     }
     // per-lane fragment base within a window (dwords): rows l&15, k-group l>>4
     const int lane_dw = ((lane & 15) * G::CIN + (lane >> 4) * 8) / 2;
+    __shared__ float lds_c1b[WG_WAVES][4 * G::C1];  // 2nd window of the pair
 
-    for (long win = blockIdx.x * WG_WAVES + wave; win < SN;
-         win += (long)gridDim.x * WG_WAVES) {
-        const unsigned int* xw =
+    // TWO windows per wave: the pair's MFMA/load chains are independent, so
+    // the in-order wave covers one chain's global-load latency with the
+    // other's MFMAs (the single-window chain was latency-exposed).
+    for (long win = (blockIdx.x * WG_WAVES + wave) * 2; win < SN;
+         win += (long)gridDim.x * WG_WAVES * 2) {
+        const long win1 = win + 1;
+        const bool has1 = win1 < SN;
+        const unsigned int* xwa =
             (const unsigned int*)(x + win * (long)(G::L * G::CIN));
+        const unsigned int* xwb =
+            (const unsigned int*)(x + (has1 ? win1 : win) *
+                                  (long)(G::L * G::CIN));
         #pragma unroll 1
         for (int mt = 0; mt < G::MTILES; ++mt) {
-            f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-            const unsigned int* base = xw + (long)mt * (16 * G::CIN / 2)
-                                       + lane_dw;
+            f32x4 acca = {0.f, 0.f, 0.f, 0.f};
+            f32x4 accb = {0.f, 0.f, 0.f, 0.f};
+            const unsigned int* basea = xwa + (long)mt * (16 * G::CIN / 2)
+                                        + lane_dw;
+            const unsigned int* baseb = xwb + (long)mt * (16 * G::CIN / 2)
+                                        + lane_dw;
             #pragma unroll
             for (int st = 0; st < G::KSTEPS; ++st) {
-                BU au;
-                au.d[0] = base[st * 16 + 0];
-                au.d[1] = base[st * 16 + 1];
-                au.d[2] = base[st * 16 + 2];
-                au.d[3] = base[st * 16 + 3];
-                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(au.v, bfr[st],
-                                                              acc, 0, 0, 0);
+                BU aa, ab;
+                aa.d[0] = basea[st * 16 + 0];
+                aa.d[1] = basea[st * 16 + 1];
+                aa.d[2] = basea[st * 16 + 2];
+                aa.d[3] = basea[st * 16 + 3];
+                ab.d[0] = baseb[st * 16 + 0];
+                ab.d[1] = baseb[st * 16 + 1];
+                ab.d[2] = baseb[st * 16 + 2];
+                ab.d[3] = baseb[st * 16 + 3];
+                acca = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aa.v, bfr[st],
+                                                               acca, 0, 0, 0);
+                accb = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ab.v, bfr[st],
+                                                               accb, 0, 0, 0);
             }
             const int c = lane & 15;
             if (c < 4) {
                 #pragma unroll
                 for (int r = 0; r < 4; ++r) {
                     const int row = mt * 16 + (lane >> 4) * 4 + r;
-                    if (row < G::C1)
-                        lds_c1[wave][c * G::C1 + row] = acc[r] + b1[c];
+                    if (row < G::C1) {
+                        lds_c1[wave][c * G::C1 + row] = acca[r] + b1[c];
+                        lds_c1b[wave][c * G::C1 + row] = accb[r] + b1[c];
+                    }
                 }
             }
         }
         wave_sync();
         conv_tail<G, true>(lane, lds_w, lds_c1[wave], lds_p1[wave],
                            lds_c2[wave], feat, win);
+        if (has1)
+            conv_tail<G, true>(lane, lds_w, lds_c1b[wave], lds_p1[wave],
+                               lds_c2[wave], feat, win1);
     }
 }
 
