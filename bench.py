@@ -72,7 +72,7 @@ def main() -> None:
                    choices=["pipeline", "infer", "train"])
     p.add_argument("--streams", type=int, default=65536,
                    help="[pipeline] concurrent patient streams per GPU")
-    p.add_argument("--seqs", type=int, default=4096,
+    p.add_argument("--seqs", type=int, default=8192,
                    help="[infer] concurrent sequences per GPU")
     p.add_argument("--batch", type=int, default=1024,
                    help="[infer] windows per sequence batch")
