@@ -159,17 +159,23 @@ def main() -> None:
         graphed = None
         tg = None
         if args.graph and dtype == torch.bfloat16:
-            from tskd_amd.engine.stream_engine import TriggerGraph
-            from tskd_amd.ops import GraphedForward
-            graphed = GraphedForward(eng, s=S, n=1, dtype=dtype,
-                                     timelast=True)
             # warm the rings to steady state, then capture the WHOLE trigger
             # (ingest -> fill -> gather -> conv -> LSTM -> advance) as ONE
-            # hipGraph (BASELINE config 4: fused preprocess+inference graph)
-            while se.nproc < se.head - se.win_buckets + 1 or se.nproc == 0:
-                se.ingest_dense(raw, chan_map=chan_map)
-            torch.cuda.synchronize()
-            tg = TriggerGraph(se, raw, chan_map, graphed, stride=12)
+            # hipGraph (BASELINE config 4: fused preprocess+inference graph).
+            # Any capture failure falls back to the eager kernel path.
+            try:
+                from tskd_amd.engine.stream_engine import TriggerGraph
+                from tskd_amd.ops import GraphedForward
+                graphed = GraphedForward(eng, s=S, n=1, dtype=dtype,
+                                         timelast=True)
+                while se.nproc < se.head - se.win_buckets + 1 or se.nproc == 0:
+                    se.ingest_dense(raw, chan_map=chan_map)
+                torch.cuda.synchronize()
+                tg = TriggerGraph(se, raw, chan_map, graphed, stride=12)
+            except Exception as e:  # pragma: no cover - fallback safety
+                print(f"[bench] hipGraph capture unavailable ({e}); "
+                      "running eager", file=sys.stderr)
+                tg = None
 
         def step():
             if tg is not None:

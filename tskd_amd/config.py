@@ -14,7 +14,7 @@ import configparser
 import os
 import sys
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 # Wire/tensor channel order — the index order IS the channel order on the wire
 # and in the (N, C, 120) model tensor (reference config.cfg:23 CHANNEL_NAMES).

@@ -18,7 +18,7 @@ Semantics (event-time, all at speed=1):
 
 from __future__ import annotations
 
-from typing import Optional, Sequence, Tuple
+from typing import Tuple
 
 import numpy as np
 
