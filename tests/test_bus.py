@@ -133,3 +133,22 @@ class TestCrossProcess:
         for tag in ("A", "B"):
             vals = [m.value.decode() for m in got if m.key.decode() == f"key{tag}"]
             assert vals == [f"{tag}:{i}" for i in range(50)]
+
+
+class TestNativeParse:
+    def test_poll_samples(self, bus):
+        bus.create_topic("HR")
+        c = Consumer(bus, starting="earliest")
+        c.subscribe(["HR"])
+        p = Producer(bus)
+        p.produce_sample("HR", "p000194", 0, 71.5, ts_us=5_000_000)
+        p.produce("HR", "p000194", "not json")  # skipped, offset advances
+        p.produce_sample("HR", "p000194", 3, -2.25, ts_us=6_000_000)
+        keys, topics, chans, vals, ts = c.poll_samples(timeout_ms=500)
+        assert keys == ["p000194", "p000194"]
+        assert topics == ["HR", "HR"]
+        assert chans.tolist() == [0, 3]
+        assert vals.tolist() == [71.5, -2.25]
+        assert ts.tolist() == [5.0, 6.0]
+        # the malformed message was consumed too (no re-delivery)
+        assert c.poll(timeout_ms=10) == []

@@ -92,3 +92,17 @@ class TestMetrics:
         assert snap["calls"] == 5 and snap["items"] == 60
         assert snap["p50_ms"] >= 1.0
         assert "preprocess" in pm.report()
+
+
+class TestCsvDemo:
+    def test_config1_csv_to_mycnn2_cpu(self, tmp_path):
+        """BASELINE config 1: data.csv replay -> MyCNN2 eager CPU."""
+        from tskd_amd.cli.makedata import make_data
+        from tskd_amd.utils import run_csv_demo
+        p = str(tmp_path / "data.csv")
+        make_data(p, seed=42, hours=2.0)
+        out = run_csv_demo(p, variant="MyCNN2")
+        assert out["n_windows"] >= 1
+        assert ((out["scores"] >= 0) & (out["scores"] <= 1)).all()
+        # 2 h at 1 Hz with 50% dropout -> full 5-s grid coverage
+        assert out["grid_points"] > 1000

@@ -100,6 +100,13 @@ class Consumer:
     def poll(self, max_msgs: int = 256, timeout_ms: int = 0) -> List[Message]:
         return self._c.poll(max_msgs, timeout_ms)
 
+    def poll_samples(self, max_msgs: int = 4096, timeout_ms: int = 0):
+        """Poll + parse the reference "[chan, value]" wire format natively
+        (C++): returns (keys, topics, chan int32[], value float32[],
+        ts float64[] seconds). Unparseable messages are skipped (their
+        offsets still advance)."""
+        return self._c.poll_samples(max_msgs, timeout_ms)
+
     def seek(self, topic: str, partition: int, offset: int) -> None:
         self._c.seek(topic, partition, offset)
 

@@ -14,7 +14,6 @@ speed-invariant.
 from __future__ import annotations
 
 import argparse
-import json
 import logging
 import signal
 import time
@@ -58,28 +57,22 @@ class ProcessStream:
         return self.pid_index[pid]
 
     def trigger(self) -> int:
-        """One micro-batch: drain bus, ingest events, emit new points."""
-        msgs = self.consumer.poll(max_msgs=65536, timeout_ms=0)
-        si, ci, ts, vv = [], [], [], []
-        for m in msgs:
-            try:
-                chan, val = json.loads(m.value)
-            except (ValueError, TypeError):
-                continue
-            si.append(self._sid(m.key.decode()))
-            ci.append(int(chan))
-            ts.append(m.ts_us / 1e6)
-            vv.append(float(val))
+        """One micro-batch: drain bus (native C++ wire-format parse), ingest
+        events, emit new points."""
+        keys, _topics, chans, vals, ts_arr = self.consumer.poll_samples(
+            max_msgs=65536, timeout_ms=0)
+        si = [self._sid(k) for k in keys]
+        n_in = len(si)
         nproc_before = self.engine.nproc
-        if ts:
-            self.hwm = max(self.hwm, max(ts))
+        if n_in:
+            self.hwm = max(self.hwm, float(ts_arr.max()))
         advance = max(0.0, self.hwm - self.watermark_s)
         if si:
             self.engine.ingest_events(
                 torch.tensor(si, dtype=torch.long),
-                torch.tensor(ci, dtype=torch.long),
-                torch.tensor(ts, dtype=torch.float64),
-                torch.tensor(vv, dtype=torch.float32),
+                torch.from_numpy(chans).long(),
+                torch.from_numpy(ts_arr),
+                torch.from_numpy(vals),
                 advance_to=advance)
         else:
             # watermark can still advance processed points
@@ -104,7 +97,7 @@ class ProcessStream:
                 emitted += 1
         self.producer.flush(self.out_topic)
         log.info("trigger: %d msgs in, %d new points x %d keys",
-                 len(msgs), np_new, emitted)
+                 n_in, np_new, emitted)
         return emitted
 
 

@@ -14,6 +14,7 @@
 // producer acks=all (here: append+commit is the ack; flush() msyncs),
 // consumer startingOffsets latest|earliest, seek to byte offset for replay.
 
+#include <pybind11/numpy.h>
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
@@ -345,6 +346,25 @@ struct Message {
     std::string key_, val_;
 };
 
+// Fast parse of the reference wire value format "[<chan>, <float>]"
+// (sendStream.py:59-64) — the Spark-SQL CAST/substring parsing replacement.
+static bool parse_sample(const char* p, size_t n, long* chan, double* val) {
+    const char* end = p + n;
+    while (p < end && isspace((unsigned char)*p)) ++p;
+    if (p >= end || *p != '[') return false;
+    ++p;
+    char* q = nullptr;
+    *chan = strtol(p, &q, 10);
+    if (q == p) return false;
+    p = q;
+    while (p < end && (isspace((unsigned char)*p) || *p == ',')) ++p;
+    *val = strtod(p, &q);
+    if (q == p) return false;
+    p = q;
+    while (p < end && isspace((unsigned char)*p)) ++p;
+    return p < end && *p == ']';
+}
+
 class Consumer {
 public:
     Consumer(std::shared_ptr<Bus> bus, std::string starting = "latest")
@@ -402,6 +422,35 @@ public:
 
     std::map<std::string, uint64_t> positions() const { return pos_; }
 
+    // Poll + parse the "[chan, value]" wire format natively: returns
+    // (keys, topics, chan int32[], value float32[], ts float64[] seconds).
+    py::tuple poll_samples(int max_msgs = 4096, int timeout_ms = 0) {
+        auto msgs = poll(max_msgs, timeout_ms);
+        std::vector<std::string> keys, topics;
+        std::vector<int> chans;
+        std::vector<float> vals;
+        std::vector<double> tss;
+        keys.reserve(msgs.size());
+        for (auto& m : msgs) {
+            long chan;
+            double val;
+            if (!parse_sample(m.val_.data(), m.val_.size(), &chan, &val))
+                continue;
+            keys.push_back(m.key_);
+            topics.push_back(m.topic);
+            chans.push_back((int)chan);
+            vals.push_back((float)val);
+            tss.push_back((double)m.ts_us / 1e6);
+        }
+        auto ca = py::array_t<int>((py::ssize_t)chans.size());
+        auto va = py::array_t<float>((py::ssize_t)vals.size());
+        auto ta = py::array_t<double>((py::ssize_t)tss.size());
+        std::memcpy(ca.mutable_data(), chans.data(), chans.size() * 4);
+        std::memcpy(va.mutable_data(), vals.data(), vals.size() * 4);
+        std::memcpy(ta.mutable_data(), tss.data(), tss.size() * 8);
+        return py::make_tuple(keys, topics, ca, va, ta);
+    }
+
 private:
     static int64_t now_us() {
         struct timespec ts{};
@@ -450,5 +499,7 @@ PYBIND11_MODULE(_tskd_bus, m) {
         .def("seek", &Consumer::seek)
         .def("poll", &Consumer::poll, py::arg("max_msgs") = 256,
              py::arg("timeout_ms") = 0)
+        .def("poll_samples", &Consumer::poll_samples,
+             py::arg("max_msgs") = 4096, py::arg("timeout_ms") = 0)
         .def("positions", &Consumer::positions);
 }

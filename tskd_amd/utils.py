@@ -107,3 +107,34 @@ def run_offline_demo(model=None, record_name: Optional[str] = None,
         scores = torch.sigmoid(model(x, age)).numpy()
     return {"n_windows": len(wins), "scores": scores,
             "grid_points": len(df)}
+
+
+def run_csv_demo(csv_path: str, variant: str = "MyCNN2",
+                 model=None) -> dict:
+    """BASELINE config 1: single-patient CSV replay (timestamp,value) ->
+    offline preprocess -> MyCNN2 eager PyTorch on CPU. No bus, no GPU.
+
+    The CSV stream is a single channel; it lands in wire channel 0 with the
+    remaining model channels zero (the reference's missing-channel rule).
+    """
+    from tskd_amd.engine.windowing import (MODEL_WIN, preprocess_series_oracle,
+                                           sliding_windows)
+    from tskd_amd.models import build_model
+    rows = np.genfromtxt(csv_path, delimiter=",", names=True)
+    ts = np.asarray(rows["timestamp"], float)
+    vals = np.asarray(rows["value"], float)
+    n_buckets = int(ts.max() // 5) + 1
+    series = preprocess_series_oracle(ts, vals, n_buckets)
+    model = model or build_model(variant)
+    model = model.eval()
+    cin = int(model.IN_CHANNELS)
+    grid = np.zeros((len(series), cin))
+    grid[:, 0] = series
+    wins = sliding_windows(grid, MODEL_WIN, 0.4)
+    if len(wins) == 0:
+        return {"n_windows": 0, "scores": np.empty(0)}
+    with torch.no_grad():
+        x = torch.from_numpy(wins).float()
+        scores = torch.sigmoid(model(x, torch.full((len(wins),), 65.0)))
+    return {"n_windows": len(wins), "grid_points": len(series),
+            "scores": scores.numpy()}
