@@ -14,6 +14,7 @@ speed-invariant.
 from __future__ import annotations
 
 import argparse
+import json
 import logging
 import signal
 import time
