@@ -444,8 +444,8 @@ __global__ __launch_bounds__(WG_THREADS) void lstm_head_kernel(
     int S, int N, float age_eps, int apply_sigmoid)
 {
     constexpr int LIN = G::LIN;
-    constexpr int LINP = (LIN + 3) / 4 * 4;  // float4-padded chunk row
-    __shared__ float lds_feat[WG_WAVES][CHUNK * LINP];
+    constexpr int PRE_R = (CHUNK * LIN + WAVE - 1) / WAVE;  // regs per lane
+    __shared__ float lds_feat[WG_WAVES][CHUNK * LIN];
 
     const int wave = threadIdx.x / WAVE;
     const int lane = threadIdx.x % WAVE;
@@ -478,17 +478,36 @@ __global__ __launch_bounds__(WG_THREADS) void lstm_head_kernel(
         #pragma unroll
         for (int u = 0; u < 16; ++u) h1v[u] = h2v[u] = 0.f;
 
+        // Double-buffered chunk pipeline: chunk c+1's global loads are
+        // ISSUED before chunk c's compute (their s_waitcnt lands at the LDS
+        // write after the compute), so the scan never stalls on HBM latency
+        // at chunk boundaries.
+        const long ntot = (long)N * LIN;
+        float pre[PRE_R];
+        #pragma unroll
+        for (int r = 0; r < PRE_R; ++r) {
+            const long idx = (long)r * WAVE + lane;
+            pre[r] = idx < ntot ? fseq[idx] : 0.f;
+        }
         for (int t0 = 0; t0 < N; t0 += CHUNK) {
             const int tn = min(CHUNK, N - t0);
-            // Prefetch a feature chunk into LDS (row stride LINP).
-            for (int i = lane; i < tn * LIN; i += WAVE) {
-                const int tt = i / LIN, j = i % LIN;
-                fw[tt * LINP + j] = fseq[(long)(t0 + tt) * LIN + j];
+            #pragma unroll
+            for (int r = 0; r < PRE_R; ++r) {
+                const int j = r * WAVE + lane;
+                if (j < CHUNK * LIN) fw[j] = pre[r];
             }
             wave_sync();
+            if (t0 + CHUNK < N) {
+                const long base = (long)(t0 + CHUNK) * LIN;
+                #pragma unroll
+                for (int r = 0; r < PRE_R; ++r) {
+                    const long idx = base + (long)r * WAVE + lane;
+                    pre[r] = idx < ntot ? fseq[idx] : 0.f;
+                }
+            }
 
             for (int tt = 0; tt < tn; ++tt) {
-                const float* xt = fw + tt * LINP;
+                const float* xt = fw + tt * LIN;
                 // ----- layer 1: LIN-dim x-dot (LDS broadcast, 2 partial
                 //       accumulators) + 16-dim h-dot (register FMA) -----
                 float ga = bl1, gb = 0.f;

@@ -34,6 +34,9 @@
 
 #define WAVE 64
 
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4_;
+typedef __attribute__((ext_vector_type(4))) float f32x4_;
+
 __device__ __forceinline__ float bf16_to_f32_(unsigned short u) {
     union { unsigned int i; float f; } v;
     v.i = ((unsigned int)u) << 16;
@@ -109,10 +112,11 @@ __global__ void ingest_dense_kernel(
                     if (!isnan(f)) { sum += f; cnt += 1.f; }
                 }
                 const int oct = (bucket_len - pre) / 8;
-                const uint4* vp = (const uint4*)((const unsigned short*)src + pre);
+                const u32x4_* vp = (const u32x4_*)((const unsigned short*)src + pre);
                 for (int p = lane; p < oct; p += ING_GL) {
-                    union { uint4 q; unsigned short h[8]; } v;
-                    v.q = vp[p];
+                    union { u32x4_ q; unsigned short h[8]; } v;
+                    // raw samples are consumed exactly once: stream past L2
+                    v.q = __builtin_nontemporal_load(&vp[p]);
                     #pragma unroll
                     for (int j = 0; j < 8; ++j) {
                         const float f = bf16_to_f32_(v.h[j]);
@@ -131,9 +135,9 @@ __global__ void ingest_dense_kernel(
                     if (!isnan(f)) { sum += f; cnt += 1.f; }
                 }
                 const int quad = (bucket_len - pre) / 4;
-                const float4* vp = (const float4*)((const float*)src + pre);
+                const f32x4_* vp = (const f32x4_*)((const float*)src + pre);
                 for (int p = lane; p < quad; p += ING_GL) {
-                    const float4 v = vp[p];
+                    const f32x4_ v = __builtin_nontemporal_load(&vp[p]);
                     if (!isnan(v.x)) { sum += v.x; cnt += 1.f; }
                     if (!isnan(v.y)) { sum += v.y; cnt += 1.f; }
                     if (!isnan(v.z)) { sum += v.z; cnt += 1.f; }
