@@ -176,3 +176,65 @@ class TestWfdb:
                                    np.array([100, 300, 500]) / 200.0)
         np.testing.assert_allclose(rec.p_signal[:, 1],
                                    np.array([-200, -400, -600]) / 100.0)
+
+
+class TestWfdbRobustness:
+    def _mk(self, d, hea, dat=None, name="r"):
+        open(f"{d}/{name}.hea", "w").write(hea)
+        if dat is not None:
+            np.asarray(dat, dtype="<i2").tofile(f"{d}/{name}.dat")
+        from tskd_amd.io import rdrecord
+        return lambda **kw: rdrecord(f"{d}/{name}", **kw)
+
+    def test_truncated_dat_reads_nan(self, tmp_path):
+        # header says 10 samples, file holds 4 -> the tail decodes as NaN
+        rd = self._mk(str(tmp_path), "r 1 125 10\nr.dat 16 10(0)/u 16 0 0 0 0 S\n",
+                      [10, 20, 30, 40])
+        rec = rd()
+        assert rec.p_signal.shape == (10, 1)
+        assert np.isfinite(rec.p_signal[:4, 0]).all()
+        assert np.isnan(rec.p_signal[4:, 0]).all()
+
+    def test_invalid_sentinel_is_nan(self, tmp_path):
+        rd = self._mk(str(tmp_path), "r 1 125 3\nr.dat 16 10(0)/u 16 0 0 0 0 S\n",
+                      [100, -32768, 200])
+        rec = rd()
+        assert np.isnan(rec.p_signal[1, 0])
+        np.testing.assert_allclose(rec.p_signal[[0, 2], 0], [10.0, 20.0])
+
+    def test_unsupported_format_raises(self, tmp_path):
+        rd = self._mk(str(tmp_path), "r 1 125 2\nr.dat 24 10(0)/u 24 0 0 0 0 S\n",
+                      [1, 2])
+        with pytest.raises(RuntimeError, match="unsupported format"):
+            rd()
+
+    def test_multisegment_rejected(self, tmp_path):
+        rd = self._mk(str(tmp_path), "r/3 2 125 100\n")
+        with pytest.raises(RuntimeError, match="multi-segment"):
+            rd()
+
+    def test_missing_files(self, tmp_path):
+        from tskd_amd.io import rdrecord
+        with pytest.raises(RuntimeError, match="cannot open"):
+            rdrecord(str(tmp_path / "nope"))
+        rd = self._mk(str(tmp_path), "r 1 125 2\nr.dat 16 10(0)/u 16 0 0 0 0 S\n")
+        with pytest.raises(RuntimeError, match="cannot open"):
+            rd()  # .hea exists, .dat missing
+
+    def test_comments_and_baseline(self, tmp_path):
+        hea = ("# a comment line\n"
+               "r 1 125 3 14:00:00 01/02/2100\n"
+               "# another\n"
+               "r.dat 16 200(100)/mV 16 0 0 0 0 ECG\n")
+        rd = self._mk(str(tmp_path), hea, [100, 300, 500])
+        rec = rd()
+        # physical = (adc - baseline 100) / gain 200
+        np.testing.assert_allclose(rec.p_signal[:, 0], [0.0, 1.0, 2.0])
+        assert rec.base_datetime.year == 2100
+
+    def test_unknown_channel_selection_skipped(self, tmp_path):
+        rd = self._mk(str(tmp_path), "r 1 125 2\nr.dat 16 10(0)/u 16 0 0 0 0 HR\n",
+                      [1, 2])
+        rec = rd(channel_names=["SpO2", "HR"])  # SpO2 absent
+        assert rec.sig_name == ["HR"]
+        assert rec.p_signal.shape == (2, 1)
