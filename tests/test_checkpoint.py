@@ -130,3 +130,20 @@ class TestReferenceCheckpoints:
         m2 = load_checkpoint(p2)
         for a, b in zip(m.state_dict().values(), m2.state_dict().values()):
             assert torch.equal(a, b)
+
+
+class TestGoldenWindow:
+    """The reference keeps a fixed 120x10 window (explore_output/X.TESTINPUT)
+    re-scored by notebook cells and plot.py as a de-facto golden fixture
+    (SURVEY.md §4); here the fixture's expected score is committed too."""
+
+    def test_golden_window_score_stable(self):
+        import numpy as np
+        x = np.loadtxt("tests/data/x_testinput.csv", delimiter=",")
+        assert x.shape == (120, 10)
+        m = load_checkpoint("tests/data/ref_mycnn5.pth")
+        with torch.no_grad():
+            xt = torch.from_numpy(x.T[None]).float()
+            y = torch.sigmoid(m(xt, torch.full((1,), 65.0)))
+        golden = float(open("tests/data/x_testinput.golden").read())
+        assert abs(float(y[0]) - golden) < 1e-6

@@ -158,6 +158,8 @@ class TestDashboard:
         preds = client.get("/api/predictions").json()
         assert preds and preds[0]["patient"] == "p000194"
         assert abs(preds[0]["risk"] - 0.42) < 1e-6
+        h = client.get("/health").json()
+        assert h["status"] == "ok" and h["predictions"] == 1
 
 
 @pytest.mark.gpu

@@ -117,6 +117,12 @@ def build_app(state: DashState):
     def index():
         return PAGE
 
+    @app.get("/health")
+    def health():
+        # the reference's db healthcheck analog (docker-compose.yml:138-140)
+        return {"status": "ok", "predictions": state.store.count(),
+                "patients": len(set(state.raw) | set(state.proc))}
+
     @app.get("/api/patients")
     def patients():
         with state.lock:

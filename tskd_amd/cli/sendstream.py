@@ -95,8 +95,16 @@ def main(argv=None) -> None:
                     help="patient records (default: config PATIENTRECORDS)")
     ap.add_argument("--bus-dir", default=None)
     ap.add_argument("--csv", default=None, help="CSV replay mode input file")
+    ap.add_argument("--log-file", default=None,
+                    help="also log to this file (the reference writes "
+                         "${DATAPATH}/producer.log, sendStream.py:15-21)")
     ap.add_argument("--topic", default="data", help="[csv mode] topic")
     args = ap.parse_args(argv)
+    if args.log_file:
+        fh = logging.FileHandler(args.log_file)
+        fh.setFormatter(logging.Formatter(
+            "%(asctime)s %(name)s %(levelname)s %(message)s"))
+        logging.getLogger().addHandler(fh)
 
     bus = Bus(args.bus_dir)
     if args.csv:
