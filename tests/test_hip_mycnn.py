@@ -36,11 +36,14 @@ def test_build_cross_compiles():
 
 @pytest.mark.gpu
 class TestHipForward:
-    @pytest.mark.parametrize("variant", ["MyCNN5", "MyCNN2", "MyCNN4"])
-    def test_fp32_parity(self, variant):
+    @pytest.mark.parametrize("variant,n", [("MyCNN5", 64), ("MyCNN5", 37),
+                                           ("MyCNN5", 1), ("MyCNN2", 64),
+                                           ("MyCNN4", 64)])
+    def test_fp32_parity(self, variant, n):
+        # n covers CHUNK-multiple, odd, and single-step scans
         m = build_model(variant).eval()
         eng = MyCNNEngine(m, device="cuda")
-        x = _x(3, 64, m.IN_CHANNELS, seed=1)
+        x = _x(3, n, m.IN_CHANNELS, seed=1)
         ref = _oracle(m, x)
         got = eng.forward(x.cuda()).cpu()
         torch.testing.assert_close(got, ref, rtol=2e-4, atol=2e-5)
