@@ -205,6 +205,8 @@ class MyCNNEngine:
             out = self.lstm_head(feat, age, apply_sigmoid)
         else:
             s, n = x.shape[0], x.shape[1]
+            if x.shape[-1] == self.cin and x.shape[-2] == 120:
+                x = x.transpose(-1, -2)  # timelast input on the CPU path
             with torch.no_grad():
                 outs = []
                 for i in range(s):
