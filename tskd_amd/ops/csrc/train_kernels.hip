@@ -512,9 +512,10 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
     for (int i = threadIdx.x; i < 21; i += 256) lw2[i] = wpack[G::OW2 + i];
     __syncthreads();
     const int wave = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
-    float gw2[20];  // lane-local conv2 grads (summed over windows)
-    #pragma unroll
-    for (int i = 0; i < 20; ++i) gw2[i] = 0.f;
+    // lane-local conv2 grad: each lane owns exactly ONE (c,k) element — a
+    // scalar, NOT a lane-indexed array (runtime-indexed per-thread arrays
+    // spill to scratch memory).
+    float gw2_own = 0.f;
     float gb1l[4] = {0.f, 0.f, 0.f, 0.f};
     float gb2l = 0.f;
     for (int i = lane; i < 4 * G::CIN * G::K1; i += WAVE) gw1[wave][i] = 0.f;
@@ -557,7 +558,7 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
             float acc = 0.f;
             for (int s = 0; s < G::C2; ++s)
                 acc = fmaf(lda2[wave][s], lp1[wave][c * G::P1 + s + k], acc);
-            gw2[lane] += acc;
+            gw2_own += acc;
         }
         if (lane == 0) {
             float acc = 0.f;
@@ -617,7 +618,7 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
     // fold into global grads
     for (int i = lane; i < 4 * G::CIN * G::K1; i += WAVE)
         atomicAdd(&grads[G::OW1 + i], gw1[wave][i]);
-    if (lane < 20) atomicAdd(&grads[G::OW2 + lane], gw2[lane]);
+    if (lane < 20) atomicAdd(&grads[G::OW2 + lane], gw2_own);
     if (lane == 0) {
         #pragma unroll
         for (int c = 0; c < 4; ++c) atomicAdd(&grads[G::OB1 + c], gb1l[c]);
