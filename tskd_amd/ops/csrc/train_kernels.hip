@@ -336,10 +336,13 @@ __global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
     int S, int B)
 {
     constexpr int LIN = G::LIN;
-    __shared__ float lwih1[64 * LIN], lwhh1[64 * 16];
-    __shared__ float lwih2[64 * 16], lwhh2[64 * 16];
-    __shared__ float gwih1[64 * LIN], gwhh1[64 * 16], gb1[64];
-    __shared__ float gwih2[64 * 16], gwhh2[64 * 16], gb2[64];
+    // 16-wide rows padded to 17 dwords: a 16-dword stride puts the whole
+    // half-wave on 2 banks (16-way conflict); gcd(17, 32) = 1 spreads it.
+    constexpr int P16 = 17;
+    __shared__ float lwih1[64 * LIN], lwhh1[64 * P16];
+    __shared__ float lwih2[64 * P16], lwhh2[64 * P16];
+    __shared__ float gwih1[64 * LIN], gwhh1[64 * P16], gb1[64];
+    __shared__ float gwih2[64 * P16], gwhh2[64 * P16], gb2[64];
     __shared__ float gout[17];
     __shared__ float lda[64];  // per-step activated-gate grads (both layers)
 
@@ -350,10 +353,11 @@ __global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
         gwih1[i] = 0.f;
     }
     for (int i = lane; i < 64 * 16; i += WAVE) {
-        lwhh1[i] = wpack[G::OWHH1 + i];
-        lwih2[i] = wpack[G::OWIH2 + i];
-        lwhh2[i] = wpack[G::OWHH2 + i];
-        gwhh1[i] = gwih2[i] = gwhh2[i] = 0.f;
+        const int pi = (i / 16) * P16 + (i % 16);
+        lwhh1[pi] = wpack[G::OWHH1 + i];
+        lwih2[pi] = wpack[G::OWIH2 + i];
+        lwhh2[pi] = wpack[G::OWHH2 + i];
+        gwhh1[pi] = gwih2[pi] = gwhh2[pi] = 0.f;
     }
     if (lane < 17) gout[lane] = 0.f;
     gb1[lane] = gb2[lane] = 0.f;
@@ -397,8 +401,8 @@ __global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
             for (int j = 0; j < 16; ++j) {
                 const float h1j = sp[G::SL_H1 + j];
                 const float h2pj = spm ? spm[G::SL_H2 + j] : 0.f;
-                gwih2[lane * 16 + j] = fmaf(da2, h1j, gwih2[lane * 16 + j]);
-                gwhh2[lane * 16 + j] = fmaf(da2, h2pj, gwhh2[lane * 16 + j]);
+                gwih2[lane * P16 + j] = fmaf(da2, h1j, gwih2[lane * P16 + j]);
+                gwhh2[lane * P16 + j] = fmaf(da2, h2pj, gwhh2[lane * P16 + j]);
             }
             gb2[lane] += da2;
             if (lane < 16) gout[lane] = fmaf(dzt, sp[G::SL_H2 + lane],
@@ -413,8 +417,8 @@ __global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
                 #pragma unroll
                 for (int r = 0; r < 16; ++r) {
                     const float d = lda[k0 + r];
-                    p1 = fmaf(d, lwih2[(k0 + r) * 16 + unit], p1);
-                    p2 = fmaf(d, lwhh2[(k0 + r) * 16 + unit], p2);
+                    p1 = fmaf(d, lwih2[(k0 + r) * P16 + unit], p1);
+                    p2 = fmaf(d, lwhh2[(k0 + r) * P16 + unit], p2);
                 }
                 #pragma unroll
                 for (int off = 16; off < 64; off <<= 1) {
@@ -450,8 +454,8 @@ __global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
                 #pragma unroll
                 for (int j = 0; j < 16; ++j) {
                     const float h1pj = spm ? spm[G::SL_H1 + j] : 0.f;
-                    gwhh1[lane * 16 + j] = fmaf(da1, h1pj,
-                                                gwhh1[lane * 16 + j]);
+                    gwhh1[lane * P16 + j] = fmaf(da1, h1pj,
+                                                gwhh1[lane * P16 + j]);
                 }
                 gb1[lane] += da1;
                 dc1n = dc1 * f1;
@@ -460,7 +464,7 @@ __global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
                 float q1 = 0.f;
                 #pragma unroll
                 for (int r = 0; r < 16; ++r)
-                    q1 = fmaf(lda[k0 + r], lwhh1[(k0 + r) * 16 + unit], q1);
+                    q1 = fmaf(lda[k0 + r], lwhh1[(k0 + r) * P16 + unit], q1);
                 #pragma unroll
                 for (int off = 16; off < 64; off <<= 1)
                     q1 += __shfl_xor(q1, off);
@@ -481,9 +485,10 @@ __global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
     for (int i = lane; i < 64 * LIN; i += WAVE)
         atomicAdd(&grads[G::OWIH1 + i], gwih1[i]);
     for (int i = lane; i < 64 * 16; i += WAVE) {
-        atomicAdd(&grads[G::OWHH1 + i], gwhh1[i]);
-        atomicAdd(&grads[G::OWIH2 + i], gwih2[i]);
-        atomicAdd(&grads[G::OWHH2 + i], gwhh2[i]);
+        const int pi = (i / 16) * P16 + (i % 16);
+        atomicAdd(&grads[G::OWHH1 + i], gwhh1[pi]);
+        atomicAdd(&grads[G::OWIH2 + i], gwih2[pi]);
+        atomicAdd(&grads[G::OWHH2 + i], gwhh2[pi]);
     }
     atomicAdd(&grads[G::OBL1 + lane], gb1[lane]);
     atomicAdd(&grads[G::OBL2 + lane], gb2[lane]);
@@ -555,10 +560,17 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
         // conv2 grads: dW2[c][k] = sum_s da2[s] * p1[c][s+k]; db2 = sum da2
         if (lane < 20) {
             const int c = lane / 5, k = lane % 5;
-            float acc = 0.f;
-            for (int s = 0; s < G::C2; ++s)
-                acc = fmaf(lda2[wave][s], lp1[wave][c * G::P1 + s + k], acc);
-            gw2_own += acc;
+            const float* da = lda2[wave];
+            const float* pr = lp1[wave] + c * G::P1 + k;
+            float a0 = 0.f, a1 = 0.f, a2 = 0.f;
+            int s = 0;
+            for (; s + 3 <= G::C2; s += 3) {
+                a0 = fmaf(da[s + 0], pr[s + 0], a0);
+                a1 = fmaf(da[s + 1], pr[s + 1], a1);
+                a2 = fmaf(da[s + 2], pr[s + 2], a2);
+            }
+            for (; s < G::C2; ++s) a0 = fmaf(da[s], pr[s], a0);
+            gw2_own += a0 + a1 + a2;
         }
         if (lane == 0) {
             float acc = 0.f;
@@ -591,16 +603,25 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
             }
         }
         twsync();
-        // conv1 grads: dW1[c][i][k] = sum_s da1[c][s] * x[i][s+k]
+        // conv1 grads: dW1[c][i][k] = sum_s da1[c][s] * x[i][s+k].
+        // 4 independent partial accumulators: keeps 8 LDS loads in flight
+        // per iteration instead of exposing LDS latency every element.
         for (int o = lane; o < 4 * G::CIN * G::K1; o += WAVE) {
             const int c = o / (G::CIN * G::K1);
             const int i = (o / G::K1) % G::CIN;
             const int k = o % G::K1;
-            float acc = 0.f;
-            for (int s = 0; s < G::C1; ++s)
-                acc = fmaf(lda1[wave][c * G::C1 + s], xw[i * G::L + s + k],
-                           acc);
-            gw1[wave][o] += acc;
+            const float* da = lda1[wave] + c * G::C1;
+            const float* xr = xw + i * G::L + k;
+            float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+            int s = 0;
+            for (; s + 4 <= G::C1; s += 4) {
+                a0 = fmaf(da[s + 0], xr[s + 0], a0);
+                a1 = fmaf(da[s + 1], xr[s + 1], a1);
+                a2 = fmaf(da[s + 2], xr[s + 2], a2);
+                a3 = fmaf(da[s + 3], xr[s + 3], a3);
+            }
+            for (; s < G::C1; ++s) a0 = fmaf(da[s], xr[s], a0);
+            gw1[wave][o] += (a0 + a1) + (a2 + a3);
         }
         #pragma unroll
         for (int c = 0; c < 4; ++c) {
