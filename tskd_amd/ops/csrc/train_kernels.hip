@@ -547,16 +547,20 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
         }
         for (int sidx = lane; sidx < G::C2; sidx += WAVE) lda2[wave][sidx] = 0.f;
         twsync();
-        // pool2 scatter (serial on lane 0: LIN small, overlaps possible)
-        if (lane == 0) {
-            for (int q = 0; q < G::LIN; ++q) {
-                const int p = q * G::PS + i2[q];
-                const float d = dfw[q] * m2s[q];       // through dropout2
-                const float ct = c2t[p];
-                lda2[wave][p] += d * (1.f - ct * ct);  // through tanh'
-            }
+        // pool2 scatter: adjacent windows (stride 2, kernel 3) can collide
+        // only between NEIGHBORING q — even/odd q phases write disjoint
+        // ranges, so each phase is fully lane-parallel.
+        #pragma unroll
+        for (int phase = 0; phase < 2; ++phase) {
+            for (int q = lane; q < G::LIN; q += WAVE)
+                if ((q & 1) == phase) {
+                    const int p = q * G::PS + i2[q];
+                    const float d = dfw[q] * m2s[q];       // through dropout2
+                    const float ct = c2t[p];
+                    lda2[wave][p] += d * (1.f - ct * ct);  // through tanh'
+                }
+            twsync();
         }
-        twsync();
         // conv2 grads: dW2[c][k] = sum_s da2[s] * p1[c][s+k]; db2 = sum da2
         if (lane < 20) {
             const int c = lane / 5, k = lane % 5;
@@ -592,17 +596,21 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
         // zero da1 then pool1 scatter (serialized per channel: 4 lanes)
         for (int o = lane; o < 4 * G::C1; o += WAVE) lda1[wave][o] = 0.f;
         twsync();
-        if (lane < 4) {
-            const int c = lane;
-            for (int q = 0; q < G::P1; ++q) {
-                const int p = c * G::C1 + q * G::PS + i1[c * G::P1 + q];
-                const float ct = c1t[p];
-                lda1[wave][p] += ldp1[wave][c * G::P1 + q] *
-                                 m1s[c * G::P1 + q] *   // through dropout1
-                                 (1.f - ct * ct);
+        // pool1 scatter, same even/odd-phase parallelization per channel
+        #pragma unroll
+        for (int phase = 0; phase < 2; ++phase) {
+            for (int o = lane; o < 4 * G::P1; o += WAVE) {
+                const int c = o / G::P1, q = o % G::P1;
+                if ((q & 1) == phase) {
+                    const int p = c * G::C1 + q * G::PS + i1[o];
+                    const float ct = c1t[p];
+                    lda1[wave][p] += ldp1[wave][o] *
+                                     m1s[o] *           // through dropout1
+                                     (1.f - ct * ct);
+                }
             }
+            twsync();
         }
-        twsync();
         // conv1 grads: dW1[c][i][k] = sum_s da1[c][s] * x[i][s+k].
         // 4 independent partial accumulators: keeps 8 LDS loads in flight
         // per iteration instead of exposing LDS latency every element.
