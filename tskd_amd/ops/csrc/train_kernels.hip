@@ -514,6 +514,7 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
     __shared__ float lda2[4][G::C2];
     __shared__ float ldp1[4][4 * G::P1];
     __shared__ float gw1[4][4 * G::CIN * G::K1];  // per-wave conv1 w grads
+    __shared__ float gsml[4][25];  // per-wave w2/b grads for the block fold
     for (int i = threadIdx.x; i < 21; i += 256) lw2[i] = wpack[G::OW2 + i];
     __syncthreads();
     const int wave = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
@@ -644,14 +645,24 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
         }
         twsync();
     }
-    // fold into global grads
-    for (int i = lane; i < 4 * G::CIN * G::K1; i += WAVE)
-        atomicAdd(&grads[G::OW1 + i], gw1[wave][i]);
-    if (lane < 20) atomicAdd(&grads[G::OW2 + lane], gw2_own);
-    if (lane == 0) {
-        #pragma unroll
-        for (int c = 0; c < 4; ++c) atomicAdd(&grads[G::OB1 + c], gb1l[c]);
-        atomicAdd(&grads[G::OB2], gb2l);
+    // fold the block's 4 per-wave accumulators first, then ONE atomicAdd
+    // set per block: with tens of thousands of waves, per-wave atomics on
+    // the 425 grad addresses serialize per-address — the block fold cuts
+    // the atomic count 4x and the grid cap (launcher) another 4x.
+    if (lane < 20) gsml[wave][lane] = gw2_own;
+    else if (lane < 24) gsml[wave][lane] = gb1l[lane - 20];
+    else if (lane == 24) gsml[wave][lane] = gb2l;
+    __syncthreads();
+    for (int i = threadIdx.x; i < 4 * G::CIN * G::K1; i += 256)
+        atomicAdd(&grads[G::OW1 + i],
+                  gw1[0][i] + gw1[1][i] + gw1[2][i] + gw1[3][i]);
+    if (threadIdx.x < 25) {
+        const float v = gsml[0][threadIdx.x] + gsml[1][threadIdx.x] +
+                        gsml[2][threadIdx.x] + gsml[3][threadIdx.x];
+        if (threadIdx.x < 20) atomicAdd(&grads[G::OW2 + threadIdx.x], v);
+        else if (threadIdx.x < 24)
+            atomicAdd(&grads[G::OB1 + threadIdx.x - 20], v);
+        else atomicAdd(&grads[G::OB2], v);
     }
 }
 
@@ -715,7 +726,7 @@ template <class G>
 int conv_bwd(const float* x, const float* stash, const float* dfeat,
              const float* wpack, float* grads, int SN, hipStream_t s) {
     if (SN <= 0) return 0;
-    int grid = min((SN + 3) / 4, 8192);
+    int grid = min((SN + 3) / 4, 2048);
     hipLaunchKernelGGL((train_conv_bwd_kernel<G>), dim3(grid), dim3(256), 0,
                        s, x, stash, dfeat, wpack, grads, SN);
     return (int)hipGetLastError();
