@@ -650,8 +650,11 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
     // the 425 grad addresses serialize per-address — the block fold cuts
     // the atomic count 4x and the grid cap (launcher) another 4x.
     if (lane < 20) gsml[wave][lane] = gw2_own;
-    else if (lane < 24) gsml[wave][lane] = gb1l[lane - 20];
-    else if (lane == 24) gsml[wave][lane] = gb2l;
+    if (lane == 0) {  // gb1l/gb2l accumulate in lane 0's registers
+        #pragma unroll
+        for (int c = 0; c < 4; ++c) gsml[wave][20 + c] = gb1l[c];
+        gsml[wave][24] = gb2l;
+    }
     __syncthreads();
     for (int i = threadIdx.x; i < 4 * G::CIN * G::K1; i += 256)
         atomicAdd(&grads[G::OW1 + i],
