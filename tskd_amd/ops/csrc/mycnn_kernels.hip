@@ -194,16 +194,22 @@ __global__ __launch_bounds__(WG_THREADS) void conv_stack_kernel(
         wave_sync();
         for (int o = lane; o < 4 * G::C1; o += WAVE) {
             const int c = o / G::C1, s = o % G::C1;
-            float acc = b1[c];
+            // two independent partial accumulators (even/odd input channel)
+            // halve the 100-deep dependent fma chain
+            float acc = b1[c], acc2 = 0.f;
             const float* wr = w1 + c * (G::CIN * G::K1);
             #pragma unroll
-            for (int i = 0; i < G::CIN; ++i) {
+            for (int i = 0; i < G::CIN; i += 2) {
                 const float* xr = xw + i * G::L + s;
+                const float* xr2 = xw + (i + 1) * G::L + s;
                 #pragma unroll
-                for (int k = 0; k < G::K1; ++k)
+                for (int k = 0; k < G::K1; ++k) {
                     acc = fmaf(wr[i * G::K1 + k], xr[k], acc);
+                    if (i + 1 < G::CIN)
+                        acc2 = fmaf(wr[(i + 1) * G::K1 + k], xr2[k], acc2);
+                }
             }
-            lds_c1[wave][o] = tanhf_(acc);
+            lds_c1[wave][o] = tanhf_(acc + acc2);
         }
         wave_sync();
         conv_tail<G, false>(lane, lds_w, lds_c1[wave], lds_p1[wave],

@@ -146,14 +146,20 @@ __global__ __launch_bounds__(256) void train_conv_fwd_kernel(
         twsync();
         for (int o = lane; o < 4 * G::C1; o += WAVE) {
             const int c = o / G::C1, s = o % G::C1;
-            float acc = b1[c];
+            // two independent partial accumulators (even/odd input channel)
+            float acc = b1[c], acc2 = 0.f;
             const float* wr = w1 + c * (G::CIN * G::K1);
             #pragma unroll
-            for (int i = 0; i < G::CIN; ++i)
+            for (int i = 0; i < G::CIN; i += 2) {
                 #pragma unroll
-                for (int k = 0; k < G::K1; ++k)
+                for (int k = 0; k < G::K1; ++k) {
                     acc = fmaf(wr[i * G::K1 + k], xw[i * G::L + s + k], acc);
-            c1t[o] = tanh_(acc);
+                    if (i + 1 < G::CIN)
+                        acc2 = fmaf(wr[(i + 1) * G::K1 + k],
+                                    xw[(i + 1) * G::L + s + k], acc2);
+                }
+            }
+            c1t[o] = tanh_(acc + acc2);
         }
         twsync();
         for (int o = lane; o < 4 * G::P1; o += WAVE) {
