@@ -215,3 +215,75 @@ class TestEndToEndGPU:
         for k in out["cpu"]:
             np.testing.assert_allclose(out["cpu"][k], out["cuda"][k],
                                        rtol=1e-4, atol=1e-5)
+
+
+class TestFusedServe:
+    def test_fused_server_cpu(self, tmp_path, cfg):
+        """serve.py: bus -> fused ingest+preprocess+infer -> store, one
+        process (the production arrangement of processStream+predictStream)."""
+        from tskd_amd.cli.serve import FusedServer
+        bus = Bus(str(tmp_path / "bus"))
+        _write_wfdb_record(cfg.wavef_path, "p000194-test", cfg.channel_names,
+                           1 / 60, 40)
+        store = PredictionStore(str(tmp_path / "pred.log"))
+        ages = AgeTable()
+        ages.set("p000194", 47.0)
+        srv = FusedServer(bus, cfg, store, ages, device="cpu",
+                          max_streams=8, ring_grid=1024, starting="earliest")
+        send_record_data(bus, "p000194-test", None, 1e6, 1.0, cfg)
+        n = srv.trigger()
+        assert n == 1
+        assert store.count() == 1
+        _, risk = store.latest("p000194")
+        assert 0.0 <= risk <= 1.0
+        # second trigger with no new data -> no duplicate prediction
+        assert srv.trigger() == 0
+        assert store.count() == 1
+
+    def test_fused_matches_two_stage(self, tmp_path, cfg):
+        """The fused daemon's risk score == the two-stage pipeline's score
+        for the same record (same windows, same model weights)."""
+        import torch as T
+        from tskd_amd.cli.serve import FusedServer
+        from tskd_amd.models import build_model
+        T.manual_seed(77)
+        model = build_model("MyCNN5").eval()
+        _write_wfdb_record(cfg.wavef_path, "p000194-test", cfg.channel_names,
+                           1 / 60, 40, seed=9)
+
+        # two-stage
+        bus1 = Bus(str(tmp_path / "bus1"))
+        send_record_data(bus1, "p000194-test", None, 1e6, 1.0, cfg)
+        ps = ProcessStream(bus1, cfg, max_streams=4, starting="earliest")
+        ps.trigger()
+        st1 = PredictionStore(str(tmp_path / "p1.log"))
+        pr = PredictStream(bus1, cfg, st1, model=model, device="cpu",
+                           starting="earliest")
+        pr.trigger()
+        _, risk_two_stage = st1.latest("p000194")
+
+        # fused
+        bus2 = Bus(str(tmp_path / "bus2"))
+        send_record_data(bus2, "p000194-test", None, 1e6, 1.0, cfg)
+        st2 = PredictionStore(str(tmp_path / "p2.log"))
+        srv = FusedServer(bus2, cfg, st2, model=model, device="cpu",
+                          max_streams=8, ring_grid=1024, starting="earliest")
+        srv.trigger()
+        _, risk_fused = st2.latest("p000194")
+        assert abs(risk_fused - risk_two_stage) < 1e-4
+
+    @pytest.mark.gpu
+    def test_fused_server_gpu(self, tmp_path, cfg):
+        from tskd_amd.cli.serve import FusedServer
+        bus = Bus(str(tmp_path / "bus"))
+        _write_wfdb_record(cfg.wavef_path, "p000194-test", cfg.channel_names,
+                           1 / 60, 40)
+        store = PredictionStore(str(tmp_path / "pred.log"))
+        srv = FusedServer(bus, cfg, store, device="cuda", max_streams=8,
+                          ring_grid=1024, starting="earliest")
+        send_record_data(bus, "p000194-test", None, 1e6, 1.0, cfg)
+        n = srv.trigger()
+        torch.cuda.synchronize()
+        assert n == 1 and store.count() == 1
+        _, risk = store.latest("p000194")
+        assert 0.0 <= risk <= 1.0

@@ -1,0 +1,208 @@
+"""Fused serving daemon — the production deployment mode.
+
+Collapses processStream + predictStream into ONE process per GPU: raw
+channel topics are drained from the bus (native wire parse), ingested into
+the GPU StreamEngine ring buffers, and every trigger runs the fused
+preprocess -> window-gather -> MFMA conv -> LSTM -> sigmoid path; risk
+scores land in the prediction store (and optionally a response topic).
+The two-stage CLIs (processstream | predictstream) remain available for the
+reference's exact topology including the `call-stream` wire contract; this
+daemon is the latency/throughput-optimal arrangement.
+
+Scale-out: under torchrun (one rank per GPU), patient streams are sharded
+by consistent hash (tskd_amd.parallel.shard_for_key) — each rank ignores
+messages outside its shard — and per-trigger predictions are all-gathered
+over RCCL/xGMI; rank 0 writes the store.
+
+Usage:
+    python -m tskd_amd.cli.serve --bus-dir ... --store-path predictions.log
+    python -m torch.distributed.run --nproc-per-node 8 --master-addr \
+        127.0.0.1 -m tskd_amd.cli.serve ...
+"""
+
+from __future__ import annotations
+
+import argparse
+import logging
+import os
+import signal
+import time
+from typing import Dict, Optional
+
+import numpy as np
+import torch
+
+from tskd_amd.bus import Bus, Consumer, Producer
+from tskd_amd.config import get_global_config
+from tskd_amd.engine import StreamEngine
+from tskd_amd.metrics import StageTimer
+from tskd_amd.models import build_model
+from tskd_amd.ops import MyCNNEngine
+from tskd_amd.parallel import (all_gather_predictions, init_distributed,
+                               shard_for_key)
+from tskd_amd.store import AgeTable, PredictionStore
+
+log = logging.getLogger("serve")
+
+
+class FusedServer:
+    def __init__(self, bus: Bus, cfg, store: Optional[PredictionStore],
+                 ages: Optional[AgeTable] = None, model=None,
+                 device: str = "cpu", max_streams: int = 1024,
+                 ring_grid: int = 4096, starting: str = "latest",
+                 response_topic: Optional[str] = None,
+                 rank: int = 0, world: int = 1):
+        self.cfg = cfg
+        self.store = store
+        self.ages = ages or AgeTable()
+        self.rank, self.world = rank, world
+        self.device = device
+        self.me = MyCNNEngine(model or build_model("MyCNN5").eval(),
+                              device=device)
+        # ring channel space = the MODEL's wire space (config may name fewer
+        # channels; the rest stay zero, like the reference's fixed (1,10,120))
+        self.se = StreamEngine(max_streams, max(cfg.n_channels, self.me.cin),
+                               ring_grid=ring_grid, device=device)
+        self.consumer = Consumer(bus, starting=starting)
+        topics = [cfg.topic_for_channel(c) for c in cfg.channel_names]
+        for t in topics:
+            bus.create_topic(t)
+        self.consumer.subscribe(topics)
+        self.producer = Producer(bus) if response_topic else None
+        self.response_topic = response_topic
+        if response_topic:
+            bus.create_topic(response_topic)
+        self.pid_index: Dict[str, int] = {}
+        self.pids: list = []
+        self.max_streams = max_streams
+        self.hwm = 0.0
+        self.watermark_s = cfg.watermark_s
+        self.timer = StageTimer("serve")
+        self.n_predictions = 0
+
+    def _sid(self, pid: str) -> Optional[int]:
+        if self.world > 1 and shard_for_key(pid, self.world) != self.rank:
+            return None  # another rank's patient
+        if pid not in self.pid_index:
+            if len(self.pid_index) >= self.max_streams:
+                raise RuntimeError("max_streams exceeded")
+            self.pid_index[pid] = len(self.pid_index)
+            self.pids.append(pid)
+        return self.pid_index[pid]
+
+    def trigger(self) -> int:
+        """Drain bus -> ingest -> fused preprocess+infer -> store."""
+        with self.timer:
+            keys, _t, chans, vals, ts = self.consumer.poll_samples(
+                max_msgs=131072, timeout_ms=0)
+            si, ci, tt, vv = [], [], [], []
+            for i, k in enumerate(keys):
+                sid = self._sid(k)
+                if sid is None:
+                    continue
+                si.append(sid)
+                ci.append(int(chans[i]))
+                tt.append(float(ts[i]))
+                vv.append(float(vals[i]))
+            if len(ts):
+                self.hwm = max(self.hwm, float(ts.max()))
+            advance = max(0.0, self.hwm - self.watermark_s)
+            nproc_before = self.se.nproc
+            if si:
+                self.se.ingest_events(
+                    torch.tensor(si, dtype=torch.long),
+                    torch.tensor(ci, dtype=torch.long),
+                    torch.tensor(tt, dtype=torch.float64),
+                    torch.tensor(vv, dtype=torch.float32),
+                    advance_to=advance)
+            elif advance / self.se.bucket_s > self.se.head:
+                self.se._clear_ahead(int(advance / self.se.bucket_s))
+                self.se.head = int(advance / self.se.bucket_s)
+                self.se._refill()
+            if self.se.nproc == nproc_before or not self.se.ready \
+                    or not self.pids:
+                return 0
+            dtype = torch.bfloat16 if self.device != "cpu" else torch.float32
+            w = self.se.windows(batch=1, stride=12, dtype=dtype,
+                                timelast=self.device != "cpu")
+            n_active = len(self.pids)
+            age = torch.tensor([[self.ages.get(p)] for p in self.pids],
+                               device=w.device)
+            probs_all = self.me.forward(w[:n_active].contiguous()
+                                        if n_active < self.se.S else w,
+                                        age, apply_sigmoid=True)
+            probs = probs_all.reshape(-1)[:n_active]
+            t_us = int(self.hwm * 1e6)
+            if self.world > 1:
+                pad = torch.zeros(self.max_streams, device=probs.device)
+                pad[:n_active] = probs
+                gathered = all_gather_predictions(pad).cpu()
+            else:
+                gathered = None
+            if self.store is not None and self.rank == 0:
+                local = probs.cpu()
+                self.store.insert_batch(self.pids,
+                                        [t_us] * n_active,
+                                        local.tolist())
+                self.n_predictions += n_active
+            if self.producer:
+                for i, pid in enumerate(self.pids):
+                    self.producer.produce(
+                        self.response_topic, pid,
+                        f'{{"t_us": {t_us}, "risk": {float(probs[i]):.6f}}}')
+            self.timer.add_items(n_active)
+            return n_active
+
+
+def main(argv=None) -> None:
+    logging.basicConfig(level=logging.INFO,
+                        format="%(asctime)s %(name)s %(levelname)s %(message)s")
+    cfg = get_global_config()
+    ap = argparse.ArgumentParser(description=__doc__)
+    ap.add_argument("--speed", type=float, default=5.0)
+    ap.add_argument("--bus-dir", default=None)
+    ap.add_argument("--store-path", default="predictions.log")
+    ap.add_argument("--model-path", default=cfg.model_path)
+    ap.add_argument("--age-table", default=None)
+    ap.add_argument("--device", default="cuda" if torch.cuda.is_available()
+                    else "cpu")
+    ap.add_argument("--max-streams", type=int, default=1024)
+    ap.add_argument("--starting", default="latest",
+                    choices=["latest", "earliest"])
+    ap.add_argument("--model-response-topic", default=None)
+    ap.add_argument("--max-triggers", type=int, default=0)
+    args = ap.parse_args(argv)
+
+    rank, world = init_distributed()
+    from tskd_amd.cli.predictstream import load_model_for_serving
+    bus = Bus(args.bus_dir)
+    store = PredictionStore(args.store_path) if rank == 0 else None
+    ages = AgeTable()
+    if args.age_table and os.path.exists(args.age_table):
+        try:
+            ages.load_cohort_csv(args.age_table)
+        except (ValueError, IndexError):
+            ages.load(args.age_table)
+    srv = FusedServer(bus, cfg, store, ages,
+                      model=load_model_for_serving(args.model_path),
+                      device=args.device, max_streams=args.max_streams,
+                      starting=args.starting,
+                      response_topic=args.model_response_topic,
+                      rank=rank, world=world)
+    period = cfg.predict_slide_s / args.speed
+    stop = []
+    signal.signal(signal.SIGTERM, lambda *a: stop.append(1))
+    n = 0
+    while not stop:
+        t0 = time.time()
+        srv.trigger()
+        n += 1
+        if n % 10 == 0:
+            log.info("metrics %s", srv.timer.log_line())
+        if args.max_triggers and n >= args.max_triggers:
+            break
+        time.sleep(max(0.0, period - (time.time() - t0)))
+
+
+if __name__ == "__main__":
+    main()
