@@ -29,7 +29,6 @@ import signal
 import time
 from typing import Dict, Optional
 
-import numpy as np
 import torch
 
 from tskd_amd.bus import Bus, Consumer, Producer
@@ -134,12 +133,14 @@ class FusedServer:
             probs = probs_all.reshape(-1)[:n_active]
             t_us = int(self.hwm * 1e6)
             if self.world > 1:
+                # every rank sees the full per-trigger prediction set
+                # (monitoring/router hook) — tiny direct all-gather on xGMI
                 pad = torch.zeros(self.max_streams, device=probs.device)
                 pad[:n_active] = probs
-                gathered = all_gather_predictions(pad).cpu()
-            else:
-                gathered = None
-            if self.store is not None and self.rank == 0:
+                self.last_gathered = all_gather_predictions(pad)
+            if self.store is not None:
+                # the mmap store is multi-process safe: each rank persists
+                # its own shard directly (no gather needed for durability)
                 local = probs.cpu()
                 self.store.insert_batch(self.pids,
                                         [t_us] * n_active,
@@ -176,7 +177,7 @@ def main(argv=None) -> None:
     rank, world = init_distributed()
     from tskd_amd.cli.predictstream import load_model_for_serving
     bus = Bus(args.bus_dir)
-    store = PredictionStore(args.store_path) if rank == 0 else None
+    store = PredictionStore(args.store_path)  # multi-process safe: all ranks
     ages = AgeTable()
     if args.age_table and os.path.exists(args.age_table):
         try:
