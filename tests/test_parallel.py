@@ -171,3 +171,72 @@ class TestCliOffsetFlow:
                             timeout=120)
         assert r2.returncode == 0, r2.stderr[-2000:]
         assert load_offsets(off) == pos1  # nothing new to consume
+
+
+def _fused_dp_worker(rank, world, port, tmpdir, wavef):
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "RANK": str(rank), "WORLD_SIZE": str(world),
+    })
+    import torch.distributed as dist
+
+    from tskd_amd.bus import Bus
+    from tskd_amd.cli.serve import FusedServer
+    from tskd_amd.cli.sendstream import send_record_data
+    from tskd_amd.config import GlobalConfig
+    from tskd_amd.models import build_model
+    from tskd_amd.store import PredictionStore
+    dist.init_process_group("gloo")
+    cfg = GlobalConfig()
+    cfg.wavef_path = wavef
+    cfg.channel_names = ["HR", "RESP", "PULSE", "SpO2"]
+    torch.manual_seed(0)
+    model = build_model("MyCNN5").eval()
+    bus = Bus(os.path.join(tmpdir, f"bus_{rank}"))
+    store = PredictionStore(os.path.join(tmpdir, "pred.log"))  # SHARED
+    srv = FusedServer(bus, cfg, store, model=model, device="cpu",
+                      max_streams=8, ring_grid=1024, starting="earliest",
+                      rank=rank, world=world)
+    # every rank sees every patient's messages; sharding filters them
+    for rec in ("p000194-test", "p000007-test", "p000021-test",
+                "p000042-test"):
+        send_record_data(bus, rec, None, 1e6, 1.0, cfg)
+    n = srv.trigger()
+    dist.barrier()
+    with open(os.path.join(tmpdir, f"rank{rank}.n"), "w") as f:
+        f.write(str(n))
+    dist.destroy_process_group()
+
+
+class TestFusedServeDP:
+    def test_sharded_serving_world2(self, tmp_path):
+        """Two fused-server ranks over the same 4 patients: disjoint shards,
+        one shared store, complete coverage."""
+        import numpy as np
+
+        from tskd_amd.parallel import shard_for_key
+        from tskd_amd.store import PredictionStore
+        sys.path.insert(0, REPO)
+        from tests.test_pipeline import _write_wfdb_record
+        wavef = str(tmp_path / "wavef")
+        pids = ["p000194", "p000007", "p000021", "p000042"]
+        for pid in pids:
+            _write_wfdb_record(wavef, f"{pid}-test",
+                               ["HR", "RESP", "PULSE", "SpO2"], 1 / 60, 40)
+        ctx = mp.get_context("spawn")
+        ps = [ctx.Process(target=_fused_dp_worker,
+                          args=(r, 2, 29537, str(tmp_path), wavef))
+              for r in range(2)]
+        for p in ps:
+            p.start()
+        for p in ps:
+            p.join(180)
+            assert p.exitcode == 0
+        counts = [int(open(tmp_path / f"rank{r}.n").read()) for r in range(2)]
+        expect = [len([p_ for p_ in pids if shard_for_key(p_, 2) == r])
+                  for r in range(2)]
+        assert counts == expect and sum(counts) == 4
+        st = PredictionStore(str(tmp_path / "pred.log"))
+        assert st.count() == 4  # both ranks persisted into ONE store
+        got = {r[0] for r in st.tail(10)}
+        assert got == set(pids)
