@@ -167,7 +167,7 @@ def main() -> None:
                 from tskd_amd.engine.stream_engine import TriggerGraph
                 from tskd_amd.ops import GraphedForward
                 graphed = GraphedForward(eng, s=S, n=1, dtype=dtype,
-                                         timelast=True)
+                                         timelast=True, capture=False)
                 while se.nproc < se.head - se.win_buckets + 1 or se.nproc == 0:
                     se.ingest_dense(raw, chan_map=chan_map)
                 torch.cuda.synchronize()

@@ -232,13 +232,16 @@ class GraphedForward:
     def __init__(self, engine: "MyCNNEngine", s: int, n: int,
                  dtype: torch.dtype = torch.bfloat16,
                  apply_sigmoid: bool = True, warmup: int = 2,
-                 timelast: bool = False):
+                 timelast: bool = False, capture: bool = True):
         assert engine.device.type == "cuda"
         self.engine = engine
         self.timelast = timelast
         self.x = alloc_windows(s, n, engine.cin, 120, timelast=timelast,
                                dtype=dtype, device=engine.device)
         self.age = torch.full((s, n), 65.0, device=engine.device)
+        self.graph = None
+        if not capture:
+            return  # buffers only (e.g. inside a TriggerGraph capture)
         stream = torch.cuda.Stream()
         with torch.cuda.stream(stream):
             for _ in range(warmup):

@@ -29,6 +29,7 @@
 //   train_adam       — fused Adam with bias correction + grad zeroing.
 
 #include <hip/hip_runtime.h>
+#include <stdlib.h>
 
 #define WAVE 64
 
@@ -735,7 +736,9 @@ template <class G>
 int conv_bwd(const float* x, const float* stash, const float* dfeat,
              const float* wpack, float* grads, int SN, hipStream_t s) {
     if (SN <= 0) return 0;
-    int grid = min((SN + 3) / 4, 2048);
+    int cap = 2048;
+    if (const char* e = getenv("TSKD_CONVBWD_GRID")) cap = atoi(e);
+    int grid = min((SN + 3) / 4, cap);
     hipLaunchKernelGGL((train_conv_bwd_kernel<G>), dim3(grid), dim3(256), 0,
                        s, x, stash, dfeat, wpack, grads, SN);
     return (int)hipGetLastError();
