@@ -34,6 +34,7 @@
 // Wavefront size is 64 on CDNA4 (not 32) — all lane math assumes it.
 
 #include <hip/hip_runtime.h>
+#include <stdlib.h>
 
 #define WAVE 64
 #define WG_WAVES 4
@@ -643,9 +644,16 @@ int launch_lstm(const float* feat, const float* age, float* out,
     if (S <= 0 || N <= 0) return 0;
     int grid = (S + WG_WAVES - 1) / WG_WAVES;
     if (grid > 8192) grid = 8192;
-    hipLaunchKernelGGL((lstm_head_kernel<G>), dim3(grid), dim3(WG_THREADS), 0,
-                       stream, feat, age, out, wpack, S, N, age_eps,
-                       apply_sigmoid);
+    int chunk = 32;
+    if (const char* e = getenv("TSKD_LSTM_CHUNK")) chunk = atoi(e);
+    if (chunk >= 64)
+        hipLaunchKernelGGL((lstm_head_kernel<G, 64>), dim3(grid),
+                           dim3(WG_THREADS), 0, stream, feat, age, out,
+                           wpack, S, N, age_eps, apply_sigmoid);
+    else
+        hipLaunchKernelGGL((lstm_head_kernel<G, 32>), dim3(grid),
+                           dim3(WG_THREADS), 0, stream, feat, age, out,
+                           wpack, S, N, age_eps, apply_sigmoid);
     return (int)hipGetLastError();
 }
 
