@@ -340,3 +340,28 @@ class TestTriggerGraph:
             torch.cuda.synchronize()
             torch.testing.assert_close(out, eager[idx], rtol=2e-3, atol=2e-3)
             idx += 1
+
+
+class TestZNormalize:
+    def test_cpu_znorm(self):
+        eng = _mk_engine(2, 3, 25.0)
+        eng.ingest_dense(_dense_raw(2, 3, int(25 * 60 * 20), seed=9))
+        w = eng.windows(batch=2, stride=12, znormalize=True)
+        m = w.mean(dim=-1)
+        sd = w.std(dim=-1, unbiased=False)
+        assert m.abs().max() < 1e-4
+        np.testing.assert_allclose(sd.numpy(), 1.0, atol=1e-3)
+
+    @pytest.mark.gpu
+    def test_gpu_matches_cpu_znorm(self):
+        S, C, fs = 2, 4, 125.0
+        raw = _dense_raw(S, C, int(fs * 60 * 16), seed=10)
+        cpu = StreamEngine(S, C, ring_grid=1024, fs=fs, device="cpu")
+        gpu = StreamEngine(S, C, ring_grid=1024, fs=fs, device="cuda")
+        cpu.ingest_dense(raw)
+        gpu.ingest_dense(raw.cuda())
+        wc = cpu.windows(batch=2, stride=12, znormalize=True)
+        wg = gpu.windows(batch=2, stride=12, znormalize=True)
+        torch.cuda.synchronize()
+        np.testing.assert_allclose(wg.cpu().numpy(), wc.numpy(), rtol=1e-3,
+                                   atol=1e-4)

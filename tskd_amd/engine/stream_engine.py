@@ -68,6 +68,11 @@ def _load_preproc_lib() -> ctypes.CDLL:
         ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_long,
         ctypes.c_int, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p,
     ]
+    lib.tskd_preproc_window_znorm.restype = ctypes.c_int
+    lib.tskd_preproc_window_znorm.argtypes = [
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_long, ctypes.c_int,
+        ctypes.c_float, ctypes.c_void_p,
+    ]
     lib.tskd_preproc_window_gather.restype = ctypes.c_int
     lib.tskd_preproc_window_gather.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
@@ -334,7 +339,8 @@ class StreamEngine:
     def windows(self, batch: int = 1, stride: int = 12,
                 dtype: torch.dtype = torch.float32,
                 out: Optional[torch.Tensor] = None,
-                timelast: bool = False) -> torch.Tensor:
+                timelast: bool = False,
+                znormalize: bool = False, zeps: float = 1e-6) -> torch.Tensor:
         """Assemble (S, batch, C, model_win) model inputs ending at the latest
         processed grid point; window b ends at nproc - (batch-1-b)*stride.
         Early windows that would reach before grid 0 are all-zero.
@@ -358,6 +364,14 @@ class StreamEngine:
                 self.S, self.C, self.G, B, WIN, stride,
                 ctypes.c_long(self.nproc), self._dstate_ptr(),
                 self._dstate_gather_extra, _sptr())
+            if rc == 0 and znormalize:
+                # opt-in z-normalize per (stream, batch, channel) window row
+                # (NOT reference semantics — see SURVEY.md §7 fidelity note).
+                assert not timelast, "znormalize needs channel-major windows"
+                rc = lib.tskd_preproc_window_znorm(
+                    ctypes.c_void_p(out.data_ptr()), is_bf16,
+                    ctypes.c_long(self.S * B * self.C), WIN,
+                    ctypes.c_float(zeps), _sptr())
             if rc != 0:
                 raise RuntimeError(f"window_gather failed: hipError {rc}")
         else:
@@ -369,6 +383,11 @@ class StreamEngine:
                 idx = (np.arange(wend - WIN, wend)) % self.G
                 w = torch.from_numpy(pr[:, :, idx]).to(dtype)
                 out[:, b] = w.transpose(-1, -2) if timelast else w
+            if znormalize:
+                assert not timelast
+                m = out.float().mean(dim=-1, keepdim=True)
+                sd = out.float().std(dim=-1, unbiased=False, keepdim=True)
+                out.copy_(((out.float() - m) / sd.clamp_min(zeps)).to(dtype))
         return out
 
     @property

@@ -359,6 +359,48 @@ __global__ void window_gather_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Optional per-window z-normalization (BASELINE.json names a z-normalize
+// stage; the REFERENCE has none — SURVEY.md §7 fidelity note — so this is
+// an opt-in post-pass, off by default): each (stream, batch, channel) row of
+// WIN points is normalized to (v - mean) / max(std, eps) in place.
+// One wave per row: shuffle-reduced mean/var, vectorized rewrite.
+// ---------------------------------------------------------------------------
+template <class OT>
+__global__ void window_znorm_kernel(OT* __restrict__ w,  // (rows, WIN)
+                                    long rows, int WIN, float eps)
+{
+    const int lane = threadIdx.x % WAVE;
+    for (long r = (long)blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+         r < rows; r += (long)gridDim.x * (blockDim.x / WAVE)) {
+        OT* row = w + r * WIN;
+        float sum = 0.f, sq = 0.f;
+        for (int t = lane; t < WIN; t += WAVE) {
+            float v;
+            if constexpr (sizeof(OT) == 2) v = bf16_to_f32_((unsigned short)row[t]);
+            else v = (float)row[t];
+            sum += v;
+            sq = fmaf(v, v, sq);
+        }
+        #pragma unroll
+        for (int off = 32; off > 0; off >>= 1) {
+            sum += __shfl_xor(sum, off);
+            sq += __shfl_xor(sq, off);
+        }
+        const float mean = sum / WIN;
+        const float var = fmaxf(sq / WIN - mean * mean, 0.f);
+        const float inv = 1.0f / fmaxf(sqrtf(var), eps);
+        for (int t = lane; t < WIN; t += WAVE) {
+            float v;
+            if constexpr (sizeof(OT) == 2) v = bf16_to_f32_((unsigned short)row[t]);
+            else v = (float)row[t];
+            v = (v - mean) * inv;
+            if constexpr (sizeof(OT) == 2) row[t] = (OT)f32_to_bf16_(v);
+            else row[t] = (OT)v;
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // extern "C" launchers
 // ---------------------------------------------------------------------------
 static inline int grid_for(long n, int block) {
@@ -464,6 +506,20 @@ int tskd_preproc_window_gather(const float* proc, void* out, int out_is_bf16,
                                (float*)out, S, C, G, B, WIN, stride, end,
                                dstate, end_extra);
     }
+    return (int)hipGetLastError();
+}
+
+int tskd_preproc_window_znorm(void* w, int is_bf16, long rows, int win,
+                              float eps, void* stream) {
+    if (rows <= 0) return 0;
+    const int grid = grid_for(rows * WAVE, 256);
+    if (is_bf16)
+        hipLaunchKernelGGL((window_znorm_kernel<unsigned short>), dim3(grid),
+                           dim3(256), 0, (hipStream_t)stream,
+                           (unsigned short*)w, rows, win, eps);
+    else
+        hipLaunchKernelGGL((window_znorm_kernel<float>), dim3(grid), dim3(256),
+                           0, (hipStream_t)stream, (float*)w, rows, win, eps);
     return (int)hipGetLastError();
 }
 
