@@ -22,6 +22,7 @@
 #include <cstring>
 #include <map>
 #include <mutex>
+#include <set>
 #include <stdexcept>
 #include <string>
 #include <vector>
@@ -309,6 +310,7 @@ public:
         for (int attempt = 0;; ++attempt) {
             try {
                 bus_->part(topic, p)->append(key, value, ts_us);
+                touched_.insert({topic, p});
                 ++produced_;
                 return;
             } catch (const std::exception&) {
@@ -320,10 +322,15 @@ public:
 
     void flush(const std::string& topic = "") {
         // acks=all analog: force page-cache sync of produced partitions.
-        for (auto& t : topic.empty() ? bus_->list_topics()
-                                     : std::vector<std::string>{topic})
-            for (int p = 0; p < bus_->topic_nparts(t); ++p)
-                bus_->part(t, p)->sync();
+        // Without a topic, only the partitions THIS producer touched are
+        // synced (the reference flushes per sample row — sendStream.py:71 —
+        // so flush must not rescan the whole bus directory every call).
+        if (topic.empty()) {
+            for (auto& [t, p] : touched_) bus_->part(t, p)->sync();
+            return;
+        }
+        for (int p = 0; p < bus_->topic_nparts(topic); ++p)
+            bus_->part(topic, p)->sync();
     }
 
     uint64_t produced() const { return produced_; }
@@ -332,6 +339,7 @@ private:
     std::shared_ptr<Bus> bus_;
     int retries_;
     uint64_t produced_ = 0;
+    std::set<std::pair<std::string, int>> touched_;
 };
 
 struct Message {
