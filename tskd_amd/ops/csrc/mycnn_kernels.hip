@@ -434,6 +434,110 @@ __global__ __launch_bounds__(WG_THREADS) void conv_stack_mfma_tlast_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Kernel 1d (bf16 TIME-LAST, workgroup-per-window): all 4 waves of a WG
+// cooperate on ONE window — waves split the 7 conv1 M-tiles, the pooled
+// tail runs WG-wide — dividing the per-window serial-latency chain ~4x at
+// the same instruction count. Tiny LDS (one window's intermediates) keeps
+// occupancy at the wave cap.
+// ---------------------------------------------------------------------------
+template <class G>
+__global__ __launch_bounds__(WG_THREADS) void conv_stack_mfma_wg_kernel(
+    const unsigned short* __restrict__ x,   // (SN, L, CIN) bf16 bits
+    float* __restrict__ feat,               // (SN, LIN)
+    const float* __restrict__ wpack,
+    const unsigned short* __restrict__ bfrag,
+    int SN)
+{
+    constexpr int NW = 4 * G::CIN * G::K1 + 4 + 20 + 1;
+    __shared__ float lds_w[NW];
+    __shared__ float lds_c1[4 * G::C1];
+    __shared__ float lds_p1[4 * G::P1];
+    __shared__ float lds_c2[G::C2];
+
+    for (int i = threadIdx.x; i < NW; i += WG_THREADS) lds_w[i] = wpack[i];
+    __syncthreads();
+
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const float* b1 = lds_w + G::OB1;
+    const float* w2 = lds_w + G::OW2;
+    const float b2 = lds_w[G::OB2];
+
+    bf16x8 bfr[G::KSTEPS];
+    union BU { unsigned int d[4]; unsigned short u[8]; bf16x8 v; };
+    #pragma unroll
+    for (int st = 0; st < G::KSTEPS; ++st) {
+        BU bu;
+        #pragma unroll
+        for (int q = 0; q < 4; ++q)
+            bu.d[q] = ((const unsigned int*)bfrag)[(st * WAVE + lane) * 4 + q];
+        bfr[st] = bu.v;
+    }
+    const int lane_dw = ((lane & 15) * G::CIN + (lane >> 4) * 8) / 2;
+
+    for (long win = blockIdx.x; win < SN; win += gridDim.x) {
+        const unsigned int* xw =
+            (const unsigned int*)(x + win * (long)(G::L * G::CIN));
+        // waves split the M-tiles
+        for (int mt = wave; mt < G::MTILES; mt += WG_WAVES) {
+            f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+            const unsigned int* base = xw + (long)mt * (16 * G::CIN / 2)
+                                       + lane_dw;
+            #pragma unroll
+            for (int st = 0; st < G::KSTEPS; ++st) {
+                BU au;
+                au.d[0] = base[st * 16 + 0];
+                au.d[1] = base[st * 16 + 1];
+                au.d[2] = base[st * 16 + 2];
+                au.d[3] = base[st * 16 + 3];
+                acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(au.v, bfr[st],
+                                                              acc, 0, 0, 0);
+            }
+            const int c = lane & 15;
+            if (c < 4) {
+                #pragma unroll
+                for (int r = 0; r < 4; ++r) {
+                    const int row = mt * 16 + (lane >> 4) * 4 + r;
+                    if (row < G::C1)
+                        lds_c1[c * G::C1 + row] = acc[r] + b1[c];
+                }
+            }
+        }
+        __syncthreads();
+        // WG-wide pooled tail (tanh after pool — monotonicity)
+        for (int o = threadIdx.x; o < 4 * G::P1; o += WG_THREADS) {
+            const int c = o / G::P1, q = o % G::P1;
+            const float* src = lds_c1 + c * G::C1 + q * G::PS;
+            float m = src[0];
+            #pragma unroll
+            for (int k = 1; k < G::PK; ++k) m = fmaxf(m, src[k]);
+            lds_p1[o] = tanhf_(m);
+        }
+        __syncthreads();
+        for (int s = threadIdx.x; s < G::C2; s += WG_THREADS) {
+            float acc = b2;
+            #pragma unroll
+            for (int c = 0; c < 4; ++c) {
+                const float* pr = lds_p1 + c * G::P1 + s;
+                #pragma unroll
+                for (int k = 0; k < 5; ++k)
+                    acc = fmaf(w2[c * 5 + k], pr[k], acc);
+            }
+            lds_c2[s] = acc;
+        }
+        __syncthreads();
+        for (int q = threadIdx.x; q < G::LIN; q += WG_THREADS) {
+            const float* src = lds_c2 + q * G::PS;
+            float m = src[0];
+            #pragma unroll
+            for (int k = 1; k < G::PK; ++k) m = fmaxf(m, src[k]);
+            feat[win * G::LIN + q] = tanhf_(m);
+        }
+        __syncthreads();
+    }
+}
+
+// ---------------------------------------------------------------------------
 // Kernel 2: fused LSTM (batch-as-time) + Linear head + age gate (+sigmoid)
 // (K6+K7+K8+K9 of SURVEY.md §2.6)
 // ---------------------------------------------------------------------------
