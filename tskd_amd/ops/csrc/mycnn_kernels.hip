@@ -720,10 +720,26 @@ int launch_conv(const void* x, int x_is_bf16, int x_timelast, float* feat,
         if (!bfrag) return -2;  // bf16 path requires packed B fragments
         if (x_timelast) {
             if constexpr (G::CIN % 2 == 0) {
-                hipLaunchKernelGGL((conv_stack_mfma_tlast_kernel<G>),
-                                   dim3(grid), dim3(WG_THREADS), 0, stream,
-                                   (const unsigned short*)x, feat, wpack,
-                                   (const unsigned short*)bfrag, SN);
+                int onewave = 0;
+                if (const char* e = getenv("TSKD_CONV_TLAST_1WAVE"))
+                    onewave = atoi(e);
+                if (onewave) {
+                    // 1-wave-per-2-windows variant (A/B reference)
+                    int g1 = (SN / 2 + WG_WAVES - 1) / WG_WAVES;
+                    if (g1 < 1) g1 = 1;
+                    if (g1 > 8192) g1 = 8192;
+                    hipLaunchKernelGGL((conv_stack_mfma_tlast_kernel<G>),
+                                       dim3(g1), dim3(WG_THREADS), 0, stream,
+                                       (const unsigned short*)x, feat, wpack,
+                                       (const unsigned short*)bfrag, SN);
+                } else {
+                    // default: 4-wave workgroup per window
+                    int g2 = SN > 16384 ? 16384 : SN;
+                    hipLaunchKernelGGL((conv_stack_mfma_wg_kernel<G>),
+                                       dim3(g2), dim3(WG_THREADS), 0, stream,
+                                       (const unsigned short*)x, feat, wpack,
+                                       (const unsigned short*)bfrag, SN);
+                }
                 return (int)hipGetLastError();
             }
             return -4;  // odd-CIN variants: use the staged layout
