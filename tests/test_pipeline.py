@@ -287,3 +287,28 @@ class TestFusedServe:
         assert n == 1 and store.count() == 1
         _, risk = store.latest("p000194")
         assert 0.0 <= risk <= 1.0
+
+    def test_fused_emit_processed_and_response(self, tmp_path, cfg):
+        from tskd_amd.cli.serve import FusedServer
+        bus = Bus(str(tmp_path / "bus"))
+        _write_wfdb_record(cfg.wavef_path, "p000194-test", cfg.channel_names,
+                           1 / 60, 40)
+        store = PredictionStore(str(tmp_path / "pred.log"))
+        srv = FusedServer(bus, cfg, store, device="cpu", max_streams=8,
+                          ring_grid=1024, starting="earliest",
+                          response_topic="model-response",
+                          emit_processed="call-stream")
+        cc = Consumer(bus, starting="earliest")
+        cc.subscribe(["call-stream", "model-response"])
+        send_record_data(bus, "p000194-test", None, 1e6, 1.0, cfg)
+        assert srv.trigger() == 1
+        msgs = cc.poll(max_msgs=256, timeout_ms=500)
+        topics = {m.topic for m in msgs}
+        assert topics == {"call-stream", "model-response"}
+        resp = [m for m in msgs if m.topic == "model-response"]
+        assert resp[0].key == b"p000194"
+        payload = json.loads(resp[0].value)
+        assert 0.0 <= payload["risk"] <= 1.0
+        # call-stream carries the two-stage wire contract
+        proc_keys = {m.key.decode() for m in msgs if m.topic == "call-stream"}
+        assert "p000194_0" in proc_keys

@@ -50,6 +50,7 @@ class FusedServer:
                  device: str = "cpu", max_streams: int = 1024,
                  ring_grid: int = 4096, starting: str = "latest",
                  response_topic: Optional[str] = None,
+                 emit_processed: Optional[str] = None,
                  rank: int = 0, world: int = 1):
         self.cfg = cfg
         self.store = store
@@ -67,10 +68,16 @@ class FusedServer:
         for t in topics:
             bus.create_topic(t)
         self.consumer.subscribe(topics)
-        self.producer = Producer(bus) if response_topic else None
+        self.producer = Producer(bus) if (response_topic or emit_processed) \
+            else None
         self.response_topic = response_topic
         if response_topic:
             bus.create_topic(response_topic)
+        # optional call-stream emission so the dashboard's processed plots
+        # work in fused mode too (the two-stage wire contract)
+        self.emit_processed = emit_processed
+        if emit_processed:
+            bus.create_topic(emit_processed)
         self.pid_index: Dict[str, int] = {}
         self.pids: list = []
         self.max_streams = max_streams
@@ -118,8 +125,19 @@ class FusedServer:
                 self.se._clear_ahead(int(advance / self.se.bucket_s))
                 self.se.head = int(advance / self.se.bucket_s)
                 self.se._refill()
-            if self.se.nproc == nproc_before or not self.se.ready \
-                    or not self.pids:
+            np_new = self.se.nproc - nproc_before
+            if self.emit_processed and np_new > 0 and self.pids:
+                import json as _json
+                proc = self.se.proc
+                for pid, sid in self.pid_index.items():
+                    for c in range(self.cfg.n_channels):
+                        pts = [float(proc[sid, c,
+                                          (nproc_before + j) % self.se.G])
+                               for j in range(np_new)]
+                        self.producer.produce(self.emit_processed,
+                                              f"{pid}_{c}", _json.dumps(pts),
+                                              ts_us=int(self.hwm * 1e6))
+            if np_new <= 0 or not self.se.ready or not self.pids:
                 return 0
             dtype = torch.bfloat16 if self.device != "cpu" else torch.float32
             w = self.se.windows(batch=1, stride=12, dtype=dtype,
@@ -171,6 +189,9 @@ def main(argv=None) -> None:
     ap.add_argument("--starting", default="latest",
                     choices=["latest", "earliest"])
     ap.add_argument("--model-response-topic", default=None)
+    ap.add_argument("--emit-processed", default=None, metavar="TOPIC",
+                    help="also publish processed grid points to TOPIC "
+                         "(e.g. call-stream) for the dashboard")
     ap.add_argument("--max-triggers", type=int, default=0)
     args = ap.parse_args(argv)
 
@@ -189,6 +210,7 @@ def main(argv=None) -> None:
                       device=args.device, max_streams=args.max_streams,
                       starting=args.starting,
                       response_topic=args.model_response_topic,
+                      emit_processed=args.emit_processed,
                       rank=rank, world=world)
     period = cfg.predict_slide_s / args.speed
     stop = []
