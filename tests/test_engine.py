@@ -365,3 +365,31 @@ class TestZNormalize:
         torch.cuda.synchronize()
         np.testing.assert_allclose(wg.cpu().numpy(), wc.numpy(), rtol=1e-3,
                                    atol=1e-4)
+
+
+@pytest.mark.gpu
+class TestHeavyEventLoad:
+    def test_half_million_events_match_cpu(self):
+        """Bulk atomically-ingested events (the sparse path under load)."""
+        fs = 5.0
+        rng = np.random.default_rng(23)
+        n = 500_000
+        S, C = 64, 10
+        ts = rng.uniform(0, 2000, n)
+        si = rng.integers(0, S, n)
+        ci = rng.integers(0, C, n)
+        vv = rng.normal(80, 10, n).astype(np.float32)
+        args = (torch.tensor(si, dtype=torch.long),
+                torch.tensor(ci, dtype=torch.long),
+                torch.tensor(ts), torch.tensor(vv))
+        gpu = StreamEngine(S, C, ring_grid=1024, fs=fs, device="cuda")
+        gpu.ingest_events(*args, advance_to=2000)
+        torch.cuda.synchronize()
+        # spot-check 8 (stream, channel) pairs against the numpy oracle
+        for s_ in (0, 17, 40, 63):
+            for c_ in (0, 7):
+                m = (si == s_) & (ci == c_)
+                want = preprocess_series_oracle(ts[m], vv[m], gpu.head)
+                got = gpu.proc[s_, c_, :gpu.nproc].cpu().numpy()
+                np.testing.assert_allclose(got, want[:gpu.nproc], rtol=2e-4,
+                                           atol=2e-4)
