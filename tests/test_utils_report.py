@@ -106,3 +106,12 @@ class TestCsvDemo:
         assert ((out["scores"] >= 0) & (out["scores"] <= 1)).all()
         # 2 h at 1 Hz with 50% dropout -> full 5-s grid coverage
         assert out["grid_points"] > 1000
+
+
+class TestScriptsImportable:
+    def test_scripts_parse(self):
+        """The operational scripts must at least import/compile on CPU."""
+        import py_compile
+        for f in ("scripts/ab_bench.py", "scripts/gpu_soak.py",
+                  "scripts/demo_e2e.py"):
+            py_compile.compile(f, doraise=True)
