@@ -312,3 +312,39 @@ class TestFusedServe:
         # call-stream carries the two-stage wire contract
         proc_keys = {m.key.decode() for m in msgs if m.topic == "call-stream"}
         assert "p000194_0" in proc_keys
+
+    def test_hot_reload(self, tmp_path, cfg):
+        """Swapping the checkpoint file between triggers changes the served
+        model (the reference loads once at start and never reloads)."""
+        from tskd_amd.cli.serve import FusedServer
+        from tskd_amd.models import build_model, save_checkpoint
+        bus = Bus(str(tmp_path / "bus"))
+        _write_wfdb_record(cfg.wavef_path, "p000194-test", cfg.channel_names,
+                           1 / 60, 52)
+        store = PredictionStore(str(tmp_path / "pred.log"))
+        ckpt = str(tmp_path / "model.pth")
+        torch.manual_seed(1)
+        save_checkpoint(build_model("MyCNN5").eval(), ckpt)
+        from tskd_amd.models import load_checkpoint
+        srv = FusedServer(bus, cfg, store, model=load_checkpoint(ckpt),
+                          device="cpu", max_streams=8, ring_grid=1024,
+                          starting="earliest")
+        srv.model_path = ckpt
+        import os as _os
+        srv._model_mtime = _os.path.getmtime(ckpt)
+        send_record_data(bus, "p000194-test", None, 1e6, 1.0, cfg)
+        assert not srv.maybe_reload_model()  # unchanged file
+        assert srv.trigger() == 1
+        _, risk1 = store.latest("p000194")
+        # retrain/replace the model on disk with different weights
+        torch.manual_seed(99)
+        save_checkpoint(build_model("MyCNN5").eval(), ckpt)
+        _os.utime(ckpt, (time.time() + 2, time.time() + 2))
+        assert srv.maybe_reload_model()
+        # more data -> a new prediction from the NEW model
+        _write_wfdb_record(cfg.wavef_path, "p000194-test", cfg.channel_names,
+                           1 / 60, 64, seed=5)
+        send_record_data(bus, "p000194-test", None, 1e6, 1.0, cfg)
+        assert srv.trigger() == 1
+        _, risk2 = store.latest("p000194")
+        assert abs(risk1 - risk2) > 1e-6  # different weights, different score

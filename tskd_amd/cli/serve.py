@@ -85,6 +85,29 @@ class FusedServer:
         self.watermark_s = cfg.watermark_s
         self.timer = StageTimer("serve")
         self.n_predictions = 0
+        self.model_path: Optional[str] = None   # set to enable hot reload
+        self._model_mtime = 0.0
+
+    def maybe_reload_model(self) -> bool:
+        """Hot-reload the checkpoint when the file changes (the reference
+        loads once at process start and never again — predictStream.py:36).
+        Weight repack is ~6 KB; the engine swaps atomically between triggers.
+        """
+        if not self.model_path or not os.path.exists(self.model_path):
+            return False
+        mtime = os.path.getmtime(self.model_path)
+        if mtime <= self._model_mtime:
+            return False
+        from tskd_amd.models import load_checkpoint
+        try:
+            model = load_checkpoint(self.model_path)
+        except Exception as e:  # partial write etc. — retry next trigger
+            log.warning("hot reload failed (%s); keeping current model", e)
+            return False
+        self.me = MyCNNEngine(model, device=self.device)
+        self._model_mtime = mtime
+        log.info("hot-reloaded model from %s", self.model_path)
+        return True
 
     def _sid(self, pid: str) -> Optional[int]:
         if self.world > 1 and shard_for_key(pid, self.world) != self.rank:
@@ -193,6 +216,8 @@ def main(argv=None) -> None:
                     help="also publish processed grid points to TOPIC "
                          "(e.g. call-stream) for the dashboard")
     ap.add_argument("--max-triggers", type=int, default=0)
+    ap.add_argument("--hot-reload", action="store_true",
+                    help="reload the checkpoint when the file changes")
     args = ap.parse_args(argv)
 
     rank, world = init_distributed()
@@ -212,12 +237,17 @@ def main(argv=None) -> None:
                       response_topic=args.model_response_topic,
                       emit_processed=args.emit_processed,
                       rank=rank, world=world)
+    if args.hot_reload and args.model_path:
+        srv.model_path = args.model_path
+        if os.path.exists(args.model_path):
+            srv._model_mtime = os.path.getmtime(args.model_path)
     period = cfg.predict_slide_s / args.speed
     stop = []
     signal.signal(signal.SIGTERM, lambda *a: stop.append(1))
     n = 0
     while not stop:
         t0 = time.time()
+        srv.maybe_reload_model()
         srv.trigger()
         n += 1
         if n % 10 == 0:
