@@ -62,6 +62,26 @@ class PredictStream:
             lambda: deque(maxlen=self.win))
         self.last_ts_us: Dict[str, int] = {}
         self.n_predictions = 0
+        self.model_path: Optional[str] = None  # set to enable hot reload
+        self._model_mtime = 0.0
+
+    def maybe_reload_model(self) -> bool:
+        """Hot-reload the checkpoint when the file changes (the reference
+        loads once at process start — predictStream.py:36)."""
+        if not self.model_path or not os.path.exists(self.model_path):
+            return False
+        mtime = os.path.getmtime(self.model_path)
+        if mtime <= self._model_mtime:
+            return False
+        try:
+            model = load_checkpoint(self.model_path)
+        except Exception as e:
+            log.warning("hot reload failed (%s); keeping current model", e)
+            return False
+        self.engine = MyCNNEngine(model, device=self.device)
+        self._model_mtime = mtime
+        log.info("hot-reloaded model from %s", self.model_path)
+        return True
 
     def trigger(self) -> int:
         msgs = self.consumer.poll(max_msgs=65536, timeout_ms=0)
@@ -147,6 +167,8 @@ def main(argv=None) -> None:
     ap.add_argument("--max-triggers", type=int, default=0)
     ap.add_argument("--offsets-file", default=None,
                     help="persist/restore consumer offsets (resume-on-restart)")
+    ap.add_argument("--hot-reload", action="store_true",
+                    help="reload the checkpoint when the file changes")
     args = ap.parse_args(argv)
 
     bus = Bus(args.bus_dir)
@@ -163,6 +185,10 @@ def main(argv=None) -> None:
                        call_topic=args.model_call_topic,
                        response_topic=args.model_response_topic,
                        starting=args.starting)
+    if args.hot_reload and args.model_path:
+        ps.model_path = args.model_path
+        if os.path.exists(args.model_path):
+            ps._model_mtime = os.path.getmtime(args.model_path)
     trigger_period = cfg.predict_slide_s / args.speed
     stop = []
     signal.signal(signal.SIGTERM, lambda *a: stop.append(1))
@@ -177,6 +203,7 @@ def main(argv=None) -> None:
     n = 0
     while not stop:
         t0 = time.time()
+        ps.maybe_reload_model()
         with timer:
             out = ps.trigger()
         timer.add_items(out)
