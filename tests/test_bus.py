@@ -152,3 +152,55 @@ class TestNativeParse:
         assert ts.tolist() == [5.0, 6.0]
         # the malformed message was consumed too (no re-delivery)
         assert c.poll(timeout_ms=10) == []
+
+
+class TestRetention:
+    def test_trim_reclaims_and_preserves_offsets(self, bus, tmp_path):
+        bus.create_topic("ret")
+        p = Producer(bus)
+        blob = "y" * 8000
+        for i in range(200):  # ~1.6 MB
+            p.produce("ret", f"k{i}", blob)
+        c = Consumer(bus, starting="earliest")
+        c.subscribe(["ret"])
+        msgs = c.poll(max_msgs=200, timeout_ms=500)
+        assert len(msgs) == 200
+        import os
+        part_file = None
+        for root, _, files in os.walk(bus.dir):
+            for f in files:
+                if f == "p0.log" and "ret" in root:
+                    part_file = os.path.join(root, f)
+        blocks_before = os.stat(part_file).st_blocks
+        cut = msgs[150].offset
+        applied = bus.trim_topic("ret", 0, cut)
+        assert 0 < applied <= cut
+        blocks_after = os.stat(part_file).st_blocks
+        assert blocks_after < blocks_before  # storage actually reclaimed
+        # new earliest consumer starts at the trim point, offsets unchanged
+        c2 = Consumer(bus, starting="earliest")
+        c2.subscribe(["ret"])
+        survivors = c2.poll(max_msgs=200, timeout_ms=500)
+        assert survivors
+        assert survivors[0].offset >= applied
+        assert survivors[-1].offset == msgs[-1].offset  # absolute offsets
+        keys = [m.key.decode() for m in survivors]
+        assert keys == [f"k{i}" for i in range(200 - len(keys), 200)]
+
+    def test_consumer_mid_gap_skips_forward(self, bus):
+        bus.create_topic("ret2")
+        p = Producer(bus)
+        for i in range(50):
+            p.produce("ret2", "k", "x" * 4000)
+        c = Consumer(bus, starting="earliest")
+        c.subscribe(["ret2"])
+        first = c.poll(max_msgs=5, timeout_ms=200)  # position now at msg 5
+        side = Consumer(bus, starting="earliest")
+        side.subscribe(["ret2"])
+        all_msgs = side.poll(max_msgs=100, timeout_ms=200)
+        bus.trim_topic("ret2", 0, all_msgs[25].offset)  # a record boundary
+        rest = c.poll(max_msgs=100, timeout_ms=200)
+        # consumer skipped the retention gap and continued
+        assert rest
+        assert rest[0].offset >= bus.trim_offset("ret2")
+        assert first[-1].next_offset < rest[0].offset

@@ -50,6 +50,17 @@ class Bus:
     def partition_for(self, topic: str, key: str) -> int:
         return self._c.partition_for(topic, key)
 
+    def trim_offset(self, topic: str, partition: int = 0) -> int:
+        return self._c.trim_offset(topic, partition)
+
+    def trim_topic(self, topic: str, partition: int = 0,
+                   before_offset: int = 0) -> int:
+        """Retention: reclaim storage for messages before ``before_offset``
+        (hole punch; offsets stay absolute — reads below the trim point are
+        a consumer-visible gap, like Kafka retention). Returns the applied
+        page-aligned trim offset."""
+        return self._c.trim_topic(topic, partition, before_offset)
+
 
 class Producer:
     """acks=all, retries=5 producer (reference utils.py:417-422)."""

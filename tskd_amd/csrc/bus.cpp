@@ -30,6 +30,8 @@
 #include <dirent.h>
 #include <fcntl.h>
 #include <pthread.h>
+#include <fcntl.h>
+#include <linux/falloc.h>
 #include <sys/mman.h>
 #include <sys/stat.h>
 #include <unistd.h>
@@ -49,6 +51,8 @@ struct PartHeader {
     pthread_mutex_t mtx;               // process-shared writer lock
     std::atomic<uint64_t> committed;   // valid data bytes after header
     std::atomic<uint64_t> next_seq;    // per-partition message sequence
+    std::atomic<uint64_t> trimmed;     // bytes before this offset reclaimed
+                                       // (retention; zero-init on old files)
 };
 static_assert(sizeof(PartHeader) <= HDR_SIZE, "header fits");
 
@@ -91,6 +95,7 @@ public:
             pthread_mutexattr_destroy(&at);
             h->committed.store(0);
             h->next_seq.store(0);
+            h->trimmed.store(0);
             h->version = 1;
             std::atomic_thread_fence(std::memory_order_release);
             h->magic = MAGIC;
@@ -113,6 +118,44 @@ public:
     uint64_t committed() {
         maybe_remap();
         return hdr()->committed.load(std::memory_order_acquire);
+    }
+
+    uint64_t trimmed() {
+        maybe_remap();
+        return hdr()->trimmed.load(std::memory_order_acquire);
+    }
+
+    // Retention: reclaim storage for data before `before_off`, which MUST
+    // be a record boundary (an offset the bus handed out — Message.offset /
+    // next_offset / end_offset). The logical trim point is stored exactly;
+    // only the punched region is page-aligned (the partial page survives).
+    // Offsets stay ABSOLUTE and stable; reads below the trim point are a
+    // consumer-visible gap (Kafka-retention semantics).
+    uint64_t trim(uint64_t before_off) {
+        auto* h = hdr();
+        int rc = pthread_mutex_lock(&h->mtx);
+        if (rc == EOWNERDEAD) pthread_mutex_consistent(&h->mtx);
+        uint64_t lim = h->committed.load(std::memory_order_relaxed);
+        if (before_off > lim) before_off = lim;
+        uint64_t old = h->trimmed.load(std::memory_order_relaxed);
+        if (before_off > old) {
+            const uint64_t punch_from = old & ~(uint64_t)4095;
+            const uint64_t punch_to = before_off & ~(uint64_t)4095;
+            if (punch_to > punch_from) {
+#ifdef FALLOC_FL_PUNCH_HOLE
+                if (fallocate(fd_, FALLOC_FL_PUNCH_HOLE | FALLOC_FL_KEEP_SIZE,
+                              (off_t)(HDR_SIZE + punch_from),
+                              (off_t)(punch_to - punch_from)) != 0)
+#endif
+                    // fs without hole support: zero instead (no reclaim,
+                    // same gap semantics)
+                    memset(data() + punch_from, 0, punch_to - punch_from);
+            }
+            h->trimmed.store(before_off, std::memory_order_release);
+        }
+        uint64_t out = h->trimmed.load(std::memory_order_relaxed);
+        pthread_mutex_unlock(&h->mtx);
+        return out;
     }
 
     void append(const std::string& key, const std::string& val, int64_t ts_us) {
@@ -146,6 +189,8 @@ public:
               int64_t& ts_us, uint64_t& next) {
         const uint64_t lim = committed();
         if (off + sizeof(MsgHeader) > lim) return false;
+        // below the retention trim point: caller must skip forward
+        if (off < hdr()->trimmed.load(std::memory_order_acquire)) return false;
         auto* m = reinterpret_cast<MsgHeader*>(data() + off);
         const size_t need = rec_size(m->klen, m->vlen);
         if (off + need > lim) return false;
@@ -258,6 +303,17 @@ public:
 
     uint64_t end_offset(const std::string& topic, int p) {
         return part(topic, p)->committed();
+    }
+
+    uint64_t trim_offset(const std::string& topic, int p) {
+        return part(topic, p)->trimmed();
+    }
+
+    // Retention: drop data before `before_offset` in every partition (or a
+    // single one); returns the applied (aligned) trim offsets.
+    uint64_t trim_topic(const std::string& topic, int p,
+                        uint64_t before_offset) {
+        return part(topic, p)->trim(before_offset);
     }
 
     int partition_for(const std::string& topic, const std::string& key) {
@@ -384,8 +440,9 @@ public:
             for (int p = 0; p < n; ++p) {
                 const std::string k = t + "/" + std::to_string(p);
                 if (pos_.count(k)) continue;
-                pos_[k] = starting_ == "earliest" ? 0
-                                                  : bus_->end_offset(t, p);
+                pos_[k] = starting_ == "earliest"
+                              ? bus_->trim_offset(t, p)
+                              : bus_->end_offset(t, p);
                 parts_.push_back({t, p});
             }
         }
@@ -406,6 +463,8 @@ public:
                 const std::string k = t + "/" + std::to_string(p);
                 auto* pm = bus_->part(t, p);
                 uint64_t off = pos_[k];
+                const uint64_t trim = pm->trimmed();
+                if (off < trim) off = trim;  // retention gap: skip forward
                 Message m;
                 while ((int)out.size() < max_msgs &&
                        pm->read(off, m.key_, m.val_, m.seq, m.ts_us,
@@ -481,6 +540,9 @@ PYBIND11_MODULE(_tskd_bus, m) {
         .def("topic_nparts", &Bus::topic_nparts)
         .def("list_topics", &Bus::list_topics)
         .def("end_offset", &Bus::end_offset)
+        .def("trim_offset", &Bus::trim_offset)
+        .def("trim_topic", &Bus::trim_topic, py::arg("topic"),
+             py::arg("partition"), py::arg("before_offset"))
         .def("partition_for", &Bus::partition_for)
         .def_property_readonly("dir", &Bus::dir);
     py::class_<Producer>(m, "Producer")
