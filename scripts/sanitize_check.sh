@@ -16,6 +16,10 @@ case "$SAN" in
   thread)  LIB="$GCCDIR/libtsan.so"; FLAG=-fsanitize=thread ;;
   *) echo "unknown sanitizer $SAN"; exit 2 ;;
 esac
+# libstdc++ must be preloaded too or the sanitizer's __cxa_throw
+# interceptor cannot resolve the real symbol (C++ exceptions from the
+# modules would abort the run)
+LIB="$LIB $(g++ -print-file-name=libstdc++.so.6)"
 
 PYINC=$(python3 -c "import sysconfig; print(sysconfig.get_paths()['include'])")
 PBINC=$(python3 -c "import pybind11; print(pybind11.get_include())")
