@@ -199,6 +199,11 @@ public:
         seq = m->seq;
         ts_us = m->ts_us;
         next = off + need;
+        // a concurrent trim may have zeroed this record mid-read (destructive
+        // retention, like Kafka segment deletion): re-check and discard.
+        if (off < hdr()->trimmed.load(std::memory_order_acquire)) return false;
+        if (m->klen == 0 && m->vlen == 0 && seq == 0 && ts_us == 0)
+            return false;  // fully-zeroed (punched) region
         return true;
     }
 
