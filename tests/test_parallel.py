@@ -240,3 +240,28 @@ class TestFusedServeDP:
         assert st.count() == 4  # both ranks persisted into ONE store
         got = {r[0] for r in st.tail(10)}
         assert got == set(pids)
+
+
+class TestRetentionFlow:
+    def test_processstream_trim_consumed(self, tmp_path):
+        """--trim-consumed reclaims raw-topic storage behind the consumer."""
+        bus_dir = str(tmp_path / "bus")
+        from tskd_amd.bus import Bus, Producer
+        bus = Bus(bus_dir)
+        bus.create_topic("HR")
+        p = Producer(bus)
+        blob = 600  # messages ~ 40 B each
+        for i in range(20000):
+            p.produce("HR", "p000194", json.dumps([0, float(i % blob)]),
+                      ts_us=int(i * 60e6))
+        end_before = bus.end_offset("HR", 0)
+        env = dict(os.environ, PYTHONPATH=REPO)
+        cmd = [sys.executable, "-m", "tskd_amd.cli.processstream",
+               "--bus-dir", bus_dir, "--starting", "earliest",
+               "--trim-consumed", "--max-triggers", "5",
+               "--device", "cpu", "--speed", "1e9", "--max-streams", "4"]
+        r = subprocess.run(cmd, env=env, capture_output=True, text=True,
+                           timeout=180)
+        assert r.returncode == 0, r.stderr[-2000:]
+        assert bus.trim_offset("HR", 0) == end_before  # fully consumed+trimmed
+        assert bus.end_offset("HR", 0) == end_before   # offsets stable

@@ -69,7 +69,7 @@ class ProcessStream:
             self.hwm = max(self.hwm, float(ts_arr.max()))
         advance = max(0.0, self.hwm - self.watermark_s)
         if si:
-            self.engine.ingest_events(
+            self.engine.ingest_events_chunked(
                 torch.tensor(si, dtype=torch.long),
                 torch.from_numpy(chans).long(),
                 torch.from_numpy(ts_arr),
@@ -85,6 +85,13 @@ class ProcessStream:
         np_new = self.engine.nproc - nproc_before
         if np_new <= 0:
             return 0
+        # catch-up past ring capacity: only the retained tail is emittable
+        max_emit = self.engine.G - self.engine.win_buckets - 1
+        if np_new > max_emit:
+            log.warning("catch-up produced %d points; emitting only the "
+                        "retained last %d", np_new, max_emit)
+            nproc_before = self.engine.nproc - max_emit
+            np_new = max_emit
         # emit the new processed points per (patient, channel)
         proc = self.engine.proc
         emitted = 0
@@ -120,6 +127,10 @@ def main(argv=None) -> None:
                     help="stop after N triggers (0 = run forever)")
     ap.add_argument("--offsets-file", default=None,
                     help="persist/restore consumer offsets (resume-on-restart)")
+    ap.add_argument("--trim-consumed", action="store_true",
+                    help="retention: reclaim bus storage behind THIS "
+                         "consumer's position after each trigger (only safe "
+                         "when no other consumer needs the raw topics)")
     args = ap.parse_args(argv)
     if args.signal_list:
         cfg.channel_names = args.signal_list
@@ -148,6 +159,10 @@ def main(argv=None) -> None:
         timer.add_items(out)
         if args.offsets_file:
             save_offsets(args.offsets_file, ps.consumer.positions())
+        if args.trim_consumed:
+            for key, off in ps.consumer.positions().items():
+                topic, _, part = key.rpartition("/")
+                bus.trim_topic(topic, int(part), off)
         n += 1
         if n % 10 == 0:
             log.info("metrics %s", timer.log_line())

@@ -138,7 +138,7 @@ class FusedServer:
             advance = max(0.0, self.hwm - self.watermark_s)
             nproc_before = self.se.nproc
             if si:
-                self.se.ingest_events(
+                self.se.ingest_events_chunked(
                     torch.tensor(si, dtype=torch.long),
                     torch.tensor(ci, dtype=torch.long),
                     torch.tensor(tt, dtype=torch.float64),

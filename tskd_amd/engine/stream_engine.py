@@ -244,6 +244,44 @@ class StreamEngine:
             self.head = new_head
         self._refill()
 
+    def ingest_events_chunked(self, stream_idx, chan_idx, ts, vals,
+                              advance_to=None) -> None:
+        """ingest_events for arbitrarily large backlogs: events are
+        processed in bucket-ordered chunks no wider than the ring allows
+        (catch-up replay after a long outage). Grid history older than the
+        ring is overwritten — only the last ~G points are retained."""
+        cap = self.G - self.win_buckets - 1
+        if len(ts) == 0:
+            if advance_to is not None:
+                self.ingest_events(stream_idx, chan_idx, ts, vals,
+                                   advance_to=advance_to)
+            return
+        bucket = torch.floor(ts.double() / self.bucket_s).long()
+        span = int(bucket.max()) - max(int(bucket.min()), self.nproc)
+        if span < cap:
+            self.ingest_events(stream_idx, chan_idx, ts, vals,
+                               advance_to=advance_to)
+            return
+        order = torch.argsort(bucket)
+        b_sorted = bucket[order]
+        lo = 0
+        n = len(order)
+        while lo < n:
+            start_bucket = max(int(b_sorted[lo]), self.nproc)
+            hi = int(torch.searchsorted(b_sorted,
+                                        torch.tensor(start_bucket + cap)))
+            hi = max(hi, lo + 1)
+            idx = order[lo:hi]
+            chunk_adv = float((int(b_sorted[min(hi, n) - 1]) + 1)
+                              * self.bucket_s)
+            self.ingest_events(stream_idx[idx], chan_idx[idx], ts[idx],
+                               vals[idx], advance_to=chunk_adv)
+            lo = hi
+        if advance_to is not None and advance_to / self.bucket_s > self.head:
+            self._clear_ahead(int(advance_to / self.bucket_s))
+            self.head = int(advance_to / self.bucket_s)
+            self._refill()
+
     def _clear_ahead(self, upto_bucket: int) -> None:
         """Zero ring slots about to be re-used (stale data from G buckets
         ago). No-op until the ring wraps."""
