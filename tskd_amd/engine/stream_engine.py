@@ -231,13 +231,15 @@ class StreamEngine:
             if rc != 0:
                 raise RuntimeError(f"ingest_events failed: hipError {rc}")
         else:
-            ok = ~torch.isnan(vals)
-            for s, c, b, v in zip(stream_idx[ok].tolist(), chan_idx[ok].tolist(),
-                                  bucket[ok].tolist(), vals[ok].tolist()):
-                if b < min_bucket:
-                    continue
-                self.bsum[s, c, b % self.G] += v
-                self.bcnt[s, c, b % self.G] += 1
+            ok = (~torch.isnan(vals)) & (bucket >= min_bucket)
+            if ok.any():
+                flat = ((stream_idx[ok].numpy() * self.C
+                         + chan_idx[ok].numpy()) * self.G
+                        + (bucket[ok].numpy() % self.G))
+                bs = self.bsum.numpy().reshape(-1)
+                bc = self.bcnt.numpy().reshape(-1)
+                np.add.at(bs, flat, vals[ok].numpy().astype(np.float64))
+                np.add.at(bc, flat, 1.0)
         hwm = float(advance_to) if advance_to is not None else float(ts.max())
         new_head = int(np.floor(hwm / self.bucket_s))
         if new_head > self.head:
@@ -358,18 +360,17 @@ class StreamEngine:
             bc = self.bcnt.numpy()
             pr = self.proc.numpy()
             lv = self.last_val.numpy()
+            # vectorized sliding sums via cumsum over the gathered bucket
+            # range [nproc, nproc + np_new + win - 1)
+            span = np_new + self.win_buckets - 1
+            gidx = (self.nproc + np.arange(span)) % self.G
+            pidx = (self.nproc + np.arange(np_new)) % self.G
             for s in range(self.S):
                 for c in range(self.C):
-                    vals = np.empty(np_new)
-                    for j in range(np_new):
-                        g0 = self.nproc + j
-                        idx = (g0 + np.arange(self.win_buckets)) % self.G
-                        cnt = bc[s, c, idx].sum()
-                        vals[j] = (bs[s, c, idx].sum() / cnt if cnt > 0
-                                   else np.nan)
+                    vals = W.window_averages(bs[s, c, gidx], bc[s, c, gidx],
+                                             self.win_buckets)
                     filled, carry = W.fill_series(vals, lv[s, c])
-                    for j in range(np_new):
-                        pr[s, c, (self.nproc + j) % self.G] = filled[j]
+                    pr[s, c, pidx] = filled
                     lv[s, c] = carry
         self.nproc = navail
 
