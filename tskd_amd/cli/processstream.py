@@ -92,15 +92,18 @@ class ProcessStream:
                         "retained last %d", np_new, max_emit)
             nproc_before = self.engine.nproc - max_emit
             np_new = max_emit
-        # emit the new processed points per (patient, channel)
-        proc = self.engine.proc
+        # emit the new processed points per (patient, channel) — one
+        # vectorized gather, then host lists (no per-point .item() calls)
+        gidx = torch.tensor([(nproc_before + j) % self.engine.G
+                             for j in range(np_new)], dtype=torch.long,
+                            device=self.engine.device)
+        block = self.engine.proc.index_select(2, gidx).cpu()
         emitted = 0
         for pid, sid in self.pid_index.items():
+            rows = block[sid].tolist()
             for c in range(self.cfg.n_channels):
-                pts = [float(proc[sid, c, (nproc_before + j) % self.engine.G])
-                       for j in range(np_new)]
                 self.producer.produce(self.out_topic, f"{pid}_{c}",
-                                      json.dumps(pts),
+                                      json.dumps(rows[c]),
                                       ts_us=int(self.hwm * 1e6))
                 emitted += 1
         self.producer.flush(self.out_topic)

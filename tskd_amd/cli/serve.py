@@ -151,14 +151,22 @@ class FusedServer:
             np_new = self.se.nproc - nproc_before
             if self.emit_processed and np_new > 0 and self.pids:
                 import json as _json
-                proc = self.se.proc
+                max_emit = self.se.G - self.se.win_buckets - 1
+                if np_new > max_emit:  # catch-up: only the retained tail
+                    nproc_emit = self.se.nproc - max_emit
+                    np_emit = max_emit
+                else:
+                    nproc_emit, np_emit = nproc_before, np_new
+                gidx = torch.tensor([(nproc_emit + j) % self.se.G
+                                     for j in range(np_emit)],
+                                    dtype=torch.long, device=self.se.device)
+                block = self.se.proc.index_select(2, gidx).cpu()
                 for pid, sid in self.pid_index.items():
+                    rows = block[sid].tolist()
                     for c in range(self.cfg.n_channels):
-                        pts = [float(proc[sid, c,
-                                          (nproc_before + j) % self.se.G])
-                               for j in range(np_new)]
                         self.producer.produce(self.emit_processed,
-                                              f"{pid}_{c}", _json.dumps(pts),
+                                              f"{pid}_{c}",
+                                              _json.dumps(rows[c]),
                                               ts_us=int(self.hwm * 1e6))
             if np_new <= 0 or not self.se.ready or not self.pids:
                 return 0
