@@ -313,6 +313,31 @@ class TestFusedServe:
         proc_keys = {m.key.decode() for m in msgs if m.topic == "call-stream"}
         assert "p000194_0" in proc_keys
 
+    def test_fused_emit_catchup_clamp(self, tmp_path, cfg):
+        """A backlog wider than the proc ring must not wrap the emitted
+        processed-point arrays: the server clamps emission to the retained
+        tail (G - win_buckets - 1 points), mirroring processstream's clamp."""
+        from tskd_amd.cli.serve import FusedServer
+        bus = Bus(str(tmp_path / "bus"))
+        # 600 samples at fs=1/60 span 36000 s = 7200 buckets >> ring G=256
+        _write_wfdb_record(cfg.wavef_path, "p000194-test", cfg.channel_names,
+                           1 / 60, 600)
+        store = PredictionStore(str(tmp_path / "pred.log"))
+        srv = FusedServer(bus, cfg, store, device="cpu", max_streams=8,
+                          ring_grid=256, starting="earliest",
+                          emit_processed="call-stream")
+        cc = Consumer(bus, starting="earliest")
+        cc.subscribe(["call-stream"])
+        send_record_data(bus, "p000194-test", None, 1e6, 1.0, cfg)
+        assert srv.trigger() >= 1
+        max_emit = srv.se.G - srv.se.win_buckets - 1
+        msgs = cc.poll(max_msgs=4096, timeout_ms=500)
+        assert msgs, "no processed points emitted"
+        for m in msgs:
+            pts = json.loads(m.value)
+            assert len(pts) == max_emit
+            assert all(np.isfinite(pts))
+
     def test_hot_reload(self, tmp_path, cfg):
         """Swapping the checkpoint file between triggers changes the served
         model (the reference loads once at start and never reloads)."""

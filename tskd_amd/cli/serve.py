@@ -195,7 +195,7 @@ class FusedServer:
                                         [t_us] * n_active,
                                         local.tolist())
                 self.n_predictions += n_active
-            if self.producer:
+            if self.producer and self.response_topic:
                 for i, pid in enumerate(self.pids):
                     self.producer.produce(
                         self.response_topic, pid,
