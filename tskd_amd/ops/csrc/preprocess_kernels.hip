@@ -30,6 +30,7 @@
 // arguments so the whole trigger step can be captured in a hipGraph.
 
 #include <hip/hip_runtime.h>
+#include <cstdlib>
 #include <math.h>
 
 #define WAVE 64
@@ -293,6 +294,76 @@ __global__ __launch_bounds__(256) void window_fill_kernel(
     }
 }
 
+
+// Steady-state variant (np <= 16): the general kernel leaves 52 of 64 lanes
+// idle when a trigger yields only NB=12 new grid points. Pack FOUR
+// (stream, channel) rows per wave in 16-lane groups; the <= 51 bucket
+// reads per row are consecutive-address (coalesced within the group) and
+// L1-resident, so no LDS staging — only the 16-value ffill scan goes
+// through LDS. Dispatch in tskd_preproc_window_fill (knob TSKD_FILL16).
+__global__ __launch_bounds__(256) void window_fill16_kernel(
+    const float* __restrict__ bsum, const float* __restrict__ bcnt,
+    float* __restrict__ proc, float* __restrict__ last_val,
+    int S, int C, int G, long phead_in, int np, int win_buckets,
+    const long long* __restrict__ dstate)
+{
+    const long phead = ring_nproc(dstate, phead_in);
+    __shared__ float lds_v[4][64];        // per wave: 4 groups x 16 values
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int grp = lane / 16, gl = lane % 16;
+    float* lv = lds_v[wave] + grp * 16;
+
+    const long nsc = (long)S * C;
+    for (long sc0 = ((long)blockIdx.x * 4 + wave) * 4; sc0 < nsc;
+         sc0 += (long)gridDim.x * 16) {
+        const long sc = sc0 + grp;
+        const bool live = sc < nsc;
+        float carry = NAN, first_val = NAN, val = NAN;
+        int nan_prefix = 0;
+        if (live) {
+            carry = last_val[sc];
+            if (gl < np) {
+                const float* bs = bsum + sc * G;
+                const float* bc = bcnt + sc * G;
+                float sum = 0.f, cnt = 0.f;
+                int idx = (int)((phead + gl) % G);
+                for (int k = 0; k < win_buckets; ++k) {
+                    sum += bs[idx];
+                    cnt += bc[idx];
+                    if (++idx == G) idx = 0;
+                }
+                val = (cnt > 0.f) ? sum / cnt : NAN;
+            }
+        }
+        lv[gl] = val;
+        wsync_();
+        if (live && gl == 0) {            // serial ffill scan per group
+            for (int j = 0; j < np; ++j) {
+                const float v = lv[j];
+                if (isnan(v)) {
+                    if (isnan(carry)) ++nan_prefix;   // before first value
+                    else lv[j] = carry;
+                } else {
+                    if (isnan(carry)) first_val = v;
+                    carry = v;
+                }
+            }
+            last_val[sc] = carry;
+        }
+        wsync_();
+        if (live && gl < np) {
+            nan_prefix = __shfl(nan_prefix, grp * 16);
+            first_val = __shfl(first_val, grp * 16);
+            float outv = lv[gl];
+            if (gl < nan_prefix)          // bfill | fillna(0) prefix
+                outv = isnan(first_val) ? 0.f : first_val;
+            proc[sc * G + (phead + gl) % G] = outv;
+        }
+        wsync_();
+    }
+}
+
 // ---------------------------------------------------------------------------
 // 3. Window gather: (S, B, C, WIN) model inputs from processed grid.
 //    Window b (b = 0..B-1) covers grid [end - (B-1-b)*stride - WIN,
@@ -460,6 +531,16 @@ int tskd_preproc_window_fill(const float* bsum, const float* bcnt, float* proc,
                              int np, int win_buckets,
                              const long long* dstate, void* stream) {
     if (np <= 0) return 0;
+    const char* f16 = getenv("TSKD_FILL16");
+    if (np <= 16 && !(f16 && f16[0] == '0')) {
+        // steady-state: 4 (stream, channel) rows per wave, 16-lane groups
+        const long nsc4 = ((long)S * C + 15) / 16;
+        hipLaunchKernelGGL(window_fill16_kernel, dim3(grid_for(nsc4, 1)),
+                           dim3(256), 0, (hipStream_t)stream, bsum, bcnt,
+                           proc, last_val, S, C, G, phead, np, win_buckets,
+                           dstate);
+        return (int)hipGetLastError();
+    }
     const long nsc = (long)S * C * WAVE;  // one wave per (stream, channel)
     hipLaunchKernelGGL(window_fill_kernel, dim3(grid_for(nsc, 256)), dim3(256),
                        0, (hipStream_t)stream, bsum, bcnt, proc, last_val, S,
