@@ -115,3 +115,42 @@ class TestScriptsImportable:
         for f in ("scripts/ab_bench.py", "scripts/gpu_soak.py",
                   "scripts/demo_e2e.py"):
             py_compile.compile(f, doraise=True)
+
+
+class TestAuxCLIs:
+    def test_download_script_builder(self, tmp_path, monkeypatch):
+        """reference bin/download.py:5-25 — wget -r command per patient."""
+        from tskd_amd.cli.download import build_commands, main
+        cmds = build_commands(["p000194", "p044083"], "data/waveform")
+        assert len(cmds) == 2
+        assert cmds[0][0] == "wget" and "-r" in cmds[0]
+        assert cmds[0][-1].endswith("/p00/p000194/")
+        assert cmds[1][-1].endswith("/p04/p044083/")
+        monkeypatch.chdir(tmp_path)
+        main(["--patients", "p000194", "--script", "dl.sh"])
+        body = (tmp_path / "dl.sh").read_text()
+        assert body.startswith("#!/bin/sh") and "wget" in body
+
+    def test_mockstream_produces_wire_format(self, tmp_path):
+        """reference bin/mock-stream.py — random per-channel samples in the
+        [chan_idx, value] wire format on every channel topic."""
+        import json
+        from tskd_amd.bus import Bus, Consumer
+        from tskd_amd.cli.mockstream import main
+        from tskd_amd.config import get_global_config
+        cfg = get_global_config()
+        bus_dir = str(tmp_path / "bus")
+        main(["--bus-dir", bus_dir, "--n", "3", "--rate-hz", "1000",
+              "--patients", "p000194"])
+        bus = Bus(bus_dir)
+        cc = Consumer(bus, starting="earliest")
+        cc.subscribe([cfg.topic_for_channel(c) for c in cfg.channel_names])
+        msgs = cc.poll(max_msgs=256, timeout_ms=200)
+        assert len(msgs) == 3 * cfg.n_channels
+        seen = set()
+        for m in msgs:
+            assert m.key == b"p000194"
+            ci, val = json.loads(m.value)
+            seen.add(ci)
+            assert isinstance(val, float)
+        assert seen == set(range(cfg.n_channels))
