@@ -208,9 +208,10 @@ class TestWfdbRobustness:
         with pytest.raises(RuntimeError, match="unsupported format"):
             rd()
 
-    def test_multisegment_rejected(self, tmp_path):
-        rd = self._mk(str(tmp_path), "r/3 2 125 100\n")
-        with pytest.raises(RuntimeError, match="multi-segment"):
+    def test_multisegment_no_readable_segment(self, tmp_path):
+        # all-gap segment list -> no layout/real segment to define signals
+        rd = self._mk(str(tmp_path), "r/2 1 125 100\n~ 60\n~ 40\n")
+        with pytest.raises(RuntimeError, match="no readable segment"):
             rd()
 
     def test_missing_files(self, tmp_path):
@@ -238,3 +239,75 @@ class TestWfdbRobustness:
         rec = rd(channel_names=["SpO2", "HR"])  # SpO2 absent
         assert rec.sig_name == ["HR"]
         assert rec.p_signal.shape == (2, 1)
+
+
+class TestWfdbMultiSegment:
+    """Variable-layout multi-segment records (the MIMIC waveform format):
+    master header `rec/nseg`, `~ n` gap segments, a `_layout 0` header
+    naming the canonical channels, per-segment headers mapping a SUBSET of
+    those channels. Gaps and unmirrored segments read as NaN
+    (ref record p000194-2112-05-23-14-34.hea has exactly this shape)."""
+
+    def _mk_multiseg(self, d):
+        # layout: channels II, V  — master: gap 4, seg a (II only, 3),
+        # seg b (V,II reordered, 2), missing seg m (2) -> 11 samples total
+        open(f"{d}/ml_layout.hea", "w").write(
+            "ml_layout 2 125 0 10:00:00\n"
+            "~ 0 10/mV 16 0 0 0 0 II\n"
+            "~ 0 20/mV 16 0 0 0 0 V\n")
+        open(f"{d}/r.hea", "w").write(
+            "r/5 2 125 11 10:00:00 01/02/2100\n"
+            "ml_layout 0\n~ 4\n"
+            "sa 3\nsb 2\nsm 2\n")
+        open(f"{d}/sa.hea", "w").write(
+            "sa 1 125 3\nsa.dat 16 10/mV 16 0 0 0 0 II\n")
+        np.asarray([10, 20, 30], dtype="<i2").tofile(f"{d}/sa.dat")
+        open(f"{d}/sb.hea", "w").write(
+            "sb 2 125 2\n"
+            "sb.dat 16 20/mV 16 0 0 0 0 V\n"
+            "sb.dat 16 10/mV 16 0 0 0 0 II\n")
+        # interleaved (V, II) x 2 samples
+        np.asarray([40, 50, 80, 90], dtype="<i2").tofile(f"{d}/sb.dat")
+        open(f"{d}/sm.hea", "w").write(
+            "sm 1 125 2\nsm.dat 16 10/mV 16 0 0 0 0 II\n")
+        # sm.dat deliberately absent (unmirrored segment)
+        from tskd_amd.io import rdrecord
+        return lambda **kw: rdrecord(f"{d}/r", **kw)
+
+    def test_stitched_layout(self, tmp_path):
+        rec = self._mk_multiseg(str(tmp_path))()
+        assert rec.n_seg == 5
+        assert rec.sig_name == ["II", "V"]
+        assert rec.p_signal.shape == (11, 2)
+        ii, v = rec.p_signal[:, 0], rec.p_signal[:, 1]
+        assert np.isnan(ii[:4]).all() and np.isnan(v[:4]).all()  # gap
+        np.testing.assert_allclose(ii[4:7], [1.0, 2.0, 3.0])     # sa /10
+        assert np.isnan(v[4:7]).all()        # V absent from segment sa
+        np.testing.assert_allclose(v[7:9], [2.0, 4.0])           # sb /20
+        np.testing.assert_allclose(ii[7:9], [5.0, 9.0])          # sb /10
+        assert np.isnan(ii[9:]).all()        # sm.dat missing -> NaN span
+        assert rec.base_datetime.year == 2100
+
+    def test_channel_selection_on_layout(self, tmp_path):
+        rec = self._mk_multiseg(str(tmp_path))(channel_names=["V"])
+        assert rec.sig_name == ["V"]
+        np.testing.assert_allclose(rec.p_signal[7:9, 0], [2.0, 4.0])
+
+    def test_reference_record(self, reference_dir):
+        import os
+        path = os.path.join(
+            reference_dir, "data/waveform/physionet.org/files/"
+            "mimic3wdb-matched/1.0/p00/p000194/p000194-2112-05-23-14-34")
+        if not os.path.exists(path + ".hea"):
+            pytest.skip("reference multi-segment record missing")
+        from tskd_amd.io import rdrecord
+        rec = rdrecord(path)
+        assert rec.n_seg == 7 and rec.fs == 125.0
+        assert rec.sig_name == ["II"] and rec.sig_len == 12192771
+        a = rec.p_signal[:, 0]
+        # exactly the three mirrored segments (0002/0003/0004) are finite
+        assert int(np.isfinite(a).sum()) == 384 + 1449229 + 255000
+        assert np.isnan(a[:15060]).all()          # leading gap segment
+        s0 = 15060 + 10467887                      # 0001.dat not mirrored
+        # fmt-80 decode of segment 0002: initval 17, gain 11
+        np.testing.assert_allclose(a[s0], 17 / 11)

@@ -6,11 +6,19 @@
 // with WFDB invalid-sample sentinels mapped to NaN. Feeds the replay
 // producer (reference sendStream.py:46 wfdb.rdrecord).
 //
+// Multi-segment records (the MIMIC waveform layout, e.g. reference record
+// p000194-2112-05-23-14-34.hea: `<record>/<nseg>` master line, `~ <n>` gap
+// segments, a `<name>_layout 0` variable-layout header) are stitched into
+// one (sig_len, n_sig) array on the layout's channel set; gap segments and
+// segments whose .dat is absent (partially mirrored directories) read as
+// NaN, matching the NaN-for-missing convention of the numerics path.
+//
 // Header grammar handled (see reference record
 // p000194-2112-05-23-14-34n.hea):
-//   <record> <nsig> <fs>[/counter] [<nsamp> [<base_time> [<base_date>]]]
+//   <record>[/<nseg>] <nsig> <fs>[/counter] [<nsamp> [<base_time> [<base_date>]]]
 //   <file> <fmt>[x][:][+] <gain>[(baseline)][/units] <adcres> <adczero>
 //          <initval> <checksum> <blocksize> <description...>
+//   (multi-segment body): <segment_name|~> <nsamp>
 
 #include <pybind11/numpy.h>
 #include <pybind11/pybind11.h>
@@ -42,13 +50,20 @@ struct SigSpec {
     std::string desc;
 };
 
+struct SegSpec {
+    std::string name;  // "~" = gap
+    long nsamp = 0;
+};
+
 struct Header {
     std::string record;
     int nsig = 0;
+    int nseg = 0;  // 0 = single-segment
     double fs = 250.0;
     long nsamp = 0;
     std::string base_time, base_date;
     std::vector<SigSpec> sigs;
+    std::vector<SegSpec> segs;
 };
 
 std::string dirname_of(const std::string& p) {
@@ -73,14 +88,23 @@ Header parse_header(const std::string& hea_path) {
             first = false;
             std::string fs_tok;
             ss >> h.record >> h.nsig >> fs_tok >> h.nsamp;
-            // record may carry /nseg (multi-segment) — reject those here
+            // record may carry /nseg (multi-segment master header)
             auto slash = h.record.find('/');
-            if (slash != std::string::npos)
-                throw std::runtime_error(
-                    "wfdb: multi-segment records not supported: " + h.record);
+            if (slash != std::string::npos) {
+                h.nseg = atoi(h.record.substr(slash + 1).c_str());
+                h.record = h.record.substr(0, slash);
+            }
             if (!fs_tok.empty())
                 h.fs = atof(fs_tok.substr(0, fs_tok.find('/')).c_str());
             ss >> h.base_time >> h.base_date;
+            continue;
+        }
+        if (h.nseg > 0) {
+            // segment list: <name|~> <nsamp>
+            if ((int)h.segs.size() >= h.nseg) break;
+            SegSpec sg;
+            ss >> sg.name >> sg.nsamp;
+            h.segs.push_back(sg);
             continue;
         }
         if ((int)h.sigs.size() >= h.nsig) break;
@@ -112,8 +136,13 @@ Header parse_header(const std::string& hea_path) {
         if (!s.baseline_set) s.baseline = s.adczero;
         h.sigs.push_back(s);
     }
-    if ((int)h.sigs.size() != h.nsig)
+    if (h.nseg > 0) {
+        if ((int)h.segs.size() != h.nseg)
+            throw std::runtime_error("wfdb: segment count mismatch in " +
+                                     hea_path);
+    } else if ((int)h.sigs.size() != h.nsig) {
         throw std::runtime_error("wfdb: signal count mismatch in " + hea_path);
+    }
     return h;
 }
 
@@ -172,6 +201,46 @@ std::vector<int32_t> decode_dat(const std::vector<uint8_t>& raw, int fmt,
     return out;
 }
 
+// Decode every signal of single-segment header `h` (its .dat files live in
+// `dir`) into phys rows [t0, t0 + n) where n = min(h.nsamp, nmax).
+// col_of[i] = output column for h.sigs[i] (-1 drops the signal). When
+// `missing_ok`, an absent .dat leaves its span NaN instead of throwing.
+void decode_into(const Header& h, const std::string& dir,
+                 const std::vector<int>& col_of,
+                 std::vector<std::vector<double>>& phys, long t0, long nmax,
+                 bool missing_ok) {
+    const long n = std::min(h.nsamp, nmax);
+    for (int i = 0; i < h.nsig;) {
+        const std::string file = h.sigs[i].file;
+        std::vector<int> group;
+        for (int j = 0; j < h.nsig; ++j)
+            if (h.sigs[j].file == file) group.push_back(j);
+        while (i < h.nsig && h.sigs[i].file == file) ++i;
+        bool wanted = false;
+        for (int g : group) wanted |= col_of[g] >= 0;
+        if (!wanted) continue;
+        std::vector<uint8_t> raw;
+        try {
+            raw = read_file(dir + "/" + file);
+        } catch (const std::runtime_error&) {
+            if (missing_ok) continue;  // span stays NaN
+            throw;
+        }
+        const int fmt = h.sigs[group[0]].fmt;
+        auto adc = decode_dat(raw, fmt, (int)group.size(), n);
+        for (size_t gi = 0; gi < group.size(); ++gi) {
+            const SigSpec& s = h.sigs[group[gi]];
+            const int col = col_of[group[gi]];
+            if (col < 0) continue;
+            for (long t = 0; t < n; ++t) {
+                const int32_t a = adc[(size_t)t * group.size() + gi];
+                phys[col][t0 + t] =
+                    (a == INT32_MIN) ? NAN : (a - s.baseline) / s.gain;
+            }
+        }
+    }
+}
+
 }  // namespace
 
 // rdrecord: read header + signals; optionally select channels by name
@@ -182,43 +251,68 @@ py::dict rdrecord(const std::string& record_path,
     Header h = parse_header(record_path + ".hea");
     const std::string dir = dirname_of(record_path);
 
-    // group signals by file (multiplexed storage)
-    std::vector<int> col_of(h.nsig);
-    std::vector<std::vector<double>> phys(h.nsig,
-                                          std::vector<double>(h.nsamp, NAN));
-    // decode per distinct file
-    for (int i = 0; i < h.nsig;) {
-        const std::string file = h.sigs[i].file;
-        std::vector<int> group;
-        for (int j = 0; j < h.nsig; ++j)
-            if (h.sigs[j].file == file) group.push_back(j);
-        const int fmt = h.sigs[group[0]].fmt;
-        auto raw = read_file(dir + "/" + file);
-        auto adc = decode_dat(raw, fmt, (int)group.size(), h.nsamp);
-        for (size_t gi = 0; gi < group.size(); ++gi) {
-            const SigSpec& s = h.sigs[group[gi]];
-            for (long t = 0; t < h.nsamp; ++t) {
-                const int32_t a = adc[(size_t)t * group.size() + gi];
-                phys[group[gi]][t] =
-                    (a == INT32_MIN) ? NAN : (a - s.baseline) / s.gain;
+    // Canonical signal list: the record's own for single-segment; the
+    // layout segment's (or first real segment's, fixed-layout) otherwise.
+    std::vector<SigSpec> canon;
+    if (h.nseg == 0) {
+        canon = h.sigs;
+    } else {
+        std::string layout_name;
+        for (auto& sg : h.segs)
+            if (sg.name != "~" && sg.nsamp == 0) { layout_name = sg.name; break; }
+        if (layout_name.empty())
+            for (auto& sg : h.segs)
+                if (sg.name != "~") { layout_name = sg.name; break; }
+        if (layout_name.empty())
+            throw std::runtime_error("wfdb: multi-segment record " + h.record +
+                                     " has no readable segment");
+        canon = parse_header(dir + "/" + layout_name + ".hea").sigs;
+    }
+    const int ncanon = (int)canon.size();
+
+    std::vector<std::vector<double>> phys(
+        ncanon, std::vector<double>(h.nsamp, NAN));
+    if (h.nseg == 0) {
+        std::vector<int> ident(ncanon);
+        for (int i = 0; i < ncanon; ++i) ident[i] = i;
+        decode_into(h, dir, ident, phys, 0, h.nsamp, /*missing_ok=*/false);
+    } else {
+        long t0 = 0;
+        for (auto& sg : h.segs) {
+            if (t0 >= h.nsamp) break;
+            if (sg.name == "~" || sg.nsamp == 0) { t0 += sg.nsamp; continue; }
+            Header seg;
+            try {
+                seg = parse_header(dir + "/" + sg.name + ".hea");
+            } catch (const std::runtime_error&) {
+                t0 += sg.nsamp;  // unmirrored segment header -> NaN span
+                continue;
             }
+            // map segment signals onto canonical columns by description
+            std::vector<int> col_of(seg.nsig, -1);
+            for (int i = 0; i < seg.nsig; ++i)
+                for (int c = 0; c < ncanon; ++c)
+                    if (seg.sigs[i].desc == canon[c].desc) {
+                        col_of[i] = c;
+                        break;
+                    }
+            decode_into(seg, dir, col_of, phys, t0,
+                        std::min(sg.nsamp, h.nsamp - t0), /*missing_ok=*/true);
+            t0 += sg.nsamp;
         }
-        // advance past processed signals
-        while (i < h.nsig && h.sigs[i].file == file) ++i;
-        // (non-contiguous same-file groups already handled above)
     }
 
     // channel selection by description
     std::vector<int> sel;
     std::vector<std::string> names;
     if (channel_names.empty()) {
-        for (int i = 0; i < h.nsig; ++i) sel.push_back(i);
+        for (int i = 0; i < ncanon; ++i) sel.push_back(i);
     } else {
         for (auto& want : channel_names)
-            for (int i = 0; i < h.nsig; ++i)
-                if (h.sigs[i].desc == want) { sel.push_back(i); break; }
+            for (int i = 0; i < ncanon; ++i)
+                if (canon[i].desc == want) { sel.push_back(i); break; }
     }
-    for (int i : sel) names.push_back(h.sigs[i].desc);
+    for (int i : sel) names.push_back(canon[i].desc);
 
     py::array_t<double> p_signal({(py::ssize_t)h.nsamp,
                                   (py::ssize_t)sel.size()});
@@ -234,8 +328,8 @@ py::dict rdrecord(const std::string& record_path,
     out["sig_name"] = names;
     py::list units, gains;
     for (int i : sel) {
-        units.append(h.sigs[i].units);
-        gains.append(h.sigs[i].gain);
+        units.append(canon[i].units);
+        gains.append(canon[i].gain);
     }
     out["units"] = units;
     out["gain"] = gains;
@@ -243,11 +337,12 @@ py::dict rdrecord(const std::string& record_path,
     out["base_time"] = h.base_time;
     out["base_date"] = h.base_date;
     out["sig_len"] = h.nsamp;
+    out["n_seg"] = h.nseg;
     return out;
 }
 
 PYBIND11_MODULE(_tskd_wfdb, m) {
-    m.doc() = "tskd first-party WFDB reader (fmt 16/80/212)";
+    m.doc() = "tskd first-party WFDB reader (fmt 16/80/212, multi-segment)";
     m.def("rdrecord", &rdrecord, py::arg("record_path"),
           py::arg("channel_names") = std::vector<std::string>{});
 }

@@ -30,6 +30,7 @@ class Record:
     base_time: str = ""
     base_date: str = ""
     sig_len: int = 0
+    n_seg: int = 0  # 0 = single-segment record
     extra: dict = field(default_factory=dict)
 
     @property
@@ -55,7 +56,7 @@ def rdrecord(record_path: str,
         sig_name=list(d["sig_name"]), units=list(d["units"]),
         gain=list(d["gain"]), p_signal=np.asarray(d["p_signal"]),
         base_time=d["base_time"], base_date=d["base_date"],
-        sig_len=d["sig_len"])
+        sig_len=d["sig_len"], n_seg=d.get("n_seg", 0))
 
 
 def get_waveform_path(record_name: str,
