@@ -158,6 +158,11 @@ class TestDashboard:
         preds = client.get("/api/predictions").json()
         assert preds and preds[0]["patient"] == "p000194"
         assert abs(preds[0]["risk"] - 0.42) < 1e-6
+        # Prometheus text exposition
+        body = client.get("/metrics").text
+        assert "tskd_stage_calls_total{stage=\"plotdata.pump\"}" in body
+        assert 'tskd_stage_latency_ms{stage="plotdata.pump",quantile="p50"}' \
+            in body
         h = client.get("/health").json()
         assert h["status"] == "ok" and h["predictions"] == 1
 

@@ -154,3 +154,14 @@ class TestAuxCLIs:
             seen.add(ci)
             assert isinstance(val, float)
         assert seen == set(range(cfg.n_channels))
+
+    def test_prometheus_text(self):
+        from tskd_amd.metrics import PipelineMetrics, prometheus_text
+        pm = PipelineMetrics()
+        with pm.stage("ingest"):
+            pass
+        pm.stage("ingest").add_items(7)
+        body = prometheus_text(pm)
+        assert body.endswith("\n")
+        assert 'tskd_stage_items_total{stage="ingest"} 7' in body
+        assert "# TYPE tskd_stage_latency_ms gauge" in body

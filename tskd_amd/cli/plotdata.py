@@ -75,6 +75,8 @@ class DashState:
         self.lock = threading.Lock()  # the reference's single mutex
         self.raw = defaultdict(lambda: defaultdict(lambda: deque(maxlen=keep)))
         self.proc = defaultdict(lambda: defaultdict(lambda: deque(maxlen=keep)))
+        from tskd_amd.metrics import StageTimer
+        self.pump_timer = StageTimer("plotdata.pump")
         self.raw_consumer = Consumer(bus, starting=starting)
         topics = [cfg.topic_for_channel(c) for c in cfg.channel_names]
         for t in topics:
@@ -87,24 +89,28 @@ class DashState:
 
     def pump(self) -> None:
         while not self._stop:
-            msgs = self.raw_consumer.poll(max_msgs=4096, timeout_ms=200)
-            pmsgs = self.proc_consumer.poll(max_msgs=4096, timeout_ms=0)
-            with self.lock:
-                for m in msgs:
-                    try:
-                        chan, val = json.loads(m.value)
-                    except (ValueError, TypeError):
-                        continue
-                    self.raw[m.key.decode()][int(chan)].append(float(val))
-                for m in pmsgs:
-                    pid, _, chan_s = m.key.decode().rpartition("_")
-                    try:
-                        pts = json.loads(m.value)
-                        self.proc[pid][int(chan_s)].extend(
-                            float(p) for p in pts)
-                    except (ValueError, TypeError):
-                        continue
-            time.sleep(0.05)
+            with self.pump_timer:
+                self._pump_once()
+
+    def _pump_once(self) -> None:
+        msgs = self.raw_consumer.poll(max_msgs=4096, timeout_ms=200)
+        pmsgs = self.proc_consumer.poll(max_msgs=4096, timeout_ms=0)
+        with self.lock:
+            for m in msgs:
+                try:
+                    chan, val = json.loads(m.value)
+                except (ValueError, TypeError):
+                    continue
+                self.raw[m.key.decode()][int(chan)].append(float(val))
+            for m in pmsgs:
+                pid, _, chan_s = m.key.decode().rpartition("_")
+                try:
+                    pts = json.loads(m.value)
+                    self.proc[pid][int(chan_s)].extend(
+                        float(p) for p in pts)
+                except (ValueError, TypeError):
+                    continue
+        time.sleep(0.05)
 
 
 def build_app(state: DashState):
@@ -122,6 +128,14 @@ def build_app(state: DashState):
         # the reference's db healthcheck analog (docker-compose.yml:138-140)
         return {"status": "ok", "predictions": state.store.count(),
                 "patients": len(set(state.raw) | set(state.proc))}
+
+    @app.get("/metrics")
+    def metrics():
+        # Prometheus text exposition (scrape target; reference has none)
+        from fastapi.responses import PlainTextResponse
+        from tskd_amd.metrics import prometheus_text
+        return PlainTextResponse(prometheus_text([state.pump_timer]),
+                                 media_type="text/plain; version=0.0.4")
 
     @app.get("/api/patients")
     def patients():

@@ -73,3 +73,28 @@ class PipelineMetrics:
 
     def report(self) -> str:
         return json.dumps({k: v.snapshot() for k, v in self.stages.items()})
+
+
+def prometheus_text(timers) -> str:
+    """Render StageTimer snapshots in the Prometheus text exposition format
+    (the reference has no scrapeable metrics — SURVEY.md §5 observability).
+    `timers` is an iterable of StageTimer (or a PipelineMetrics)."""
+    if isinstance(timers, PipelineMetrics):
+        timers = timers.stages.values()
+    lines = [
+        "# HELP tskd_stage_calls_total Stage invocations.",
+        "# TYPE tskd_stage_calls_total counter",
+        "# HELP tskd_stage_items_total Items (windows/messages) processed.",
+        "# TYPE tskd_stage_items_total counter",
+        "# HELP tskd_stage_latency_ms Stage latency quantiles (rolling).",
+        "# TYPE tskd_stage_latency_ms gauge",
+    ]
+    for t in timers:
+        snap = t.snapshot()
+        lbl = f'stage="{snap["stage"]}"'
+        lines.append(f'tskd_stage_calls_total{{{lbl}}} {snap["calls"]}')
+        lines.append(f'tskd_stage_items_total{{{lbl}}} {snap["items"]}')
+        for q in ("p50", "p99"):
+            lines.append(f'tskd_stage_latency_ms{{{lbl},quantile="{q}"}} '
+                         f'{snap[q + "_ms"]:.6f}')
+    return "\n".join(lines) + "\n"
