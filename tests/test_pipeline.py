@@ -378,3 +378,26 @@ class TestFusedServe:
         assert srv.trigger() == 1
         _, risk2 = store.latest("p000194")
         assert abs(risk1 - risk2) > 1e-6  # different weights, different score
+
+
+class TestMetricsServer:
+    def test_prometheus_scrape(self):
+        import urllib.request
+        from tskd_amd.cli.serve import start_metrics_server
+        from tskd_amd.metrics import StageTimer
+        t = StageTimer("serve")
+        with t:
+            pass
+        t.add_items(3)
+        srv = start_metrics_server([t], 0)  # port 0 = ephemeral
+        port = srv.server_address[1]
+        try:
+            body = urllib.request.urlopen(
+                f"http://127.0.0.1:{port}/metrics", timeout=5).read().decode()
+            assert 'tskd_stage_items_total{stage="serve"} 3' in body
+            import urllib.error
+            with pytest.raises(urllib.error.HTTPError):
+                urllib.request.urlopen(f"http://127.0.0.1:{port}/nope",
+                                       timeout=5)
+        finally:
+            srv.shutdown()

@@ -204,6 +204,36 @@ class FusedServer:
             return n_active
 
 
+
+def start_metrics_server(timers, port: int):
+    """Serve the Prometheus text exposition on 127.0.0.1:port/metrics in a
+    daemon thread (stdlib only; scrape target for production serving)."""
+    import threading
+    from http.server import BaseHTTPRequestHandler, HTTPServer
+
+    from tskd_amd.metrics import prometheus_text
+
+    class H(BaseHTTPRequestHandler):
+        def do_GET(self):
+            if self.path != "/metrics":
+                self.send_response(404)
+                self.end_headers()
+                return
+            body = prometheus_text(timers).encode()
+            self.send_response(200)
+            self.send_header("Content-Type", "text/plain; version=0.0.4")
+            self.send_header("Content-Length", str(len(body)))
+            self.end_headers()
+            self.wfile.write(body)
+
+        def log_message(self, *a):  # quiet
+            pass
+
+    srv = HTTPServer(("127.0.0.1", port), H)
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    return srv
+
+
 def main(argv=None) -> None:
     logging.basicConfig(level=logging.INFO,
                         format="%(asctime)s %(name)s %(levelname)s %(message)s")
@@ -226,6 +256,8 @@ def main(argv=None) -> None:
     ap.add_argument("--max-triggers", type=int, default=0)
     ap.add_argument("--hot-reload", action="store_true",
                     help="reload the checkpoint when the file changes")
+    ap.add_argument("--metrics-port", type=int, default=0,
+                    help="expose Prometheus /metrics on 127.0.0.1:PORT")
     args = ap.parse_args(argv)
 
     rank, world = init_distributed()
@@ -249,6 +281,11 @@ def main(argv=None) -> None:
         srv.model_path = args.model_path
         if os.path.exists(args.model_path):
             srv._model_mtime = os.path.getmtime(args.model_path)
+    metrics_srv = None
+    if args.metrics_port:
+        metrics_srv = start_metrics_server([srv.timer],
+                                           args.metrics_port + rank)
+        log.info("metrics on 127.0.0.1:%d/metrics", args.metrics_port + rank)
     period = cfg.predict_slide_s / args.speed
     stop = []
     signal.signal(signal.SIGTERM, lambda *a: stop.append(1))
