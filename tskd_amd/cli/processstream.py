@@ -126,6 +126,8 @@ def main(argv=None) -> None:
     ap.add_argument("--max-streams", type=int, default=64)
     ap.add_argument("--starting", default="latest",
                     choices=["latest", "earliest"])
+    ap.add_argument("--metrics-port", type=int, default=0,
+                    help="expose Prometheus /metrics on 127.0.0.1:PORT")
     ap.add_argument("--max-triggers", type=int, default=0,
                     help="stop after N triggers (0 = run forever)")
     ap.add_argument("--offsets-file", default=None,
@@ -154,6 +156,10 @@ def main(argv=None) -> None:
                      args.offsets_file)
     from tskd_amd.metrics import StageTimer
     timer = StageTimer(ap.prog or "stage")
+    if args.metrics_port:
+        from tskd_amd.cli.serve import start_metrics_server
+        start_metrics_server([timer], args.metrics_port)
+        log.info("metrics on 127.0.0.1:%d/metrics", args.metrics_port)
     n = 0
     while not stop:
         t0 = time.time()
