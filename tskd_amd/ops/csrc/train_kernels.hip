@@ -112,7 +112,7 @@ __device__ __forceinline__ float drop_mult_(float p, unsigned long long seed,
 // ---------------------------------------------------------------------------
 // conv forward with stash (fp32; one wave per window, 4 waves per block)
 // ---------------------------------------------------------------------------
-template <class G>
+template <class G, bool STAGE_X = true>
 __global__ __launch_bounds__(256) void train_conv_fwd_kernel(
     const float* __restrict__ x,      // (SN, CIN, L)
     float* __restrict__ feat,         // (SN, LIN)
@@ -122,7 +122,9 @@ __global__ __launch_bounds__(256) void train_conv_fwd_kernel(
 {
     constexpr int NW = 4 * G::CIN * G::K1 + 4 + 20 + 1;
     __shared__ float lw[NW];
-    __shared__ float lx[4][G::CIN * G::L];
+    // lx only exists in the staged instantiation (19 KB of LDS otherwise
+    // wasted against occupancy)
+    __shared__ float lx[STAGE_X ? 4 : 1][STAGE_X ? G::CIN * G::L : 1];
     __shared__ float lp1[4][4 * G::P1];
     for (int i = threadIdx.x; i < NW; i += 256) lw[i] = wpack[i];
     __syncthreads();
@@ -142,9 +144,19 @@ __global__ __launch_bounds__(256) void train_conv_fwd_kernel(
         int* i2 = (int*)(st + G::SC_I2);
         float* m1 = st + G::SC_M1;
         float* m2 = st + G::SC_M2;
-        float* xw = lx[wave];
-        for (int i = lane; i < G::CIN * G::L; i += WAVE) xw[i] = xin[i];
-        twsync();
+        // STAGE_X=false reads the window straight from L1-resident global
+        // memory (PMC: the LDS-staged variant spends 32% of wave cycles in
+        // s_waitcnt lgkmcnt — the same lesson as the inference tlast conv);
+        // the backward pass still stages (it sweeps x twice).
+        const float* xw;
+        if (STAGE_X) {
+            float* xs = lx[STAGE_X ? wave : 0];
+            for (int i = lane; i < G::CIN * G::L; i += WAVE) xs[i] = xin[i];
+            twsync();
+            xw = xs;
+        } else {
+            xw = xin;
+        }
         for (int o = lane; o < 4 * G::C1; o += WAVE) {
             const int c = o / G::C1, s = o % G::C1;
             // two independent partial accumulators (even/odd input channel)
@@ -515,7 +527,9 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
     float* __restrict__ grads, int SN)
 {
     __shared__ float lw2[21];  // conv2 weights + bias
-    __shared__ float lx[4][G::CIN * G::L];
+    // row stride L+1: L=120 dwords is 24 mod 32 banks (partial conflicts
+    // on the stride-L xw reads in the conv1-grad loop); +1 spreads banks
+    __shared__ float lx[4][G::CIN * (G::L + 1)];
     __shared__ float lp1[4][4 * G::P1];
     __shared__ float lda1[4][4 * G::C1];
     __shared__ float lda2[4][G::C2];
@@ -546,7 +560,8 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
         const float* dfw = dfeat + win * G::LIN;
         const float* xin = x + win * (G::CIN * G::L);
         float* xw = lx[wave];
-        for (int i = lane; i < G::CIN * G::L; i += WAVE) xw[i] = xin[i];
+        for (int i = lane; i < G::CIN * G::L; i += WAVE)
+            xw[i + i / G::L] = xin[i];  // padded-row store
         // recompute the (dropout-masked) pool1 output = conv2's input
         for (int o = lane; o < 4 * G::P1; o += WAVE) {
             const int c = o / G::P1, q = o % G::P1;
@@ -627,7 +642,7 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
             const int i = (o / G::K1) % G::CIN;
             const int k = o % G::K1;
             const float* da = lda1[wave] + c * G::C1;
-            const float* xr = xw + i * G::L + k;
+            const float* xr = xw + i * (G::L + 1) + k;
             float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
             int s = 0;
             for (; s + 4 <= G::C1; s += 4) {
@@ -708,8 +723,15 @@ int conv_fwd(const float* x, float* feat, float* stash, const float* wpack,
              hipStream_t s) {
     if (SN <= 0) return 0;
     int grid = min((SN + 3) / 4, 8192);
-    hipLaunchKernelGGL((train_conv_fwd_kernel<G>), dim3(grid), dim3(256), 0,
-                       s, x, feat, stash, wpack, SN, d1, d2, seed);
+    const char* sx = getenv("TSKD_TRAIN_STAGE_X");
+    if (sx && sx[0] == '1')
+        hipLaunchKernelGGL((train_conv_fwd_kernel<G, true>), dim3(grid),
+                           dim3(256), 0, s, x, feat, stash, wpack, SN, d1, d2,
+                           seed);
+    else
+        hipLaunchKernelGGL((train_conv_fwd_kernel<G, false>), dim3(grid),
+                           dim3(256), 0, s, x, feat, stash, wpack, SN, d1, d2,
+                           seed);
     return (int)hipGetLastError();
 }
 template <class G>
