@@ -723,13 +723,16 @@ int conv_fwd(const float* x, float* feat, float* stash, const float* wpack,
              hipStream_t s) {
     if (SN <= 0) return 0;
     int grid = min((SN + 3) / 4, 8192);
+    // Measured: the LDS-staged variant beats direct-global reads by ~8%
+    // here (no transpose cost, and the dependent FMA chain prefers LDS
+    // latency) — the OPPOSITE of the inference tlast conv. Default staged.
     const char* sx = getenv("TSKD_TRAIN_STAGE_X");
-    if (sx && sx[0] == '1')
-        hipLaunchKernelGGL((train_conv_fwd_kernel<G, true>), dim3(grid),
+    if (sx && sx[0] == '0')
+        hipLaunchKernelGGL((train_conv_fwd_kernel<G, false>), dim3(grid),
                            dim3(256), 0, s, x, feat, stash, wpack, SN, d1, d2,
                            seed);
     else
-        hipLaunchKernelGGL((train_conv_fwd_kernel<G, false>), dim3(grid),
+        hipLaunchKernelGGL((train_conv_fwd_kernel<G, true>), dim3(grid),
                            dim3(256), 0, s, x, feat, stash, wpack, SN, d1, d2,
                            seed);
     return (int)hipGetLastError();
