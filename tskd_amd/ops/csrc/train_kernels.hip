@@ -518,7 +518,7 @@ __global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
 // ---------------------------------------------------------------------------
 // conv backward: dfeat -> conv/bias grads (atomics). One wave per window.
 // ---------------------------------------------------------------------------
-template <class G>
+template <class G, int XP = 1>  // XP: lx row pad (bank spread A/B)
 __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
     const float* __restrict__ x,        // (SN, CIN, L)
     const float* __restrict__ stash,    // (SN, SC_SIZE)
@@ -529,7 +529,7 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
     __shared__ float lw2[21];  // conv2 weights + bias
     // row stride L+1: L=120 dwords is 24 mod 32 banks (partial conflicts
     // on the stride-L xw reads in the conv1-grad loop); +1 spreads banks
-    __shared__ float lx[4][G::CIN * (G::L + 1)];
+    __shared__ float lx[4][G::CIN * (G::L + XP)];
     __shared__ float lp1[4][4 * G::P1];
     __shared__ float lda1[4][4 * G::C1];
     __shared__ float lda2[4][G::C2];
@@ -561,7 +561,7 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
         const float* xin = x + win * (G::CIN * G::L);
         float* xw = lx[wave];
         for (int i = lane; i < G::CIN * G::L; i += WAVE)
-            xw[i + i / G::L] = xin[i];  // padded-row store
+            xw[i + XP * (i / G::L)] = xin[i];  // padded-row store
         // recompute the (dropout-masked) pool1 output = conv2's input
         for (int o = lane; o < 4 * G::P1; o += WAVE) {
             const int c = o / G::P1, q = o % G::P1;
@@ -642,7 +642,7 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
             const int i = (o / G::K1) % G::CIN;
             const int k = o % G::K1;
             const float* da = lda1[wave] + c * G::C1;
-            const float* xr = xw + i * (G::L + 1) + k;
+            const float* xr = xw + i * (G::L + XP) + k;
             float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
             int s = 0;
             for (; s + 4 <= G::C1; s += 4) {
@@ -764,8 +764,13 @@ int conv_bwd(const float* x, const float* stash, const float* dfeat,
     int cap = 2048;
     if (const char* e = getenv("TSKD_CONVBWD_GRID")) cap = atoi(e);
     int grid = min((SN + 3) / 4, cap);
-    hipLaunchKernelGGL((train_conv_bwd_kernel<G>), dim3(grid), dim3(256), 0,
-                       s, x, stash, dfeat, wpack, grads, SN);
+    const char* xp = getenv("TSKD_CONVBWD_PAD");
+    if (xp && xp[0] == '0')
+        hipLaunchKernelGGL((train_conv_bwd_kernel<G, 0>), dim3(grid),
+                           dim3(256), 0, s, x, stash, dfeat, wpack, grads, SN);
+    else
+        hipLaunchKernelGGL((train_conv_bwd_kernel<G, 1>), dim3(grid),
+                           dim3(256), 0, s, x, stash, dfeat, wpack, grads, SN);
     return (int)hipGetLastError();
 }
 }  // namespace
