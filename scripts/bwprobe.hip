@@ -1,0 +1,63 @@
+// HBM read-bandwidth ceiling probe (calibration tool, not a product kernel).
+//
+// Measures the best-achievable pure-read rate on gfx950 across load widths,
+// nontemporal hints and grid sizes, to price the pipeline's ingest stage
+// against the REAL platform ceiling (profiles/r01: ingest 4.7 TB/s vs ATen
+// copy 4.62 TB/s — is there more?).
+//
+// Build: hipcc --offload-arch=gfx950 -O3 -shared -fPIC scripts/bwprobe.hip
+//        -o scripts/_bwprobe.so
+#include <hip/hip_runtime.h>
+
+typedef float f32x4_ __attribute__((ext_vector_type(4)));
+
+// Each variant sums `n` floats (16 B-aligned base) and folds the result to
+// out[block] so the loads cannot be optimized away.
+template <int VEC, bool NT>
+__global__ void read_probe(const float* __restrict__ p, long n4,
+                           float* __restrict__ out) {
+    const f32x4_* v = (const f32x4_*)p;
+    f32x4_ acc = {0.f, 0.f, 0.f, 0.f};
+    const long stride = (long)gridDim.x * blockDim.x;
+    long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    for (; i + (VEC - 1) * stride < n4; i += VEC * stride) {
+        #pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+            f32x4_ t = NT ? __builtin_nontemporal_load(&v[i + j * stride])
+                          : v[i + j * stride];
+            acc += t;
+        }
+    }
+    for (; i < n4; i += stride) acc += NT
+        ? __builtin_nontemporal_load(&v[i]) : v[i];
+    float s = acc.x + acc.y + acc.z + acc.w;
+    __shared__ float red[256];
+    red[threadIdx.x] = s;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float t = 0.f;
+        for (int j = 0; j < 256; ++j) t += red[j];
+        out[blockIdx.x] = t;
+    }
+}
+
+extern "C" int bw_probe(const float* p, long n, float* out, int variant,
+                        int grid, void* stream) {
+    const long n4 = n / 4;
+    hipStream_t st = (hipStream_t)stream;
+    switch (variant) {
+        case 0: hipLaunchKernelGGL((read_probe<1, true>), dim3(grid),
+                                   dim3(256), 0, st, p, n4, out); break;
+        case 1: hipLaunchKernelGGL((read_probe<2, true>), dim3(grid),
+                                   dim3(256), 0, st, p, n4, out); break;
+        case 2: hipLaunchKernelGGL((read_probe<4, true>), dim3(grid),
+                                   dim3(256), 0, st, p, n4, out); break;
+        case 3: hipLaunchKernelGGL((read_probe<1, false>), dim3(grid),
+                                   dim3(256), 0, st, p, n4, out); break;
+        case 4: hipLaunchKernelGGL((read_probe<2, false>), dim3(grid),
+                                   dim3(256), 0, st, p, n4, out); break;
+        case 5: hipLaunchKernelGGL((read_probe<4, false>), dim3(grid),
+                                   dim3(256), 0, st, p, n4, out); break;
+    }
+    return (int)hipGetLastError();
+}
