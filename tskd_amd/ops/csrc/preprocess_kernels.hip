@@ -75,7 +75,7 @@ __device__ __forceinline__ void wsync_() {
 // busy (78/16 = 4.9 balanced iterations per group).
 #define ING_GRP 4   // buckets per wave
 #define ING_GL 16   // lanes per bucket group
-template <class DT>
+template <class DT, bool ILP = true>
 __global__ void ingest_dense_kernel(
     const DT* __restrict__ raw,     // (S, CIN, T)
     float* __restrict__ bsum,       // (S, C, G)
@@ -114,14 +114,41 @@ __global__ void ingest_dense_kernel(
                 }
                 const int oct = (bucket_len - pre) / 8;
                 const u32x4_* vp = (const u32x4_*)((const unsigned short*)src + pre);
-                for (int p = lane; p < oct; p += ING_GL) {
-                    union { u32x4_ q; unsigned short h[8]; } v;
-                    // raw samples are consumed exactly once: stream past L2
-                    v.q = __builtin_nontemporal_load(&vp[p]);
-                    #pragma unroll
-                    for (int j = 0; j < 8; ++j) {
-                        const float f = bf16_to_f32_(v.h[j]);
-                        if (!isnan(f)) { sum += f; cnt += 1.f; }
+                if constexpr (ILP) {
+                    // 5 independent nontemporal loads in flight per lane
+                    // (read-ceiling probe: 1-deep streams measure ~5.9 TB/s,
+                    // 4..5-deep ~7.0 TB/s). Out-of-range lanes re-load chunk
+                    // 0 (clamped index) so the loads issue unconditionally,
+                    // and their contribution is masked after.
+                    for (int base = lane; base < oct; base += 5 * ING_GL) {
+                        union { u32x4_ q; unsigned short h[8]; } v[5];
+                        #pragma unroll
+                        for (int u = 0; u < 5; ++u) {
+                            const int pp = base + u * ING_GL;
+                            v[u].q = __builtin_nontemporal_load(
+                                &vp[pp < oct ? pp : 0]);
+                        }
+                        #pragma unroll
+                        for (int u = 0; u < 5; ++u) {
+                            if (base + u * ING_GL < oct) {
+                                #pragma unroll
+                                for (int j = 0; j < 8; ++j) {
+                                    const float f = bf16_to_f32_(v[u].h[j]);
+                                    if (!isnan(f)) { sum += f; cnt += 1.f; }
+                                }
+                            }
+                        }
+                    }
+                } else {
+                    for (int p = lane; p < oct; p += ING_GL) {
+                        union { u32x4_ q; unsigned short h[8]; } v;
+                        // raw samples are consumed once: stream past L2
+                        v.q = __builtin_nontemporal_load(&vp[p]);
+                        #pragma unroll
+                        for (int j = 0; j < 8; ++j) {
+                            const float f = bf16_to_f32_(v.h[j]);
+                            if (!isnan(f)) { sum += f; cnt += 1.f; }
+                        }
                     }
                 }
                 for (int i = pre + oct * 8 + lane; i < bucket_len; i += ING_GL) {
@@ -137,12 +164,33 @@ __global__ void ingest_dense_kernel(
                 }
                 const int quad = (bucket_len - pre) / 4;
                 const f32x4_* vp = (const f32x4_*)((const float*)src + pre);
-                for (int p = lane; p < quad; p += ING_GL) {
-                    const f32x4_ v = __builtin_nontemporal_load(&vp[p]);
-                    if (!isnan(v.x)) { sum += v.x; cnt += 1.f; }
-                    if (!isnan(v.y)) { sum += v.y; cnt += 1.f; }
-                    if (!isnan(v.z)) { sum += v.z; cnt += 1.f; }
-                    if (!isnan(v.w)) { sum += v.w; cnt += 1.f; }
+                if constexpr (ILP) {
+                    for (int base = lane; base < quad; base += 5 * ING_GL) {
+                        f32x4_ v[5];
+                        #pragma unroll
+                        for (int u = 0; u < 5; ++u) {
+                            const int pp = base + u * ING_GL;
+                            v[u] = __builtin_nontemporal_load(
+                                &vp[pp < quad ? pp : 0]);
+                        }
+                        #pragma unroll
+                        for (int u = 0; u < 5; ++u) {
+                            if (base + u * ING_GL < quad) {
+                                if (!isnan(v[u].x)) { sum += v[u].x; cnt += 1.f; }
+                                if (!isnan(v[u].y)) { sum += v[u].y; cnt += 1.f; }
+                                if (!isnan(v[u].z)) { sum += v[u].z; cnt += 1.f; }
+                                if (!isnan(v[u].w)) { sum += v[u].w; cnt += 1.f; }
+                            }
+                        }
+                    }
+                } else {
+                    for (int p = lane; p < quad; p += ING_GL) {
+                        const f32x4_ v = __builtin_nontemporal_load(&vp[p]);
+                        if (!isnan(v.x)) { sum += v.x; cnt += 1.f; }
+                        if (!isnan(v.y)) { sum += v.y; cnt += 1.f; }
+                        if (!isnan(v.z)) { sum += v.z; cnt += 1.f; }
+                        if (!isnan(v.w)) { sum += v.w; cnt += 1.f; }
+                    }
                 }
                 for (int i = pre + quad * 4 + lane; i < bucket_len; i += ING_GL) {
                     const float f = (float)src[i];
@@ -493,15 +541,33 @@ int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
     if (NB <= 0 || S <= 0) return 0;
     const long nwaves = (long)S * CIN * ((NB + ING_GRP - 1) / ING_GRP);
     const int grid = grid_for(nwaves * WAVE, 256);
-    if (raw_is_bf16)
-        hipLaunchKernelGGL((ingest_dense_kernel<unsigned short>), dim3(grid),
-                           dim3(256), 0, st, (const unsigned short*)raw, bsum,
-                           bcnt, chan_map, S, CIN, C, T, G, bucket_len, head,
-                           dstate);
-    else
-        hipLaunchKernelGGL((ingest_dense_kernel<float>), dim3(grid), dim3(256),
-                           0, st, (const float*)raw, bsum, bcnt, chan_map, S,
-                           CIN, C, T, G, bucket_len, head, dstate);
+    const char* ilp = getenv("TSKD_INGEST_ILP");
+    const bool deep = !(ilp && ilp[0] == '0');
+    if (raw_is_bf16) {
+        if (deep)
+            hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, true>),
+                               dim3(grid), dim3(256), 0, st,
+                               (const unsigned short*)raw, bsum, bcnt,
+                               chan_map, S, CIN, C, T, G, bucket_len, head,
+                               dstate);
+        else
+            hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, false>),
+                               dim3(grid), dim3(256), 0, st,
+                               (const unsigned short*)raw, bsum, bcnt,
+                               chan_map, S, CIN, C, T, G, bucket_len, head,
+                               dstate);
+    } else {
+        if (deep)
+            hipLaunchKernelGGL((ingest_dense_kernel<float, true>), dim3(grid),
+                               dim3(256), 0, st, (const float*)raw, bsum,
+                               bcnt, chan_map, S, CIN, C, T, G, bucket_len,
+                               head, dstate);
+        else
+            hipLaunchKernelGGL((ingest_dense_kernel<float, false>), dim3(grid),
+                               dim3(256), 0, st, (const float*)raw, bsum,
+                               bcnt, chan_map, S, CIN, C, T, G, bucket_len,
+                               head, dstate);
+    }
     return (int)hipGetLastError();
 }
 
