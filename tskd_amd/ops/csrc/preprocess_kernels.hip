@@ -120,8 +120,17 @@ __global__ void ingest_dense_kernel(
                     // 4..5-deep ~7.0 TB/s). Out-of-range lanes re-load chunk
                     // 0 (clamped index) so the loads issue unconditionally,
                     // and their contribution is masked after.
+                    //
+                    // NaN handling is DEFERRED: the fast path unpacks each
+                    // dword into its two bf16 lanes with 2 bit-ops + 2 adds
+                    // (vs 5 VALU/element for the select chain — at 7 TB/s
+                    // the select version is VALU-issue-bound). A NaN input
+                    // poisons the group sum; the group then redoes its
+                    // bucket with the exact per-element masked loop (rare:
+                    // NaN means a dropped sample).
+                    float bA = 0.f, bB = 0.f;
                     for (int base = lane; base < oct; base += 5 * ING_GL) {
-                        union { u32x4_ q; unsigned short h[8]; } v[5];
+                        union { u32x4_ q; unsigned int d[4]; } v[5];
                         #pragma unroll
                         for (int u = 0; u < 5; ++u) {
                             const int pp = base + u * ING_GL;
@@ -132,10 +141,32 @@ __global__ void ingest_dense_kernel(
                         for (int u = 0; u < 5; ++u) {
                             if (base + u * ING_GL < oct) {
                                 #pragma unroll
-                                for (int j = 0; j < 8; ++j) {
-                                    const float f = bf16_to_f32_(v[u].h[j]);
-                                    if (!isnan(f)) { sum += f; cnt += 1.f; }
+                                for (int j = 0; j < 4; ++j) {
+                                    const unsigned int d = v[u].d[j];
+                                    bA += __uint_as_float(d << 16);
+                                    bB += __uint_as_float(d & 0xffff0000u);
                                 }
+                            }
+                        }
+                    }
+                    float body = bA + bB;
+                    float chk = body;
+                    #pragma unroll
+                    for (int off = 8; off > 0; off >>= 1)
+                        chk += __shfl_xor(chk, off);
+                    if (!isnan(chk)) {
+                        sum += body;
+                        cnt += 8.f * (float)(oct > lane
+                                             ? (oct - 1 - lane) / ING_GL + 1
+                                             : 0);
+                    } else {  // rare: masked redo of this bucket's body
+                        for (int p = lane; p < oct; p += ING_GL) {
+                            union { u32x4_ q; unsigned short h[8]; } v;
+                            v.q = __builtin_nontemporal_load(&vp[p]);
+                            #pragma unroll
+                            for (int j = 0; j < 8; ++j) {
+                                const float f = bf16_to_f32_(v.h[j]);
+                                if (!isnan(f)) { sum += f; cnt += 1.f; }
                             }
                         }
                     }
@@ -165,6 +196,8 @@ __global__ void ingest_dense_kernel(
                 const int quad = (bucket_len - pre) / 4;
                 const f32x4_* vp = (const f32x4_*)((const float*)src + pre);
                 if constexpr (ILP) {
+                    // same deferred-NaN fast path as the bf16 body
+                    float bA = 0.f, bB = 0.f;
                     for (int base = lane; base < quad; base += 5 * ING_GL) {
                         f32x4_ v[5];
                         #pragma unroll
@@ -176,11 +209,28 @@ __global__ void ingest_dense_kernel(
                         #pragma unroll
                         for (int u = 0; u < 5; ++u) {
                             if (base + u * ING_GL < quad) {
-                                if (!isnan(v[u].x)) { sum += v[u].x; cnt += 1.f; }
-                                if (!isnan(v[u].y)) { sum += v[u].y; cnt += 1.f; }
-                                if (!isnan(v[u].z)) { sum += v[u].z; cnt += 1.f; }
-                                if (!isnan(v[u].w)) { sum += v[u].w; cnt += 1.f; }
+                                bA += v[u].x + v[u].z;
+                                bB += v[u].y + v[u].w;
                             }
+                        }
+                    }
+                    float body = bA + bB;
+                    float chk = body;
+                    #pragma unroll
+                    for (int off = 8; off > 0; off >>= 1)
+                        chk += __shfl_xor(chk, off);
+                    if (!isnan(chk)) {
+                        sum += body;
+                        cnt += 4.f * (float)(quad > lane
+                                             ? (quad - 1 - lane) / ING_GL + 1
+                                             : 0);
+                    } else {
+                        for (int p = lane; p < quad; p += ING_GL) {
+                            const f32x4_ v = __builtin_nontemporal_load(&vp[p]);
+                            if (!isnan(v.x)) { sum += v.x; cnt += 1.f; }
+                            if (!isnan(v.y)) { sum += v.y; cnt += 1.f; }
+                            if (!isnan(v.z)) { sum += v.z; cnt += 1.f; }
+                            if (!isnan(v.w)) { sum += v.w; cnt += 1.f; }
                         }
                     }
                 } else {
@@ -540,7 +590,11 @@ int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
     const int NB = T / bucket_len;
     if (NB <= 0 || S <= 0) return 0;
     const long nwaves = (long)S * CIN * ((NB + ING_GRP - 1) / ING_GRP);
-    const int grid = grid_for(nwaves * WAVE, 256);
+    // read-ceiling probe: 32768 blocks measures ~3% above the 16384 cap
+    long blocks = (nwaves + 3) / 4;
+    if (blocks > 32768) blocks = 32768;
+    if (blocks < 1) blocks = 1;
+    const int grid = (int)blocks;
     const char* ilp = getenv("TSKD_INGEST_ILP");
     const bool deep = !(ilp && ilp[0] == '0');
     if (raw_is_bf16) {
