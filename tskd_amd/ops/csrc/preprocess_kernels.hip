@@ -75,7 +75,7 @@ __device__ __forceinline__ void wsync_() {
 // busy (78/16 = 4.9 balanced iterations per group).
 #define ING_GRP 4   // buckets per wave
 #define ING_GL 16   // lanes per bucket group
-template <class DT, bool ILP = true>
+template <class DT, int IMODE = 2>
 __global__ void ingest_dense_kernel(
     const DT* __restrict__ raw,     // (S, CIN, T)
     float* __restrict__ bsum,       // (S, C, G)
@@ -114,7 +114,28 @@ __global__ void ingest_dense_kernel(
                 }
                 const int oct = (bucket_len - pre) / 8;
                 const u32x4_* vp = (const u32x4_*)((const unsigned short*)src + pre);
-                if constexpr (ILP) {
+                if constexpr (IMODE == 1) {
+                    // simple 5-deep ILP with per-element NaN selects
+                    for (int base = lane; base < oct; base += 5 * ING_GL) {
+                        union { u32x4_ q; unsigned short h[8]; } v[5];
+                        #pragma unroll
+                        for (int u = 0; u < 5; ++u) {
+                            const int pp = base + u * ING_GL;
+                            v[u].q = __builtin_nontemporal_load(
+                                &vp[pp < oct ? pp : 0]);
+                        }
+                        #pragma unroll
+                        for (int u = 0; u < 5; ++u) {
+                            if (base + u * ING_GL < oct) {
+                                #pragma unroll
+                                for (int j = 0; j < 8; ++j) {
+                                    const float f = bf16_to_f32_(v[u].h[j]);
+                                    if (!isnan(f)) { sum += f; cnt += 1.f; }
+                                }
+                            }
+                        }
+                    }
+                } else if constexpr (IMODE == 2) {
                     // 5 independent nontemporal loads in flight per lane
                     // (read-ceiling probe: 1-deep streams measure ~5.9 TB/s,
                     // 4..5-deep ~7.0 TB/s). Out-of-range lanes re-load chunk
@@ -195,7 +216,26 @@ __global__ void ingest_dense_kernel(
                 }
                 const int quad = (bucket_len - pre) / 4;
                 const f32x4_* vp = (const f32x4_*)((const float*)src + pre);
-                if constexpr (ILP) {
+                if constexpr (IMODE == 1) {
+                    for (int base = lane; base < quad; base += 5 * ING_GL) {
+                        f32x4_ v[5];
+                        #pragma unroll
+                        for (int u = 0; u < 5; ++u) {
+                            const int pp = base + u * ING_GL;
+                            v[u] = __builtin_nontemporal_load(
+                                &vp[pp < quad ? pp : 0]);
+                        }
+                        #pragma unroll
+                        for (int u = 0; u < 5; ++u) {
+                            if (base + u * ING_GL < quad) {
+                                if (!isnan(v[u].x)) { sum += v[u].x; cnt += 1.f; }
+                                if (!isnan(v[u].y)) { sum += v[u].y; cnt += 1.f; }
+                                if (!isnan(v[u].z)) { sum += v[u].z; cnt += 1.f; }
+                                if (!isnan(v[u].w)) { sum += v[u].w; cnt += 1.f; }
+                            }
+                        }
+                    }
+                } else if constexpr (IMODE == 2) {
                     // same deferred-NaN fast path as the bf16 body
                     float bA = 0.f, bB = 0.f;
                     for (int base = lane; base < quad; base += 5 * ING_GL) {
@@ -595,32 +635,41 @@ int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
     if (blocks > 32768) blocks = 32768;
     if (blocks < 1) blocks = 1;
     const int grid = (int)blocks;
-    const char* ilp = getenv("TSKD_INGEST_ILP");
-    const bool deep = !(ilp && ilp[0] == '0');
+    // TSKD_INGEST_ILP: 0 = 1-deep select loop, 1 = 5-deep ILP + selects,
+    // 2 (default) = 5-deep ILP + deferred-NaN packed adds
+    int mode = 2;
+    if (const char* ilp = getenv("TSKD_INGEST_ILP")) mode = atoi(ilp);
     if (raw_is_bf16) {
-        if (deep)
-            hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, true>),
-                               dim3(grid), dim3(256), 0, st,
-                               (const unsigned short*)raw, bsum, bcnt,
+        const unsigned short* rp = (const unsigned short*)raw;
+        if (mode == 0)
+            hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, 0>),
+                               dim3(grid), dim3(256), 0, st, rp, bsum, bcnt,
+                               chan_map, S, CIN, C, T, G, bucket_len, head,
+                               dstate);
+        else if (mode == 1)
+            hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, 1>),
+                               dim3(grid), dim3(256), 0, st, rp, bsum, bcnt,
                                chan_map, S, CIN, C, T, G, bucket_len, head,
                                dstate);
         else
-            hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, false>),
-                               dim3(grid), dim3(256), 0, st,
-                               (const unsigned short*)raw, bsum, bcnt,
+            hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, 2>),
+                               dim3(grid), dim3(256), 0, st, rp, bsum, bcnt,
                                chan_map, S, CIN, C, T, G, bucket_len, head,
                                dstate);
     } else {
-        if (deep)
-            hipLaunchKernelGGL((ingest_dense_kernel<float, true>), dim3(grid),
-                               dim3(256), 0, st, (const float*)raw, bsum,
-                               bcnt, chan_map, S, CIN, C, T, G, bucket_len,
-                               head, dstate);
+        const float* rp = (const float*)raw;
+        if (mode == 0)
+            hipLaunchKernelGGL((ingest_dense_kernel<float, 0>), dim3(grid),
+                               dim3(256), 0, st, rp, bsum, bcnt, chan_map, S,
+                               CIN, C, T, G, bucket_len, head, dstate);
+        else if (mode == 1)
+            hipLaunchKernelGGL((ingest_dense_kernel<float, 1>), dim3(grid),
+                               dim3(256), 0, st, rp, bsum, bcnt, chan_map, S,
+                               CIN, C, T, G, bucket_len, head, dstate);
         else
-            hipLaunchKernelGGL((ingest_dense_kernel<float, false>), dim3(grid),
-                               dim3(256), 0, st, (const float*)raw, bsum,
-                               bcnt, chan_map, S, CIN, C, T, G, bucket_len,
-                               head, dstate);
+            hipLaunchKernelGGL((ingest_dense_kernel<float, 2>), dim3(grid),
+                               dim3(256), 0, st, rp, bsum, bcnt, chan_map, S,
+                               CIN, C, T, G, bucket_len, head, dstate);
     }
     return (int)hipGetLastError();
 }
