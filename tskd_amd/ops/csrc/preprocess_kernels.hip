@@ -303,6 +303,116 @@ __global__ void ingest_dense_kernel(
     }
 }
 
+
+// IMODE=3 experiment: each 16-lane group owns TWO buckets (b and b+4) so a
+// wave keeps 10 independent oct-loads in flight and amortizes the per-task
+// index math / reduce / store bubble over twice the bytes. bf16 only (the
+// serving path); fp32 falls back to the 5-deep kernel in the launcher.
+__global__ void ingest_dense_pair_kernel(
+    const unsigned short* __restrict__ raw, float* __restrict__ bsum,
+    float* __restrict__ bcnt, const int* __restrict__ chan_map,
+    int S, int CIN, int C, int T, int G, int bucket_len, long head_in,
+    const long long* __restrict__ dstate)
+{
+    const long head = ring_head(dstate, head_in);
+    const int NB = T / bucket_len;
+    const int SPAN = 2 * ING_GRP;                 // 8 buckets per wave
+    const int NBG = (NB + SPAN - 1) / SPAN;
+    const long nwaves = (long)S * CIN * NBG;
+    const int wlane = threadIdx.x % WAVE;
+    const int grp = wlane / ING_GL;
+    const int lane = wlane % ING_GL;
+    for (long w = (long)blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
+         w < nwaves; w += (long)gridDim.x * (blockDim.x / WAVE)) {
+        const int bg = (int)(w % NBG);
+        const int cin = (int)((w / NBG) % CIN);
+        const int s = (int)(w / ((long)NBG * CIN));
+        const int b0 = bg * SPAN + grp, b1 = b0 + ING_GRP;
+        const bool a0 = b0 < NB, a1 = b1 < NB;
+        const unsigned short* row = raw + ((long)s * CIN + cin) * T;
+        const unsigned short* src0 = row + (long)(a0 ? b0 : 0) * bucket_len;
+        const unsigned short* src1 = row + (long)(a1 ? b1 : 0) * bucket_len;
+        float sum0 = 0.f, cnt0 = 0.f, sum1 = 0.f, cnt1 = 0.f;
+        int pre0 = (int)(((16 - ((size_t)src0 & 15)) & 15) / 2);
+        int pre1 = (int)(((16 - ((size_t)src1 & 15)) & 15) / 2);
+        if (pre0 > bucket_len) pre0 = bucket_len;
+        if (pre1 > bucket_len) pre1 = bucket_len;
+        if (a0)
+            for (int i = lane; i < pre0; i += ING_GL) {
+                const float f = bf16_to_f32_(src0[i]);
+                if (!isnan(f)) { sum0 += f; cnt0 += 1.f; }
+            }
+        if (a1)
+            for (int i = lane; i < pre1; i += ING_GL) {
+                const float f = bf16_to_f32_(src1[i]);
+                if (!isnan(f)) { sum1 += f; cnt1 += 1.f; }
+            }
+        const int oct0 = a0 ? (bucket_len - pre0) / 8 : 0;
+        const int oct1 = a1 ? (bucket_len - pre1) / 8 : 0;
+        const u32x4_* vp0 = (const u32x4_*)(src0 + pre0);
+        const u32x4_* vp1 = (const u32x4_*)(src1 + pre1);
+        const int octm = oct0 > oct1 ? oct0 : oct1;
+        for (int base = lane; base < octm; base += 5 * ING_GL) {
+            union U { u32x4_ q; unsigned short h[8]; } v0[5], v1[5];
+            #pragma unroll
+            for (int u = 0; u < 5; ++u) {
+                const int pp = base + u * ING_GL;
+                v0[u].q = __builtin_nontemporal_load(&vp0[pp < oct0 ? pp : 0]);
+                v1[u].q = __builtin_nontemporal_load(&vp1[pp < oct1 ? pp : 0]);
+            }
+            #pragma unroll
+            for (int u = 0; u < 5; ++u) {
+                const int pp = base + u * ING_GL;
+                if (pp < oct0) {
+                    #pragma unroll
+                    for (int j = 0; j < 8; ++j) {
+                        const float f = bf16_to_f32_(v0[u].h[j]);
+                        if (!isnan(f)) { sum0 += f; cnt0 += 1.f; }
+                    }
+                }
+                if (pp < oct1) {
+                    #pragma unroll
+                    for (int j = 0; j < 8; ++j) {
+                        const float f = bf16_to_f32_(v1[u].h[j]);
+                        if (!isnan(f)) { sum1 += f; cnt1 += 1.f; }
+                    }
+                }
+            }
+        }
+        if (a0)
+            for (int i = pre0 + oct0 * 8 + lane; i < bucket_len; i += ING_GL) {
+                const float f = bf16_to_f32_(src0[i]);
+                if (!isnan(f)) { sum0 += f; cnt0 += 1.f; }
+            }
+        if (a1)
+            for (int i = pre1 + oct1 * 8 + lane; i < bucket_len; i += ING_GL) {
+                const float f = bf16_to_f32_(src1[i]);
+                if (!isnan(f)) { sum1 += f; cnt1 += 1.f; }
+            }
+        #pragma unroll
+        for (int off = 8; off > 0; off >>= 1) {
+            sum0 += __shfl_xor(sum0, off);
+            cnt0 += __shfl_xor(cnt0, off);
+            sum1 += __shfl_xor(sum1, off);
+            cnt1 += __shfl_xor(cnt1, off);
+        }
+        if (lane == 0) {
+            const int c = chan_map[cin];
+            const long rowo = ((long)s * C + c) * G;
+            if (a0) {
+                const long idx = rowo + (head + b0) % G;
+                bsum[idx] = sum0;
+                bcnt[idx] = cnt0;
+            }
+            if (a1) {
+                const long idx = rowo + (head + b1) % G;
+                bsum[idx] = sum1;
+                bcnt[idx] = cnt1;
+            }
+        }
+    }
+}
+
 // ---------------------------------------------------------------------------
 // 1b. Sparse ingest: event tuples (stream, chan, grid_bucket, value)
 // ---------------------------------------------------------------------------
@@ -643,6 +753,12 @@ int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
     if (const char* ilp = getenv("TSKD_INGEST_ILP")) mode = atoi(ilp);
     if (raw_is_bf16) {
         const unsigned short* rp = (const unsigned short*)raw;
+        if (mode == 3) {
+            hipLaunchKernelGGL(ingest_dense_pair_kernel, dim3(grid), dim3(256),
+                               0, st, rp, bsum, bcnt, chan_map, S, CIN, C, T,
+                               G, bucket_len, head, dstate);
+            return (int)hipGetLastError();
+        }
         if (mode == 0)
             hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, 0>),
                                dim3(grid), dim3(256), 0, st, rp, bsum, bcnt,
