@@ -27,10 +27,11 @@ def main() -> None:
     n = 2 * 1024 * 1024 * 1024  # floats -> 8 GiB
     x = torch.randn(n, device="cuda")
     out = torch.empty(65536, device="cuda")
-    names = ["x4 nt", "x8 nt", "x16 nt", "x4", "x8", "x16"]
+    names = ["x4 nt", "x8 nt", "x16 nt", "x4", "x8", "x16",
+         "grouped", "contig"]
     stream = torch.cuda.current_stream().cuda_stream
     best = (0.0, "")
-    for variant in range(6):
+    for variant in range(8):
         for grid in (2048, 4096, 8192, 16384, 32768):
             rc = lib.bw_probe(x.data_ptr(), n, out.data_ptr(), variant,
                               grid, stream)

@@ -41,6 +41,53 @@ __global__ void read_probe(const float* __restrict__ p, long n4,
     }
 }
 
+
+// Variant 6: mirror ingest_dense's request pattern — 16-lane groups each
+// sweeping a separate 1248 B region (4 regions per wave-task), 5-deep.
+// Variant 7: SAME task shape (4992 B span per wave-task) but whole-wave
+// contiguous lane mapping. 6 vs 7 isolates the request-pattern variable.
+__global__ void read_probe_grouped(const float* __restrict__ p, long nbytes,
+                                   float* __restrict__ out, int contiguous) {
+    const int wlane = threadIdx.x % 64;
+    const int grp = wlane / 16, gl = wlane % 16;
+    const long ntasks = nbytes / 4992;
+    f32x4_ acc = {0.f, 0.f, 0.f, 0.f};
+    for (long t = (long)blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
+         t < ntasks; t += (long)gridDim.x * (blockDim.x / 64)) {
+        if (contiguous) {
+            const f32x4_* base = (const f32x4_*)((const char*)p + t * 4992);
+            const int oct = 312;
+            for (int b = wlane; b < oct; b += 5 * 64) {
+                #pragma unroll
+                for (int u = 0; u < 5; ++u) {
+                    const int pp = b + u * 64;
+                    acc += __builtin_nontemporal_load(&base[pp < oct ? pp : 0]);
+                }
+            }
+        } else {
+            const f32x4_* base = (const f32x4_*)((const char*)p + t * 4992 +
+                                                 grp * 1248);
+            const int oct = 78;
+            for (int b = gl; b < oct; b += 5 * 16) {
+                #pragma unroll
+                for (int u = 0; u < 5; ++u) {
+                    const int pp = b + u * 16;
+                    acc += __builtin_nontemporal_load(&base[pp < oct ? pp : 0]);
+                }
+            }
+        }
+    }
+    float s = acc.x + acc.y + acc.z + acc.w;
+    __shared__ float red[256];
+    red[threadIdx.x] = s;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float t2 = 0.f;
+        for (int j = 0; j < 256; ++j) t2 += red[j];
+        out[blockIdx.x] = t2;
+    }
+}
+
 extern "C" int bw_probe(const float* p, long n, float* out, int variant,
                         int grid, void* stream) {
     const long n4 = n / 4;
@@ -58,6 +105,10 @@ extern "C" int bw_probe(const float* p, long n, float* out, int variant,
                                    dim3(256), 0, st, p, n4, out); break;
         case 5: hipLaunchKernelGGL((read_probe<4, false>), dim3(grid),
                                    dim3(256), 0, st, p, n4, out); break;
+        case 6: hipLaunchKernelGGL(read_probe_grouped, dim3(grid), dim3(256),
+                                   0, st, p, n * 4L, out, 0); break;
+        case 7: hipLaunchKernelGGL(read_probe_grouped, dim3(grid), dim3(256),
+                                   0, st, p, n * 4L, out, 1); break;
     }
     return (int)hipGetLastError();
 }
