@@ -10,6 +10,7 @@
 #include <hip/hip_runtime.h>
 
 typedef float f32x4_ __attribute__((ext_vector_type(4)));
+typedef unsigned int u32x4_ __attribute__((ext_vector_type(4)));
 
 // Each variant sums `n` floats (16 B-aligned base) and folds the result to
 // out[block] so the loads cannot be optimized away.
@@ -88,6 +89,55 @@ __global__ void read_probe_grouped(const float* __restrict__ p, long nbytes,
     }
 }
 
+
+// Variant 8: grouped pattern + the REAL ingest consumption (bf16 unpack,
+// NaN select, count) — isolates the VALU-consumption cost from the
+// request pattern (variant 6) and the pure-read ceiling (variant 7).
+__device__ __forceinline__ float bf16u_(unsigned short h) {
+    union { unsigned int i; float f; } u;
+    u.i = ((unsigned int)h) << 16;
+    return u.f;
+}
+__global__ void read_probe_consume(const float* __restrict__ p, long nbytes,
+                                   float* __restrict__ out) {
+    const int wlane = threadIdx.x % 64;
+    const int grp = wlane / 16, gl = wlane % 16;
+    const long ntasks = nbytes / 4992;
+    float sum = 0.f, cnt = 0.f;
+    for (long t = (long)blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
+         t < ntasks; t += (long)gridDim.x * (blockDim.x / 64)) {
+        const char* base = (const char*)p + t * 4992 + grp * 1248;
+        const int oct = 78;
+        for (int b = gl; b < oct; b += 5 * 16) {
+            union { u32x4_ q; unsigned short h[8]; } v[5];
+            #pragma unroll
+            for (int u = 0; u < 5; ++u) {
+                const int pp = b + u * 16;
+                v[u].q = __builtin_nontemporal_load(
+                    (const u32x4_*)(base + (pp < oct ? pp : 0) * 16));
+            }
+            #pragma unroll
+            for (int u = 0; u < 5; ++u) {
+                if (b + u * 16 < oct) {
+                    #pragma unroll
+                    for (int j = 0; j < 8; ++j) {
+                        const float f = bf16u_(v[u].h[j]);
+                        if (!isnan(f)) { sum += f; cnt += 1.f; }
+                    }
+                }
+            }
+        }
+    }
+    __shared__ float red[256];
+    red[threadIdx.x] = sum + cnt;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float t2 = 0.f;
+        for (int j = 0; j < 256; ++j) t2 += red[j];
+        out[blockIdx.x] = t2;
+    }
+}
+
 extern "C" int bw_probe(const float* p, long n, float* out, int variant,
                         int grid, void* stream) {
     const long n4 = n / 4;
@@ -109,6 +159,8 @@ extern "C" int bw_probe(const float* p, long n, float* out, int variant,
                                    0, st, p, n * 4L, out, 0); break;
         case 7: hipLaunchKernelGGL(read_probe_grouped, dim3(grid), dim3(256),
                                    0, st, p, n * 4L, out, 1); break;
+        case 8: hipLaunchKernelGGL(read_probe_consume, dim3(grid), dim3(256),
+                                   0, st, p, n * 4L, out); break;
     }
     return (int)hipGetLastError();
 }
