@@ -28,10 +28,10 @@ def main() -> None:
     x = torch.randn(n, device="cuda")
     out = torch.empty(65536, device="cuda")
     names = ["x4 nt", "x8 nt", "x16 nt", "x4", "x8", "x16",
-         "grouped", "contig", "consume"]
+         "grouped", "contig", "consume", "packed"]
     stream = torch.cuda.current_stream().cuda_stream
     best = (0.0, "")
-    for variant in range(9):
+    for variant in range(10):
         for grid in (2048, 4096, 8192, 16384, 32768):
             rc = lib.bw_probe(x.data_ptr(), n, out.data_ptr(), variant,
                               grid, stream)

@@ -138,6 +138,50 @@ __global__ void read_probe_consume(const float* __restrict__ p, long nbytes,
     }
 }
 
+
+// Variant 9: grouped + PACKED consumption (1 shift + 1 and + 2 adds per
+// dword, no NaN selects, no counts) — the mode-2 ingest fast path's cost.
+__global__ void read_probe_packed(const float* __restrict__ p, long nbytes,
+                                  float* __restrict__ out) {
+    const int wlane = threadIdx.x % 64;
+    const int grp = wlane / 16, gl = wlane % 16;
+    const long ntasks = nbytes / 4992;
+    float a = 0.f, b2 = 0.f;
+    for (long t = (long)blockIdx.x * (blockDim.x / 64) + threadIdx.x / 64;
+         t < ntasks; t += (long)gridDim.x * (blockDim.x / 64)) {
+        const char* base = (const char*)p + t * 4992 + grp * 1248;
+        const int oct = 78;
+        for (int b = gl; b < oct; b += 5 * 16) {
+            union { u32x4_ q; unsigned int d[4]; } v[5];
+            #pragma unroll
+            for (int u = 0; u < 5; ++u) {
+                const int pp = b + u * 16;
+                v[u].q = __builtin_nontemporal_load(
+                    (const u32x4_*)(base + (pp < oct ? pp : 0) * 16));
+            }
+            #pragma unroll
+            for (int u = 0; u < 5; ++u) {
+                if (b + u * 16 < oct) {
+                    #pragma unroll
+                    for (int j = 0; j < 4; ++j) {
+                        const unsigned int d = v[u].d[j];
+                        a += __uint_as_float(d << 16);
+                        b2 += __uint_as_float(d & 0xffff0000u);
+                    }
+                }
+            }
+        }
+    }
+    __shared__ float red[256];
+    red[threadIdx.x] = a + b2;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float t2 = 0.f;
+        for (int j = 0; j < 256; ++j) t2 += red[j];
+        out[blockIdx.x] = t2;
+    }
+}
+
 extern "C" int bw_probe(const float* p, long n, float* out, int variant,
                         int grid, void* stream) {
     const long n4 = n / 4;
@@ -160,6 +204,8 @@ extern "C" int bw_probe(const float* p, long n, float* out, int variant,
         case 7: hipLaunchKernelGGL(read_probe_grouped, dim3(grid), dim3(256),
                                    0, st, p, n * 4L, out, 1); break;
         case 8: hipLaunchKernelGGL(read_probe_consume, dim3(grid), dim3(256),
+                                   0, st, p, n * 4L, out); break;
+        case 9: hipLaunchKernelGGL(read_probe_packed, dim3(grid), dim3(256),
                                    0, st, p, n * 4L, out); break;
     }
     return (int)hipGetLastError();

@@ -86,6 +86,7 @@ __global__ void ingest_dense_kernel(
     const long long* __restrict__ dstate)
 {
     const long head = ring_head(dstate, head_in);
+    const int head_mod = (int)(head % G);  // hoisted: no 64-bit mod per task
     const int NB = T / bucket_len;
     const int NBG = (NB + ING_GRP - 1) / ING_GRP;
     const long nwaves = (long)S * CIN * NBG;
@@ -94,9 +95,14 @@ __global__ void ingest_dense_kernel(
     const int lane = wlane % ING_GL;    // lane within the bucket group
     for (long w = (long)blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
          w < nwaves; w += (long)gridDim.x * (blockDim.x / WAVE)) {
-        const int bg = (int)(w % NBG);
-        const int cin = (int)((w / NBG) % CIN);
-        const int s = (int)(w / ((long)NBG * CIN));
+        // 32-bit decode: nwaves < 2^31 in any real config; 64-bit div/mod
+        // per task measured as a visible fraction of the wave's cycle
+        // budget (12 tasks x ~5 KB each)
+        const int wi = (int)w;
+        const int bg = wi % NBG;
+        const int wrem = wi / NBG;
+        const int cin = wrem % CIN;
+        const int s = wrem / CIN;
         const int b = bg * ING_GRP + grp;
         const bool active = b < NB;
         const DT* src = raw + ((long)s * CIN + cin) * T +
@@ -296,7 +302,9 @@ __global__ void ingest_dense_kernel(
         }
         if (lane == 0 && active) {
             const int c = chan_map[cin];
-            const long idx = ((long)s * C + c) * G + (head + b) % G;
+            int bi = head_mod + b;
+            if (bi >= G) bi -= G;      // b < NB <= G: one subtract suffices
+            const long idx = ((long)s * C + c) * G + bi;
             bsum[idx] = sum;
             bcnt[idx] = cnt;
         }
@@ -315,6 +323,7 @@ __global__ void ingest_dense_pair_kernel(
     const long long* __restrict__ dstate)
 {
     const long head = ring_head(dstate, head_in);
+    const int head_mod = (int)(head % G);
     const int NB = T / bucket_len;
     const int SPAN = 2 * ING_GRP;                 // 8 buckets per wave
     const int NBG = (NB + SPAN - 1) / SPAN;
@@ -324,9 +333,11 @@ __global__ void ingest_dense_pair_kernel(
     const int lane = wlane % ING_GL;
     for (long w = (long)blockIdx.x * (blockDim.x / WAVE) + threadIdx.x / WAVE;
          w < nwaves; w += (long)gridDim.x * (blockDim.x / WAVE)) {
-        const int bg = (int)(w % NBG);
-        const int cin = (int)((w / NBG) % CIN);
-        const int s = (int)(w / ((long)NBG * CIN));
+        const int wi = (int)w;
+        const int bg = wi % NBG;
+        const int wrem = wi / NBG;
+        const int cin = wrem % CIN;
+        const int s = wrem / CIN;
         const int b0 = bg * SPAN + grp, b1 = b0 + ING_GRP;
         const bool a0 = b0 < NB, a1 = b1 < NB;
         const unsigned short* row = raw + ((long)s * CIN + cin) * T;
@@ -400,14 +411,16 @@ __global__ void ingest_dense_pair_kernel(
             const int c = chan_map[cin];
             const long rowo = ((long)s * C + c) * G;
             if (a0) {
-                const long idx = rowo + (head + b0) % G;
-                bsum[idx] = sum0;
-                bcnt[idx] = cnt0;
+                int bi = head_mod + b0;
+                if (bi >= G) bi -= G;
+                bsum[rowo + bi] = sum0;
+                bcnt[rowo + bi] = cnt0;
             }
             if (a1) {
-                const long idx = rowo + (head + b1) % G;
-                bsum[idx] = sum1;
-                bcnt[idx] = cnt1;
+                int bi = head_mod + b1;
+                if (bi >= G) bi -= G;
+                bsum[rowo + bi] = sum1;
+                bcnt[rowo + bi] = cnt1;
             }
         }
     }
