@@ -720,11 +720,14 @@ int launch_conv(const void* x, int x_is_bf16, int x_timelast, float* feat,
         if (!bfrag) return -2;  // bf16 path requires packed B fragments
         if (x_timelast) {
             if constexpr (G::CIN % 2 == 0) {
-                int onewave = 0;
+                // DEFAULT: 1-wave-per-2-windows. The 4-wave-WG variant
+                // measured equal at 4096x256 but 17% SLOWER at the bench
+                // shape 8192x1024 (gpurun_out/val_infknobs.log) — always
+                // re-A/B at the production shape.
+                int onewave = 1;
                 if (const char* e = getenv("TSKD_CONV_TLAST_1WAVE"))
                     onewave = atoi(e);
                 if (onewave) {
-                    // 1-wave-per-2-windows variant (A/B reference)
                     int g1 = (SN / 2 + WG_WAVES - 1) / WG_WAVES;
                     if (g1 < 1) g1 = 1;
                     if (g1 > 8192) g1 = 8192;
@@ -733,7 +736,7 @@ int launch_conv(const void* x, int x_is_bf16, int x_timelast, float* feat,
                                        (const unsigned short*)x, feat, wpack,
                                        (const unsigned short*)bfrag, SN);
                 } else {
-                    // default: 4-wave workgroup per window
+                    // 4-wave workgroup per window (A/B reference)
                     int g2 = SN > 16384 ? 16384 : SN;
                     hipLaunchKernelGGL((conv_stack_mfma_wg_kernel<G>),
                                        dim3(g2), dim3(WG_THREADS), 0, stream,
