@@ -67,9 +67,8 @@ def _dp_worker(rank, world, port, tmpdir):
 
 
 class TestGlooDP:
-    def test_dp_serving_world2(self, tmp_path):
-        world = 2
-        port = 29531
+    @pytest.mark.parametrize("world,port", [(2, 29531), (4, 29533)])
+    def test_dp_serving(self, tmp_path, world, port):
         ctx = mp.get_context("spawn")
         ps = [ctx.Process(target=_dp_worker,
                           args=(r, world, port, str(tmp_path)))
