@@ -23,30 +23,40 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch  # noqa: E402
 
 
-def build_step(mode: str, device):
+def build_step(mode: str, device, seqs: int, batch: int, streams: int):
+    # NOTE: A/B verdicts are SHAPE-SPECIFIC (the conv WG-vs-1wave schedules
+    # measured equal at 4096x256 and 17% apart at 8192x1024). Default
+    # shapes here match bench.py's production shapes; override with
+    # --seqs/--batch/--streams only to study scaling.
     from tskd_amd.models import build_model
     from tskd_amd.ops import MyCNNEngine
     torch.manual_seed(0)
     model = build_model("MyCNN5").eval()
     if mode == "train":
         from tskd_amd.train.hip_trainer import MyCNNHipTrainer
+        S = seqs or 1024
         tr = MyCNNHipTrainer(model, device=device, lr=1e-5)
-        x = torch.randn(512, 64, 10, 120, device=device)
-        age = torch.full((512, 64), 65.0, device=device)
-        y = (torch.rand(512, 64, device=device) < 0.3).float()
+        x = torch.randn(S, 64, 10, 120, device=device)
+        age = torch.full((S, 64), 65.0, device=device)
+        y = (torch.rand(S, 64, device=device) < 0.3).float()
         return lambda: tr.step(x, age, y)
     if mode == "infer":
+        from tskd_amd.ops import alloc_windows
+        S, B = seqs or 8192, batch or 1024
         eng = MyCNNEngine(model, device=device)
-        x = torch.randn(4096, 256, 10, 120, device=device,
-                        dtype=torch.bfloat16)
-        age = torch.full((4096, 256), 65.0, device=device)
+        x = alloc_windows(S, B, 10, timelast=True, dtype=torch.bfloat16,
+                          device=device)
+        x.copy_(torch.randn(S, B, 120, 10, device=device,
+                            dtype=torch.bfloat16))
+        age = torch.full((S, B), 65.0, device=device)
         return lambda: eng.forward(x, age, apply_sigmoid=True)
     if mode == "pipeline":
         from tskd_amd.engine import StreamEngine
+        S = streams or 16384
         eng = MyCNNEngine(model, device=device)
-        se = StreamEngine(16384, 10, ring_grid=2048, fs=125.0, device=device)
-        raw = torch.randn(16384, 8, 7500, device=device, dtype=torch.bfloat16)
-        age = torch.full((16384, 1), 65.0, device=device)
+        se = StreamEngine(S, 10, ring_grid=2048, fs=125.0, device=device)
+        raw = torch.randn(S, 8, 7500, device=device, dtype=torch.bfloat16)
+        age = torch.full((S, 1), 65.0, device=device)
 
         def step():
             se.ingest_dense(raw, chan_map=list(range(8)))
@@ -66,9 +76,12 @@ def main() -> None:
     ap.add_argument("--rounds", type=int, default=8)
     ap.add_argument("--steps", type=int, default=8)
     ap.add_argument("--warmup", type=int, default=6)
+    ap.add_argument("--seqs", type=int, default=0)
+    ap.add_argument("--batch", type=int, default=0)
+    ap.add_argument("--streams", type=int, default=0)
     args = ap.parse_args()
     assert torch.cuda.is_available()
-    step = build_step(args.mode, "cuda")
+    step = build_step(args.mode, "cuda", args.seqs, args.batch, args.streams)
     for _ in range(args.warmup):
         step()
     torch.cuda.synchronize()
