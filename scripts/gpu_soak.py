@@ -88,6 +88,9 @@ def soak_cli_stages(n_triggers=30):
 
 if __name__ == "__main__":
     assert torch.cuda.is_available()
-    soak_trigger_graph(int(sys.argv[1]) if len(sys.argv) > 1 else 2000)
-    soak_cli_stages()
+    n = int(sys.argv[1]) if len(sys.argv) > 1 else 2000
+    S = int(sys.argv[2]) if len(sys.argv) > 2 else 4096
+    soak_trigger_graph(n, S=S)
+    if S <= 8192:  # CLI-stage soak is stream-count independent
+        soak_cli_stages()
     print("[soak] ALL OK")
