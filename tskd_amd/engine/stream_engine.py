@@ -208,18 +208,26 @@ class StreamEngine:
         seconds (defaults to max ts) — buckets strictly before it become
         eligible for processing.
         """
-        bucket = torch.floor(ts.double() / self.bucket_s).long()
         min_bucket = self.nproc  # processed grid is immutable
-        if len(bucket) and int(bucket.max()) - max(int(bucket.min()), min_bucket) \
-                >= self.G - self.win_buckets:
-            raise ValueError("event batch spans more buckets than the ring "
-                             "holds; ingest in smaller chunks")
-        self._clear_ahead(int(bucket.max().item()) + 1 if len(bucket) else 0)
         if self._gpu:
+            # Bucketing/min/max on the GPU: host float64 math over millions
+            # of events was the event path's dominant cost (event_bench).
+            tsd = ts.double().to(self.device) if not ts.is_cuda else ts.double()
+            bucket_d = torch.floor(tsd / self.bucket_s).long()
+            if len(bucket_d):
+                bmax = int(bucket_d.max().item())
+                bmin = int(bucket_d.min().item())
+                if bmax - max(bmin, min_bucket) >= self.G - self.win_buckets:
+                    raise ValueError(
+                        "event batch spans more buckets than the ring "
+                        "holds; ingest in smaller chunks")
+                self._clear_ahead(bmax + 1)
+            else:
+                self._clear_ahead(0)
             lib = _load_preproc_lib()
             si = stream_idx.to(torch.int32).contiguous().to(self.device)
             ci = chan_idx.to(torch.int32).contiguous().to(self.device)
-            bi = bucket.contiguous().to(self.device)
+            bi = bucket_d.contiguous()
             vi = vals.float().contiguous().to(self.device)
             rc = lib.tskd_preproc_ingest_events(
                 ctypes.c_void_p(si.data_ptr()), ctypes.c_void_p(ci.data_ptr()),
@@ -230,7 +238,17 @@ class StreamEngine:
                 ctypes.c_long(min_bucket), _sptr())
             if rc != 0:
                 raise RuntimeError(f"ingest_events failed: hipError {rc}")
+            if advance_to is None and len(tsd):
+                advance_to = float(tsd.max().item())
         else:
+            bucket = torch.floor(ts.double() / self.bucket_s).long()
+            if len(bucket) and int(bucket.max()) - max(int(bucket.min()),
+                                                       min_bucket) \
+                    >= self.G - self.win_buckets:
+                raise ValueError("event batch spans more buckets than the "
+                                 "ring holds; ingest in smaller chunks")
+            self._clear_ahead(int(bucket.max().item()) + 1 if len(bucket)
+                              else 0)
             ok = (~torch.isnan(vals)) & (bucket >= min_bucket)
             if ok.any():
                 flat = ((stream_idx[ok].numpy() * self.C
