@@ -342,6 +342,91 @@ class TestTriggerGraph:
             idx += 1
 
 
+@pytest.mark.gpu
+class TestTriggerGraphOverlap:
+    def test_overlap_matches_eager_with_refresh(self):
+        """Two-stream overlapped TriggerGraph, raw refreshed per trigger
+        (raw_refresh=True serializes the refresh against ingest): outputs
+        must match the eager twin exactly per trigger."""
+        from tskd_amd.engine.stream_engine import TriggerGraph
+        from tskd_amd.models import build_model
+        from tskd_amd.ops import GraphedForward, MyCNNEngine
+        fs, S = 125.0, 8
+        torch.manual_seed(11)
+        me = MyCNNEngine(build_model("MyCNN5").eval(), device="cuda")
+        chunks = [torch.randn(S, 8, int(fs * 60), device="cuda",
+                              dtype=torch.bfloat16) for _ in range(20)]
+        cm = list(range(8))
+        e1 = StreamEngine(S, 10, ring_grid=512, fs=fs, device="cuda")
+        eager = []
+        for ch in chunks:
+            e1.ingest_dense(ch, chan_map=cm)
+            w = e1.windows(batch=1, stride=12, dtype=torch.bfloat16)
+            eager.append(me.forward(
+                w, torch.full((S, 1), 65.0, device="cuda"),
+                apply_sigmoid=True).clone())
+        e2 = StreamEngine(S, 10, ring_grid=512, fs=fs, device="cuda")
+        raw_buf = chunks[0].clone()
+        idx = 0
+        while e2.nproc == 0 or e2.nproc < e2.head - e2.win_buckets + 1:
+            raw_buf.copy_(chunks[idx])
+            e2.ingest_dense(raw_buf, chan_map=cm)
+            idx += 1
+        torch.cuda.synchronize()
+        gf = GraphedForward(me, s=S, n=1, dtype=torch.bfloat16, timelast=True)
+        raw_buf.copy_(chunks[idx])  # consumed by the warm trigger
+        tg = TriggerGraph(e2, raw_buf, cm, gf, stride=12, overlap=True)
+        idx += 1
+        while idx < len(chunks):
+            raw_buf.copy_(chunks[idx])
+            out = tg.replay(raw_refresh=True)
+            torch.cuda.synchronize()
+            torch.testing.assert_close(out, eager[idx], rtol=2e-3,
+                                       atol=2e-3)
+            idx += 1
+        assert e2.head == e1.head and e2.nproc == e1.nproc
+
+    def test_overlap_pipelined_constant_raw_ring_wrap(self):
+        """Fully-pipelined replays (no refresh, no per-trigger sync —
+        ingest(T+1) genuinely overlaps model(T)) across several ring turns
+        with constant raw: every trigger's output must equal the eager
+        twin's — any ingest/fill race would corrupt it."""
+        from tskd_amd.engine.stream_engine import TriggerGraph
+        from tskd_amd.models import build_model
+        from tskd_amd.ops import GraphedForward, MyCNNEngine
+        fs, S, G = 25.0, 4, 192
+        torch.manual_seed(12)
+        me = MyCNNEngine(build_model("MyCNN5").eval(), device="cuda")
+        chunk = torch.randn(S, 8, int(fs * 60), device="cuda",
+                            dtype=torch.bfloat16)
+        cm = list(range(8))
+        n_triggers = 56  # ~3.5 ring turns at 12 buckets/trigger, G=192
+        e1 = StreamEngine(S, 10, ring_grid=G, fs=fs, device="cuda")
+        eager = []
+        for _ in range(n_triggers):
+            e1.ingest_dense(chunk, chan_map=cm)
+            w = e1.windows(batch=1, stride=12, dtype=torch.bfloat16)
+            eager.append(me.forward(w, None, apply_sigmoid=True).clone())
+        e2 = StreamEngine(S, 10, ring_grid=G, fs=fs, device="cuda")
+        idx = 0
+        while e2.nproc == 0 or e2.nproc < e2.head - e2.win_buckets + 1:
+            e2.ingest_dense(chunk, chan_map=cm)
+            idx += 1
+        torch.cuda.synchronize()
+        gf = GraphedForward(me, s=S, n=1, dtype=torch.bfloat16, timelast=True)
+        gf.age.zero_()
+        tg = TriggerGraph(e2, chunk, cm, gf, stride=12, overlap=True)
+        idx += 1
+        outs = []
+        while idx < n_triggers:
+            outs.append((idx, tg.replay().clone()))  # clone on cur: ordered
+            idx += 1
+        torch.cuda.synchronize()
+        for i, out in outs:
+            torch.testing.assert_close(out, eager[i], rtol=2e-3, atol=2e-3)
+        assert e2.head == e1.head and e2.nproc == e1.nproc
+
+
 class TestZNormalize:
     def test_cpu_znorm(self):
         eng = _mk_engine(2, 3, 25.0)

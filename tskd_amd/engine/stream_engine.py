@@ -323,9 +323,11 @@ class StreamEngine:
             self.bcnt[:, :, idx] = 0
         self._cleared = upto_bucket
 
-    def ingest_dense_graph_body(self, raw: torch.Tensor, chan_map, nb: int
-                                 ) -> None:
-        """Capture-safe ingest + fill: kernels only, indices from dstate."""
+    def ingest_graph_kernel(self, raw: torch.Tensor, chan_map) -> None:
+        """Capture-safe dense ingest only (bucket writes; head from dstate).
+        Writes buckets [head, head+NB) — DISJOINT from anything the fill
+        stage of the PREVIOUS trigger reads, which is what makes the
+        two-stream overlapped TriggerGraph legal."""
         lib = _load_preproc_lib()
         cin, t = raw.shape[1], raw.shape[2]
         key = tuple(chan_map)
@@ -345,6 +347,12 @@ class StreamEngine:
             ctypes.c_long(0), self._dstate_ptr(), _sptr())
         if rc != 0:
             raise RuntimeError(f"ingest(graph) failed: {rc}")
+
+    def fill_graph_kernel(self, nb: int) -> None:
+        """Capture-safe window fill of nb new grid points (nproc from
+        dstate; reads buckets [nproc, nproc+nb+win-1) = through the buckets
+        the SAME trigger's ingest just wrote)."""
+        lib = _load_preproc_lib()
         rc = lib.tskd_preproc_window_fill(
             ctypes.c_void_p(self.bsum.data_ptr()),
             ctypes.c_void_p(self.bcnt.data_ptr()),
@@ -354,6 +362,12 @@ class StreamEngine:
             self.win_buckets, self._dstate_ptr(), _sptr())
         if rc != 0:
             raise RuntimeError(f"fill(graph) failed: {rc}")
+
+    def ingest_dense_graph_body(self, raw: torch.Tensor, chan_map, nb: int
+                                 ) -> None:
+        """Capture-safe ingest + fill: kernels only, indices from dstate."""
+        self.ingest_graph_kernel(raw, chan_map)
+        self.fill_graph_kernel(nb)
 
     # ----------------------------------------------------------------- process
     def _refill(self) -> None:
@@ -467,7 +481,8 @@ class TriggerGraph:
     """
 
     def __init__(self, engine: "StreamEngine", raw: torch.Tensor,
-                 chan_map, graphed_forward, stride: int = 12):
+                 chan_map, graphed_forward, stride: int = 12,
+                 overlap: bool = False):
         assert engine._gpu, "TriggerGraph needs a CUDA StreamEngine"
         cin, t = raw.shape[1], raw.shape[2]
         self.nb = t // engine.bucket_len
@@ -478,11 +493,15 @@ class TriggerGraph:
         self.gf = graphed_forward
         self.stride = stride
         self.chan_map = list(chan_map)
+        self.overlap = overlap
         dev = engine.device
         self.dstate = torch.tensor([engine.head, engine.nproc],
                                    dtype=torch.int64, device=dev)
         lib = _load_preproc_lib()
         self._lib = lib
+        if overlap:
+            self._init_overlap()
+            return
 
         def trigger_body():
             engine._dstate = self.dstate
@@ -517,8 +536,97 @@ class TriggerGraph:
         with torch.cuda.graph(self.graph):
             self.out = trigger_body()
 
-    def replay(self) -> torch.Tensor:
+    def _init_overlap(self) -> None:
+        """Two-stream software-pipelined trigger: ingest(T+1) overlaps the
+        fill/gather/model chain of trigger T on a second HIP stream.
+
+        Legality: ingest(T+1) writes buckets [head_T, head_T + NB) while
+        fill(T) reads [nproc_T, nproc_T + NB + win - 1) = up to head_T - 1
+        — disjoint ring regions (and G >> the combined span). The device
+        ring indices split: the ingest graph advances dstate[0] (head)
+        only, the model graph advances dstate[1] (nproc) only, so each
+        stream owns its own index. A CUDA event orders model(T) after
+        ingest(T); nothing orders ingest(T+1) after model(T).
+
+        Raises per-trigger latency not at all (probe with replay()+sync);
+        raises steady-state throughput by overlapping the HBM-bound ingest
+        with the compute-bound model chain.
+        """
+        engine, lib = self.engine, self._lib
+        nb = self.nb
+
+        def ingest_body():
+            engine._dstate = self.dstate
+            try:
+                engine.ingest_graph_kernel(self.raw, self.chan_map)
+                rc = lib.tskd_preproc_advance_state(
+                    ctypes.c_void_p(self.dstate.data_ptr()), nb, 0, _sptr())
+                if rc != 0:
+                    raise RuntimeError(f"advance_state(head): {rc}")
+            finally:
+                engine._dstate = None
+
+        def model_body():
+            engine._dstate = self.dstate
+            engine._dstate_gather_extra = nb
+            try:
+                engine.fill_graph_kernel(nb)
+                engine.windows(batch=1, stride=self.stride,
+                               dtype=self.gf.x.dtype, out=self.gf.x,
+                               timelast=getattr(self.gf, "timelast", False))
+                out = self.gf.engine.forward(self.gf.x, self.gf.age,
+                                             apply_sigmoid=True)
+                rc = lib.tskd_preproc_advance_state(
+                    ctypes.c_void_p(self.dstate.data_ptr()), 0, nb, _sptr())
+                if rc != 0:
+                    raise RuntimeError(f"advance_state(nproc): {rc}")
+                return out
+            finally:
+                engine._dstate = None
+                engine._dstate_gather_extra = 0
+
+        self._sA = torch.cuda.Stream()
+        self._sB = torch.cuda.Stream()
+        # warm one full trigger (ingest then model), mirroring host indices
+        with torch.cuda.stream(self._sA):
+            ingest_body()
+        self._sB.wait_stream(self._sA)
+        with torch.cuda.stream(self._sB):
+            model_body()
+        torch.cuda.current_stream().wait_stream(self._sB)
+        engine.head += nb
+        engine.nproc += nb
+        self.graph_i = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph_i, stream=self._sA):
+            ingest_body()
+        self.graph_m = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph_m, stream=self._sB):
+            self.out = model_body()
+        self._evA = torch.cuda.Event()
+
+    def replay(self, raw_refresh: bool = False) -> torch.Tensor:
         """One serving trigger: caller refreshed self.raw in place."""
+        if self.overlap:
+            # ingest(T) launches on stream A; model(T) on stream B waits it
+            # (event). The NEXT call's ingest(T+1) waits only stream A's own
+            # order — it overlaps model(T). Pass raw_refresh=True when the
+            # caller rewrote self.raw since the previous call: it orders
+            # ingest(T) after the caller's stream (serializing the pipeline
+            # but keeping the refresh race-free).
+            cur = torch.cuda.current_stream()
+            if raw_refresh:
+                self._sA.wait_stream(cur)
+            with torch.cuda.stream(self._sA):
+                self.graph_i.replay()
+                self._evA.record(self._sA)
+            self._sB.wait_event(self._evA)
+            self._sB.wait_stream(cur)   # previous output consumed on cur
+            with torch.cuda.stream(self._sB):
+                self.graph_m.replay()
+            cur.wait_stream(self._sB)   # consumers on cur see trigger T
+            self.engine.head += self.nb
+            self.engine.nproc += self.nb
+            return self.out
         self.graph.replay()
         self.engine.head += self.nb      # host mirrors (bookkeeping only)
         self.engine.nproc += self.nb
