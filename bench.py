@@ -77,11 +77,11 @@ def main() -> None:
     p.add_argument("--batch", type=int, default=1024,
                    help="[infer] windows per sequence batch")
     p.add_argument("--variant", default="MyCNN5")
-    p.add_argument("--no-overlap", action="store_true",
-                   help="[pipeline] disable the two-stream ingest/model "
-                        "overlap (trigger T+1 ingest overlaps trigger T "
-                        "model chain; per-trigger latency is probed and "
-                        "reported separately)")
+    p.add_argument("--overlap", action="store_true",
+                   help="[pipeline] two-stream ingest/model software "
+                        "pipelining (measured NEUTRAL: both stages fill "
+                        "all 256 CUs, so streams cannot co-schedule — "
+                        "kept as an A/B reference)")
     p.add_argument("--graph", action="store_true", default=True,
                    help="hipGraph-capture the model forward (pipeline mode)")
     p.add_argument("--no-graph", dest="graph", action="store_false")
@@ -177,7 +177,7 @@ def main() -> None:
                     se.ingest_dense(raw, chan_map=chan_map)
                 torch.cuda.synchronize()
                 tg = TriggerGraph(se, raw, chan_map, graphed, stride=12,
-                                  overlap=not args.no_overlap)
+                                  overlap=args.overlap)
             except Exception as e:  # pragma: no cover - fallback safety
                 print(f"[bench] hipGraph capture unavailable ({e}); "
                       "running eager", file=sys.stderr)
