@@ -753,9 +753,14 @@ int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
     const int NB = T / bucket_len;
     if (NB <= 0 || S <= 0) return 0;
     const long nwaves = (long)S * CIN * ((NB + ING_GRP - 1) / ING_GRP);
-    // read-ceiling probe: 32768 blocks measures ~3% above the 16384 cap
+    // read-ceiling probe: 32768 blocks measures ~3% above the 16384 cap.
+    // TSKD_INGEST_GRID caps the block count: the probe saturates HBM from
+    // ~2048 blocks, so a small grid leaves CU wave slots free for a
+    // co-scheduled model chain (TriggerGraph overlap mode).
+    long cap = 32768;
+    if (const char* gc = getenv("TSKD_INGEST_GRID")) cap = atol(gc);
     long blocks = (nwaves + 3) / 4;
-    if (blocks > 32768) blocks = 32768;
+    if (blocks > cap) blocks = cap;
     if (blocks < 1) blocks = 1;
     const int grid = (int)blocks;
     // TSKD_INGEST_ILP: 0 = 1-deep select loop, 1 (default) = 5-deep ILP +
