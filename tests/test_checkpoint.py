@@ -147,3 +147,29 @@ class TestGoldenWindow:
             y = torch.sigmoid(m(xt, torch.full((1,), 65.0)))
         golden = float(open("tests/data/x_testinput.golden").read())
         assert abs(float(y[0]) - golden) < 1e-6
+
+
+class TestRobustness:
+    def test_truncated_file_raises_cleanly(self, tmp_path):
+        from tskd_amd.models import build_model, load_checkpoint, \
+            save_checkpoint
+        p = str(tmp_path / "m.pth")
+        save_checkpoint(build_model("MyCNN5"), p)
+        blob = open(p, "rb").read()
+        open(p, "wb").write(blob[:len(blob) // 3])
+        with pytest.raises(Exception):  # clean error, not a crash/hang
+            load_checkpoint(p)
+
+    def test_zipfile_format_also_loads(self, tmp_path):
+        """A checkpoint written with torch's DEFAULT zipfile serialization
+        (what a user's own training script would produce) must load too —
+        not only the reference's legacy format."""
+        import torch
+        from tskd_amd.models import build_model, load_checkpoint
+        m = build_model("MyCNN4")
+        p = str(tmp_path / "zip.pth")
+        torch.save(m, p)  # zipfile format, tskd_amd GLOBALs
+        m2 = load_checkpoint(p)
+        assert type(m2).__name__ == "MyCNN4"
+        for a, b in zip(m.state_dict().values(), m2.state_dict().values()):
+            torch.testing.assert_close(a, b)
