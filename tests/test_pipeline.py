@@ -343,6 +343,33 @@ class TestFusedServe:
             assert len(pts) == max_emit
             assert all(np.isfinite(pts))
 
+    def test_pipelined_mode_defers_then_matches(self, tmp_path, cfg):
+        """pipelined=True: the bus poll of trigger T+1 overlaps trigger T's
+        GPU work, so persistence lags one trigger — and after flush() the
+        scores equal the unpipelined server's exactly."""
+        from tskd_amd.cli.serve import FusedServer
+        bus = Bus(str(tmp_path / "bus"))
+        _write_wfdb_record(cfg.wavef_path, "p000194-test", cfg.channel_names,
+                           1 / 60, 40)
+        torch.manual_seed(3)
+        from tskd_amd.models import build_model
+        model = build_model("MyCNN5").eval()
+        s1 = FusedServer(bus, cfg, PredictionStore(str(tmp_path / "a.log")),
+                         model=model, device="cpu", max_streams=8,
+                         ring_grid=1024, starting="earliest")
+        s2 = FusedServer(bus, cfg, PredictionStore(str(tmp_path / "b.log")),
+                         model=model, device="cpu", max_streams=8,
+                         ring_grid=1024, starting="earliest", pipelined=True)
+        send_record_data(bus, "p000194-test", None, 1e6, 1.0, cfg)
+        assert s1.trigger() == 1 and s1.store.count() == 1
+        assert s2.trigger() == 1
+        assert s2.store.count() == 0          # deferred one trigger
+        s2.flush()
+        assert s2.store.count() == 1
+        _, r1 = s1.store.latest("p000194")
+        _, r2 = s2.store.latest("p000194")
+        assert abs(r1 - r2) < 1e-6
+
     def test_hot_reload(self, tmp_path, cfg):
         """Swapping the checkpoint file between triggers changes the served
         model (the reference loads once at start and never reloads)."""

@@ -51,7 +51,8 @@ class FusedServer:
                  ring_grid: int = 4096, starting: str = "latest",
                  response_topic: Optional[str] = None,
                  emit_processed: Optional[str] = None,
-                 rank: int = 0, world: int = 1):
+                 rank: int = 0, world: int = 1,
+                 pipelined: bool = False):
         self.cfg = cfg
         self.store = store
         self.ages = ages or AgeTable()
@@ -87,6 +88,13 @@ class FusedServer:
         self.n_predictions = 0
         self.model_path: Optional[str] = None   # set to enable hot reload
         self._model_mtime = 0.0
+        # pipelined mode: the bus poll of trigger T+1 overlaps trigger T's
+        # asynchronously-launched GPU work; T's predictions persist at the
+        # START of trigger T+1 (or at flush()). One trigger of persistence
+        # lag bought for consume/compute overlap (north-star: "inference
+        # overlapped with Kafka consume").
+        self.pipelined = pipelined
+        self._pending = None  # (probs_dev, t_us, pids_snapshot)
 
     def maybe_reload_model(self) -> bool:
         """Hot-reload the checkpoint when the file changes (the reference
@@ -124,6 +132,8 @@ class FusedServer:
         with self.timer:
             keys, _t, chans, vals, ts = self.consumer.poll_samples(
                 max_msgs=131072, timeout_ms=0)
+            self.flush()  # pipelined: persist trigger T-1 AFTER the poll
+                          # overlapped its GPU tail (no-op otherwise)
             si, ci, tt, vv = [], [], [], []
             for i, k in enumerate(keys):
                 sid = self._sid(k)
@@ -187,21 +197,37 @@ class FusedServer:
                 pad = torch.zeros(self.max_streams, device=probs.device)
                 pad[:n_active] = probs
                 self.last_gathered = all_gather_predictions(pad)
-            if self.store is not None:
-                # the mmap store is multi-process safe: each rank persists
-                # its own shard directly (no gather needed for durability)
-                local = probs.cpu()
-                self.store.insert_batch(self.pids,
-                                        [t_us] * n_active,
-                                        local.tolist())
-                self.n_predictions += n_active
-            if self.producer and self.response_topic:
-                for i, pid in enumerate(self.pids):
-                    self.producer.produce(
-                        self.response_topic, pid,
-                        f'{{"t_us": {t_us}, "risk": {float(probs[i]):.6f}}}')
+            if self.pipelined:
+                # defer the device sync: persist at the next trigger start,
+                # after the bus poll has overlapped this trigger's GPU work
+                self._pending = (probs, t_us, list(self.pids))
+            else:
+                self._persist(probs, t_us, self.pids)
             self.timer.add_items(n_active)
             return n_active
+
+    def _persist(self, probs, t_us: int, pids) -> None:
+        n = len(pids)
+        if self.store is not None:
+            # the mmap store is multi-process safe: each rank persists
+            # its own shard directly (no gather needed for durability)
+            local = probs.reshape(-1)[:n].cpu()
+            self.store.insert_batch(pids, [t_us] * n, local.tolist())
+            self.n_predictions += n
+        if self.producer and self.response_topic:
+            vals = probs.reshape(-1)[:n].cpu()
+            for i, pid in enumerate(pids):
+                self.producer.produce(
+                    self.response_topic, pid,
+                    f'{{"t_us": {t_us}, "risk": {float(vals[i]):.6f}}}')
+
+    def flush(self) -> None:
+        """Persist the deferred trigger (pipelined mode; call on shutdown
+        or whenever durability must catch up to compute)."""
+        if self._pending is not None:
+            probs, t_us, pids = self._pending
+            self._pending = None
+            self._persist(probs, t_us, pids)
 
 
 
@@ -258,6 +284,9 @@ def main(argv=None) -> None:
                     help="reload the checkpoint when the file changes")
     ap.add_argument("--metrics-port", type=int, default=0,
                     help="expose Prometheus /metrics on 127.0.0.1:PORT")
+    ap.add_argument("--pipelined", action="store_true",
+                    help="overlap the bus poll of trigger T+1 with trigger "
+                         "T's GPU work (persistence lags one trigger)")
     args = ap.parse_args(argv)
 
     rank, world = init_distributed()
@@ -276,7 +305,7 @@ def main(argv=None) -> None:
                       starting=args.starting,
                       response_topic=args.model_response_topic,
                       emit_processed=args.emit_processed,
-                      rank=rank, world=world)
+                      rank=rank, world=world, pipelined=args.pipelined)
     if args.hot_reload and args.model_path:
         srv.model_path = args.model_path
         if os.path.exists(args.model_path):
@@ -300,6 +329,7 @@ def main(argv=None) -> None:
         if args.max_triggers and n >= args.max_triggers:
             break
         time.sleep(max(0.0, period - (time.time() - t0)))
+    srv.flush()  # pipelined mode: persist the final deferred trigger
 
 
 if __name__ == "__main__":
