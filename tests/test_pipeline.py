@@ -278,16 +278,19 @@ class TestFusedServe:
         assert abs(risk_fused - risk_two_stage) < 1e-4
 
     @pytest.mark.gpu
-    def test_fused_server_gpu(self, tmp_path, cfg):
+    @pytest.mark.parametrize("pipelined", [False, True])
+    def test_fused_server_gpu(self, tmp_path, cfg, pipelined):
         from tskd_amd.cli.serve import FusedServer
         bus = Bus(str(tmp_path / "bus"))
         _write_wfdb_record(cfg.wavef_path, "p000194-test", cfg.channel_names,
                            1 / 60, 40)
         store = PredictionStore(str(tmp_path / "pred.log"))
         srv = FusedServer(bus, cfg, store, device="cuda", max_streams=8,
-                          ring_grid=1024, starting="earliest")
+                          ring_grid=1024, starting="earliest",
+                          pipelined=pipelined)
         send_record_data(bus, "p000194-test", None, 1e6, 1.0, cfg)
         n = srv.trigger()
+        srv.flush()  # pipelined: persist the deferred trigger (no-op else)
         torch.cuda.synchronize()
         assert n == 1 and store.count() == 1
         _, risk = store.latest("p000194")
