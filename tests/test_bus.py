@@ -334,3 +334,35 @@ class TestTcpRelay:
             rp.close()
         finally:
             srv.close()
+
+    def test_sendstream_over_relay(self, bus, tmp_path):
+        """The replay producer streams a WFDB record to a REMOTE node's bus
+        via --relay host:port (cross-host ingest, end to end)."""
+        from tskd_amd.bus.relay import RelayServer
+        from tskd_amd.cli import sendstream
+        from tests.test_pipeline import _write_wfdb_record
+        from tskd_amd.config import GlobalConfig
+        cfg = GlobalConfig()
+        cfg.wavef_path = str(tmp_path / "wavef")
+        cfg.channel_names = ["HR", "RESP"]
+        cfg.patient_records = ["p000194-test"]
+        _write_wfdb_record(cfg.wavef_path, "p000194-test", cfg.channel_names,
+                           1 / 60, 8)
+        srv = RelayServer(bus, port=0)
+        try:
+            import unittest.mock as mock
+            with mock.patch("tskd_amd.cli.sendstream.get_global_config",
+                            return_value=cfg):
+                sendstream.main(["--relay", f"127.0.0.1:{srv.port}",
+                                 "--speed", "1e6"])
+            c = Consumer(bus, starting="earliest")
+            c.subscribe(["HR", "RESP"])
+            got = []
+            for _ in range(50):
+                got += c.poll(max_msgs=64, timeout_ms=100)
+                if len(got) == 16:
+                    break
+            assert len(got) == 16  # 8 samples x 2 channels
+            assert {m.topic for m in got} == {"HR", "RESP"}
+        finally:
+            srv.close()

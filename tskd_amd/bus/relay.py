@@ -149,6 +149,10 @@ class RelayProducer:
         _send_frame(self._sock, {"op": "produce", "topic": topic,
                                  "key": key, "value": value, "ts_us": ts_us})
 
+    def flush(self) -> None:
+        """sendall() is already synchronous; durability is the server's
+        local log (drop-in for Producer.flush in the replay CLIs)."""
+
     def close(self) -> None:
         self._sock.close()
 

@@ -29,8 +29,10 @@ log = logging.getLogger("sendStream")
 
 
 def send_record_data(bus: Bus, record_id: str, signal_list: Optional[List[str]],
-                     speed: float, frequency: float, cfg) -> int:
-    producer = Producer(bus, retries=cfg.producer_retries)
+                     speed: float, frequency: float, cfg,
+                     producer_factory=None) -> int:
+    producer = producer_factory() if producer_factory else \
+        Producer(bus, retries=cfg.producer_retries)
     patient_id = record_id[0:7]
     path = get_waveform_path(record_id, cfg)
     record = rdrecord(path, channel_names=signal_list or cfg.channel_names)
@@ -57,9 +59,10 @@ def send_record_data(bus: Bus, record_id: str, signal_list: Optional[List[str]],
 
 
 def send_csv_data(bus: Bus, csv_path: str, topic: str, key: str,
-                  speed: float, cfg) -> int:
+                  speed: float, cfg, producer_factory=None) -> int:
     """data.csv replay mode (the upstream demo lineage: timestamp,value)."""
-    producer = Producer(bus, retries=cfg.producer_retries)
+    producer = producer_factory() if producer_factory else \
+        Producer(bus, retries=cfg.producer_retries)
     data = np.genfromtxt(csv_path, delimiter=",", names=True,
                          dtype=None, encoding="utf-8")
     sent = 0
@@ -99,6 +102,9 @@ def main(argv=None) -> None:
                     help="also log to this file (the reference writes "
                          "${DATAPATH}/producer.log, sendStream.py:15-21)")
     ap.add_argument("--topic", default="data", help="[csv mode] topic")
+    ap.add_argument("--relay", default=None, metavar="HOST:PORT",
+                    help="produce over TCP to a remote node's bus relay "
+                         "instead of a local bus directory")
     args = ap.parse_args(argv)
     if args.log_file:
         fh = logging.FileHandler(args.log_file)
@@ -106,9 +112,17 @@ def main(argv=None) -> None:
             "%(asctime)s %(name)s %(levelname)s %(message)s"))
         logging.getLogger().addHandler(fh)
 
-    bus = Bus(args.bus_dir)
+    factory = None
+    bus = None
+    if args.relay:
+        from tskd_amd.bus.relay import RelayProducer
+        host, _, port = args.relay.rpartition(":")
+        factory = lambda: RelayProducer(host or "127.0.0.1", int(port))  # noqa: E731
+    else:
+        bus = Bus(args.bus_dir)
     if args.csv:
-        n = send_csv_data(bus, args.csv, args.topic, "csv", args.speed, cfg)
+        n = send_csv_data(bus, args.csv, args.topic, "csv", args.speed, cfg,
+                          producer_factory=factory)
         log.info("csv replay done: %d messages", n)
         return
 
@@ -116,7 +130,8 @@ def main(argv=None) -> None:
     threads = [
         threading.Thread(target=send_record_data,
                          args=(bus, r, args.signal_list, args.speed,
-                               args.frequency, cfg), daemon=True)
+                               args.frequency, cfg),
+                         kwargs={"producer_factory": factory}, daemon=True)
         for r in records
     ]
     for t in threads:
