@@ -335,6 +335,32 @@ class TestTcpRelay:
         finally:
             srv.close()
 
+    def test_relay_unicode_and_large_values(self, bus):
+        """Framing robustness: multi-frame-sized values and non-ASCII keys
+        round-trip exactly."""
+        from tskd_amd.bus.relay import (RelayConsumer, RelayProducer,
+                                        RelayServer)
+        srv = RelayServer(bus, port=0)
+        try:
+            rp = RelayProducer("127.0.0.1", srv.port)
+            big = "x" * 300000  # > one socket buffer
+            rp.produce("t", "péti€nt-1", big)
+            rp.produce("t", "k", "sm\u00e5ll")
+            rc = RelayConsumer("127.0.0.1", srv.port, ["t"],
+                               starting="earliest")
+            frames = []
+            for _ in range(50):
+                frames += rc.poll(max_msgs=16, timeout_s=0.5)
+                if len(frames) == 2:
+                    break
+            assert frames[0]["key"] == "péti€nt-1"
+            assert frames[0]["value"] == big
+            assert frames[1]["value"] == "sm\u00e5ll"
+            rc.close()
+            rp.close()
+        finally:
+            srv.close()
+
     def test_sendstream_over_relay(self, bus, tmp_path):
         """The replay producer streams a WFDB record to a REMOTE node's bus
         via --relay host:port (cross-host ingest, end to end)."""
