@@ -60,7 +60,9 @@ def build_step(mode: str, device, seqs: int, batch: int, streams: int):
 
         def step():
             se.ingest_dense(raw, chan_map=list(range(8)))
-            w = se.windows(batch=1, stride=12, dtype=torch.bfloat16)
+            # timelast: the serving engine's production window layout
+            w = se.windows(batch=1, stride=12, dtype=torch.bfloat16,
+                           timelast=True)
             eng.forward(w, age, apply_sigmoid=True)
         return step
     raise ValueError(mode)
