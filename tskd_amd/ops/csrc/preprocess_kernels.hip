@@ -636,13 +636,35 @@ __global__ __launch_bounds__(256) void window_fill16_kernel(
             if (gl < np) {
                 const float* bs = bsum + sc * G;
                 const float* bc = bcnt + sc * G;
-                float sum = 0.f, cnt = 0.f;
-                int idx = (int)((phead + gl) % G);
-                for (int k = 0; k < win_buckets; ++k) {
-                    sum += bs[idx];
-                    cnt += bc[idx];
-                    if (++idx == G) idx = 0;
+                // split the (possibly wrapping) 36-bucket window into two
+                // linear segments and sum with 4 independent accumulators:
+                // the single dependent add chain was the lane's critical
+                // path (the loads are L1/L2 hits shared across the group)
+                const int start = (int)((phead + gl) % G);
+                const int first = min(win_buckets, G - start);
+                const int rem = win_buckets - first;
+                float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+                float c0 = 0.f, c1 = 0.f, c2 = 0.f, c3 = 0.f;
+                const float* p1 = bs + start;
+                const float* q1 = bc + start;
+                int k = 0;
+                for (; k + 4 <= first; k += 4) {
+                    a0 += p1[k];     c0 += q1[k];
+                    a1 += p1[k + 1]; c1 += q1[k + 1];
+                    a2 += p1[k + 2]; c2 += q1[k + 2];
+                    a3 += p1[k + 3]; c3 += q1[k + 3];
                 }
+                for (; k < first; ++k) { a0 += p1[k]; c0 += q1[k]; }
+                int j = 0;
+                for (; j + 4 <= rem; j += 4) {
+                    a0 += bs[j];     c0 += bc[j];
+                    a1 += bs[j + 1]; c1 += bc[j + 1];
+                    a2 += bs[j + 2]; c2 += bc[j + 2];
+                    a3 += bs[j + 3]; c3 += bc[j + 3];
+                }
+                for (; j < rem; ++j) { a0 += bs[j]; c0 += bc[j]; }
+                const float sum = (a0 + a1) + (a2 + a3);
+                const float cnt = (c0 + c1) + (c2 + c3);
                 val = (cnt > 0.f) ? sum / cnt : NAN;
             }
         }
