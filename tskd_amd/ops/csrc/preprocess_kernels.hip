@@ -120,8 +120,12 @@ __global__ void ingest_dense_kernel(
                 }
                 const int oct = (bucket_len - pre) / 8;
                 const u32x4_* vp = (const u32x4_*)((const unsigned short*)src + pre);
-                if constexpr (IMODE == 1) {
-                    // simple 5-deep ILP with per-element NaN selects
+                if constexpr (IMODE == 1 || IMODE == 4) {
+                    // 5-deep load ILP with per-element NaN selects; mode 4
+                    // splits the accumulation across two chains (even/odd
+                    // element) — the single 40-add chain is the lane's
+                    // critical path once loads are in flight
+                    float s1 = 0.f, n1 = 0.f;
                     for (int base = lane; base < oct; base += 5 * ING_GL) {
                         union { u32x4_ q; unsigned short h[8]; } v[5];
                         #pragma unroll
@@ -136,11 +140,20 @@ __global__ void ingest_dense_kernel(
                                 #pragma unroll
                                 for (int j = 0; j < 8; ++j) {
                                     const float f = bf16_to_f32_(v[u].h[j]);
+                                    if constexpr (IMODE == 4) {
+                                        if (j & 1) {
+                                            if (!isnan(f)) { s1 += f;
+                                                             n1 += 1.f; }
+                                            continue;
+                                        }
+                                    }
                                     if (!isnan(f)) { sum += f; cnt += 1.f; }
                                 }
                             }
                         }
                     }
+                    sum += s1;
+                    cnt += n1;
                 } else if constexpr (IMODE == 2) {
                     // 5 independent nontemporal loads in flight per lane
                     // (read-ceiling probe: 1-deep streams measure ~5.9 TB/s,
@@ -849,6 +862,11 @@ int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
                                dstate);
         else if (mode == 1)
             hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, 1>),
+                               dim3(grid), dim3(256), 0, st, rp, bsum, bcnt,
+                               chan_map, S, CIN, C, T, G, bucket_len, head,
+                               dstate);
+        else if (mode == 4)
+            hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, 4>),
                                dim3(grid), dim3(256), 0, st, rp, bsum, bcnt,
                                chan_map, S, CIN, C, T, G, bucket_len, head,
                                dstate);
