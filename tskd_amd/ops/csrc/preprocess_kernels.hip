@@ -111,7 +111,46 @@ __global__ void ingest_dense_kernel(
         // 16B-aligned vector body with scalar head/tail peel (bucket offsets
         // like 625 samples are not 16B-aligned).
         if (active) {
-            if constexpr (sizeof(DT) == 2) {
+            if constexpr (IMODE == 5 && sizeof(DT) == 2) {
+                // peel-free: cover the bucket with ALIGNED oct-chunks and
+                // mask the (at most two) boundary chunks in-register —
+                // removes both scalar peel loops per bucket
+                const unsigned short* su = (const unsigned short*)src;
+                const int mis = (int)(((size_t)su & 15) / 2);  // elems before
+                const u32x4_* vp = (const u32x4_*)(su - mis);
+                const int nch = (mis + bucket_len + 7) / 8;
+                for (int base = lane; base < nch; base += 5 * ING_GL) {
+                    union { u32x4_ q; unsigned short h[8]; } v[5];
+                    #pragma unroll
+                    for (int u = 0; u < 5; ++u) {
+                        const int pp = base + u * ING_GL;
+                        v[u].q = __builtin_nontemporal_load(
+                            &vp[pp < nch ? pp : 0]);
+                    }
+                    #pragma unroll
+                    for (int u = 0; u < 5; ++u) {
+                        const int pp = base + u * ING_GL;
+                        if (pp >= nch) continue;
+                        const int idx0 = pp * 8 - mis;  // sample idx of h[0]
+                        if (idx0 >= 0 && idx0 + 8 <= bucket_len) {
+                            #pragma unroll
+                            for (int j = 0; j < 8; ++j) {
+                                const float f = bf16_to_f32_(v[u].h[j]);
+                                if (!isnan(f)) { sum += f; cnt += 1.f; }
+                            }
+                        } else {  // boundary chunk: mask by sample index
+                            #pragma unroll
+                            for (int j = 0; j < 8; ++j) {
+                                const unsigned k = (unsigned)(idx0 + j);
+                                if (k < (unsigned)bucket_len) {
+                                    const float f = bf16_to_f32_(v[u].h[j]);
+                                    if (!isnan(f)) { sum += f; cnt += 1.f; }
+                                }
+                            }
+                        }
+                    }
+                }
+            } else if constexpr (sizeof(DT) == 2) {
                 int pre = (int)(((16 - ((size_t)src & 15)) & 15) / 2);
                 if (pre > bucket_len) pre = bucket_len;
                 for (int i = lane; i < pre; i += ING_GL) {
@@ -867,6 +906,11 @@ int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
                                dstate);
         else if (mode == 4)
             hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, 4>),
+                               dim3(grid), dim3(256), 0, st, rp, bsum, bcnt,
+                               chan_map, S, CIN, C, T, G, bucket_len, head,
+                               dstate);
+        else if (mode == 5)
+            hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, 5>),
                                dim3(grid), dim3(256), 0, st, rp, bsum, bcnt,
                                chan_map, S, CIN, C, T, G, bucket_len, head,
                                dstate);
