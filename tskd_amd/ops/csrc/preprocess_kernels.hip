@@ -118,7 +118,11 @@ __global__ void ingest_dense_kernel(
                 const unsigned short* su = (const unsigned short*)src;
                 const int mis = (int)(((size_t)su & 15) / 2);  // elems before
                 const u32x4_* vp = (const u32x4_*)(su - mis);
-                const int nch = (mis + bucket_len + 7) / 8;
+                // floor: every chunk stays inside [su - mis, su + bucket_len)
+                // (the head read-back is covered by the previous bucket/row;
+                // a CEIL here would read past the raw tensor's final byte);
+                // the < 8 leftover samples take the scalar tail below
+                const int nch = (mis + bucket_len) / 8;
                 for (int base = lane; base < nch; base += 5 * ING_GL) {
                     union { u32x4_ q; unsigned short h[8]; } v[5];
                     #pragma unroll
@@ -149,6 +153,11 @@ __global__ void ingest_dense_kernel(
                             }
                         }
                     }
+                }
+                for (int i = nch * 8 - mis + lane; i < bucket_len;
+                     i += ING_GL) {  // < 8 samples past the last full chunk
+                    const float f = bf16_to_f32_(su[i]);
+                    if (!isnan(f)) { sum += f; cnt += 1.f; }
                 }
             } else if constexpr (sizeof(DT) == 2) {
                 int pre = (int)(((16 - ((size_t)src & 15)) & 15) / 2);
