@@ -75,7 +75,7 @@ __device__ __forceinline__ void wsync_() {
 // busy (78/16 = 4.9 balanced iterations per group).
 #define ING_GRP 4   // buckets per wave
 #define ING_GL 16   // lanes per bucket group
-template <class DT, int IMODE = 1>
+template <class DT, int IMODE = 5>
 __global__ void ingest_dense_kernel(
     const DT* __restrict__ raw,     // (S, CIN, T)
     float* __restrict__ bsum,       // (S, C, G)
@@ -896,11 +896,11 @@ int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
     if (blocks > cap) blocks = cap;
     if (blocks < 1) blocks = 1;
     const int grid = (int)blocks;
-    // TSKD_INGEST_ILP: 0 = 1-deep select loop, 1 (default) = 5-deep ILP +
-    // selects, 2 = 5-deep ILP + deferred-NaN packed adds. Within-run A/B:
-    // 1 beats 0 by 4.6% and 2 by 1.3% (the NaN-select chain is NOT the
-    // limiter; the deferred check's extra group reduce costs more).
-    int mode = 1;
+    // TSKD_INGEST_ILP: 0 = 1-deep select loop, 1 = 5-deep ILP + scalar
+    // peel, 2 = deferred-NaN packed, 3 = paired-bucket, 4 = dual-acc,
+    // 5 (default, bf16) = 5-deep ILP + masked boundary chunks (no scalar
+    // peel; within-run A/B: beats 1 by 6.9%; fp32 falls back to 1).
+    int mode = 5;
     if (const char* ilp = getenv("TSKD_INGEST_ILP")) mode = atoi(ilp);
     if (raw_is_bf16) {
         const unsigned short* rp = (const unsigned short*)raw;
@@ -937,6 +937,7 @@ int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
                                dstate);
     } else {
         const float* rp = (const float*)raw;
+        if (mode >= 3) mode = 1;  // fp32: modes 3-5 are bf16-only
         if (mode == 0)
             hipLaunchKernelGGL((ingest_dense_kernel<float, 0>), dim3(grid),
                                dim3(256), 0, st, rp, bsum, bcnt, chan_map, S,
