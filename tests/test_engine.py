@@ -215,6 +215,27 @@ class TestStreamEngineGPU:
         np.testing.assert_allclose(wg.cpu().numpy(), wc.numpy(), rtol=1e-4,
                                    atol=1e-5)
 
+    def test_gpu_bf16_dense_with_nans_matches_cpu(self):
+        """bf16 dense ingest takes the masked-boundary kernel (mode 5) —
+        its NaN-skip/count path must match the CPU oracle fed the same
+        bf16-rounded samples (10% NaNs: dropped samples, like wfdb
+        invalids)."""
+        S, C, fs = 4, 10, 125.0
+        T = int(fs * 60 * 8)
+        raw = _dense_raw(S, 8, T, seed=17, nan_frac=0.1).to(torch.bfloat16)
+        cpu = StreamEngine(S, C, ring_grid=1024, fs=fs, device="cpu")
+        gpu = StreamEngine(S, C, ring_grid=1024, fs=fs, device="cuda")
+        cm = list(range(8))
+        cpu.ingest_dense(raw.float(), chan_map=cm)  # same rounded values
+        gpu.ingest_dense(raw.cuda(), chan_map=cm)
+        torch.cuda.synchronize()
+        assert gpu.nproc == cpu.nproc
+        np.testing.assert_allclose(
+            gpu.bcnt.cpu().numpy(), cpu.bcnt.numpy())  # NaN counts EXACT
+        np.testing.assert_allclose(
+            gpu.proc[:, :, :gpu.nproc].cpu().numpy(),
+            cpu.proc[:, :, :cpu.nproc].numpy(), rtol=2e-3, atol=1e-3)
+
     def test_gpu_bf16_windows_and_pipeline(self):
         # Full fused pipeline: raw -> preprocess -> windows -> MyCNN5 engine.
         from tskd_amd.models import build_model
