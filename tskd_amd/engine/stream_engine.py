@@ -22,6 +22,8 @@ import ctypes
 from typing import Optional, Sequence
 
 import numpy as np
+import os
+
 import torch
 
 from tskd_amd.engine import windowing as W
@@ -45,7 +47,7 @@ def _load_preproc_lib() -> ctypes.CDLL:
         ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int,
         ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_long,
-        ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
     ]
     lib.tskd_preproc_advance_state.restype = ctypes.c_int
     lib.tskd_preproc_advance_state.argtypes = [
@@ -55,18 +57,20 @@ def _load_preproc_lib() -> ctypes.CDLL:
     lib.tskd_preproc_ingest_events.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
-        ctypes.c_long, ctypes.c_long, ctypes.c_void_p,
+        ctypes.c_long, ctypes.c_long, ctypes.c_int, ctypes.c_void_p,
     ]
     lib.tskd_preproc_clear_buckets.restype = ctypes.c_int
     lib.tskd_preproc_clear_buckets.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
-        ctypes.c_int, ctypes.c_long, ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_int, ctypes.c_long, ctypes.c_int, ctypes.c_int,
+        ctypes.c_void_p,
     ]
     lib.tskd_preproc_window_fill.restype = ctypes.c_int
     lib.tskd_preproc_window_fill.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_long,
-        ctypes.c_int, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_int, ctypes.c_int, ctypes.c_void_p, ctypes.c_int,
+        ctypes.c_void_p,
     ]
     lib.tskd_preproc_window_znorm.restype = ctypes.c_int
     lib.tskd_preproc_window_znorm.argtypes = [
@@ -112,11 +116,23 @@ class StreamEngine:
         self.nproc = 0   # processed grid points produced
         self._cleared = 0  # ring slots zeroed up to this bucket (event path)
         dev = self.device
-        self.bsum = torch.zeros(self.S, self.C, self.G, device=dev)
-        self.bcnt = torch.zeros(self.S, self.C, self.G, device=dev)
+        # Packed layout (GPU): (sum, cnt) interleaved as adjacent floats so
+        # window_fill reads each bucket with ONE f32x2 load. bsum/bcnt stay
+        # the public (S, C, G) views; kernels take the element stride.
+        self._gpu = dev.type == "cuda"
+        self._packed = self._gpu and \
+            os.environ.get("TSKD_PACKED_BUCKETS", "1") != "0"
+        if self._packed:
+            self._bkt = torch.zeros(self.S, self.C, self.G, 2, device=dev)
+            self.bsum = self._bkt[..., 0]
+            self.bcnt = self._bkt[..., 1]
+            self._bst = 2
+        else:
+            self.bsum = torch.zeros(self.S, self.C, self.G, device=dev)
+            self.bcnt = torch.zeros(self.S, self.C, self.G, device=dev)
+            self._bst = 1
         self.proc = torch.zeros(self.S, self.C, self.G, device=dev)
         self.last_val = torch.full((self.S, self.C), float("nan"), device=dev)
-        self._gpu = dev.type == "cuda"
         self._dstate = None  # device [head, nproc] during graph capture/replay
         self._dstate_gather_extra = 0
         if self._gpu:
@@ -160,7 +176,8 @@ class StreamEngine:
                 ctypes.c_void_p(self.bcnt.data_ptr()),
                 ctypes.c_void_p(cm.data_ptr()),
                 self.S, cin, self.C, t, self.G, self.bucket_len,
-                ctypes.c_long(self.head), self._dstate_ptr(), _sptr())
+                ctypes.c_long(self.head), self._dstate_ptr(), self._bst,
+                _sptr())
             if rc != 0:
                 raise RuntimeError(f"ingest_dense failed: hipError {rc}")
             self._clear_stale_channels(chan_map, nb)
@@ -235,7 +252,7 @@ class StreamEngine:
                 ctypes.c_void_p(self.bsum.data_ptr()),
                 ctypes.c_void_p(self.bcnt.data_ptr()),
                 self.C, self.G, ctypes.c_long(len(vi)),
-                ctypes.c_long(min_bucket), _sptr())
+                ctypes.c_long(min_bucket), self._bst, _sptr())
             if rc != 0:
                 raise RuntimeError(f"ingest_events failed: hipError {rc}")
             if advance_to is None and len(tsd):
@@ -314,7 +331,8 @@ class StreamEngine:
             rc = lib.tskd_preproc_clear_buckets(
                 ctypes.c_void_p(self.bsum.data_ptr()),
                 ctypes.c_void_p(self.bcnt.data_ptr()),
-                self.S, self.C, self.G, ctypes.c_long(start), nb, _sptr())
+                self.S, self.C, self.G, ctypes.c_long(start), nb,
+                self._bst, _sptr())
             if rc != 0:
                 raise RuntimeError(f"clear_buckets failed: hipError {rc}")
         else:
@@ -344,7 +362,7 @@ class StreamEngine:
             ctypes.c_void_p(self.bcnt.data_ptr()),
             ctypes.c_void_p(cm.data_ptr()),
             self.S, cin, self.C, t, self.G, self.bucket_len,
-            ctypes.c_long(0), self._dstate_ptr(), _sptr())
+            ctypes.c_long(0), self._dstate_ptr(), self._bst, _sptr())
         if rc != 0:
             raise RuntimeError(f"ingest(graph) failed: {rc}")
 
@@ -359,7 +377,7 @@ class StreamEngine:
             ctypes.c_void_p(self.proc.data_ptr()),
             ctypes.c_void_p(self.last_val.data_ptr()),
             self.S, self.C, self.G, ctypes.c_long(0), nb,
-            self.win_buckets, self._dstate_ptr(), _sptr())
+            self.win_buckets, self._dstate_ptr(), self._bst, _sptr())
         if rc != 0:
             raise RuntimeError(f"fill(graph) failed: {rc}")
 
@@ -384,7 +402,7 @@ class StreamEngine:
                 ctypes.c_void_p(self.proc.data_ptr()),
                 ctypes.c_void_p(self.last_val.data_ptr()),
                 self.S, self.C, self.G, ctypes.c_long(self.nproc), np_new,
-                self.win_buckets, self._dstate_ptr(), _sptr())
+                self.win_buckets, self._dstate_ptr(), self._bst, _sptr())
             if rc != 0:
                 raise RuntimeError(f"window_fill failed: hipError {rc}")
         else:

@@ -60,6 +60,8 @@ __global__ void advance_state_kernel(long long* dstate, int nb, int np) {
     }
 }
 
+typedef float f32x2p_ __attribute__((ext_vector_type(2)));
+
 // Same-wave LDS RAW fence (see mycnn_kernels.hip wave_sync).
 __device__ __forceinline__ void wsync_() {
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -78,12 +80,12 @@ __device__ __forceinline__ void wsync_() {
 template <class DT, int IMODE = 5>
 __global__ void ingest_dense_kernel(
     const DT* __restrict__ raw,     // (S, CIN, T)
-    float* __restrict__ bsum,       // (S, C, G)
-    float* __restrict__ bcnt,
+    float* __restrict__ bsum,       // (S, C, G), element stride BST
+    float* __restrict__ bcnt,       // (= bsum+1 when packed (sum,cnt) pairs)
     const int* __restrict__ chan_map,  // (CIN) raw row -> wire channel
     int S, int CIN, int C, int T, int G,
     int bucket_len, long head_in,   // buckets written at [head, head+NB)
-    const long long* __restrict__ dstate)
+    const long long* __restrict__ dstate, int BST)
 {
     const long head = ring_head(dstate, head_in);
     const int head_mod = (int)(head % G);  // hoisted: no 64-bit mod per task
@@ -372,7 +374,7 @@ __global__ void ingest_dense_kernel(
             const int c = chan_map[cin];
             int bi = head_mod + b;
             if (bi >= G) bi -= G;      // b < NB <= G: one subtract suffices
-            const long idx = ((long)s * C + c) * G + bi;
+            const long idx = (((long)s * C + c) * G + bi) * BST;
             bsum[idx] = sum;
             bcnt[idx] = cnt;
         }
@@ -388,7 +390,7 @@ __global__ void ingest_dense_pair_kernel(
     const unsigned short* __restrict__ raw, float* __restrict__ bsum,
     float* __restrict__ bcnt, const int* __restrict__ chan_map,
     int S, int CIN, int C, int T, int G, int bucket_len, long head_in,
-    const long long* __restrict__ dstate)
+    const long long* __restrict__ dstate, int BST)
 {
     const long head = ring_head(dstate, head_in);
     const int head_mod = (int)(head % G);
@@ -481,14 +483,14 @@ __global__ void ingest_dense_pair_kernel(
             if (a0) {
                 int bi = head_mod + b0;
                 if (bi >= G) bi -= G;
-                bsum[rowo + bi] = sum0;
-                bcnt[rowo + bi] = cnt0;
+                bsum[(rowo + bi) * BST] = sum0;
+                bcnt[(rowo + bi) * BST] = cnt0;
             }
             if (a1) {
                 int bi = head_mod + b1;
                 if (bi >= G) bi -= G;
-                bsum[rowo + bi] = sum1;
-                bcnt[rowo + bi] = cnt1;
+                bsum[(rowo + bi) * BST] = sum1;
+                bcnt[(rowo + bi) * BST] = cnt1;
             }
         }
     }
@@ -501,7 +503,7 @@ __global__ void ingest_events_kernel(
     const int* __restrict__ ev_stream, const int* __restrict__ ev_chan,
     const long* __restrict__ ev_bucket, const float* __restrict__ ev_val,
     float* __restrict__ bsum, float* __restrict__ bcnt,
-    int C, int G, long n_events, long min_bucket)
+    int C, int G, long n_events, long min_bucket, int BST)
 {
     for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n_events;
          i += (long)gridDim.x * blockDim.x) {
@@ -509,7 +511,8 @@ __global__ void ingest_events_kernel(
         if (b < min_bucket) continue;  // behind the watermark: dropped (late data)
         const float v = ev_val[i];
         if (isnan(v)) continue;
-        const long idx = ((long)ev_stream[i] * C + ev_chan[i]) * G + b % G;
+        const long idx = (((long)ev_stream[i] * C + ev_chan[i]) * G
+                          + b % G) * BST;
         atomicAdd(&bsum[idx], v);
         atomicAdd(&bcnt[idx], 1.f);
     }
@@ -518,14 +521,14 @@ __global__ void ingest_events_kernel(
 // Zero the bucket slots about to be (re)used: [head, head+nb) mod G.
 __global__ void clear_buckets_kernel(
     float* __restrict__ bsum, float* __restrict__ bcnt,
-    int S, int C, int G, long head, int nb)
+    int S, int C, int G, long head, int nb, int BST)
 {
     const long n = (long)S * C * nb;
     for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
          i += (long)gridDim.x * blockDim.x) {
         const int j = (int)(i % nb);
         const long sc = i / nb;
-        const long idx = sc * G + (head + j) % G;
+        const long idx = (sc * G + (head + j) % G) * BST;
         bsum[idx] = 0.f;
         bcnt[idx] = 0.f;
     }
@@ -546,7 +549,7 @@ __global__ __launch_bounds__(256) void window_fill_kernel(
     const float* __restrict__ bsum, const float* __restrict__ bcnt,
     float* __restrict__ proc, float* __restrict__ last_val,
     int S, int C, int G, long phead_in, int np, int win_buckets,
-    const long long* __restrict__ dstate)
+    const long long* __restrict__ dstate, int BST)
 {
     const long phead = ring_nproc(dstate, phead_in);
     constexpr int CHUNK = 64;             // output points per iteration
@@ -564,8 +567,8 @@ __global__ __launch_bounds__(256) void window_fill_kernel(
     const long nsc = (long)S * C;
     for (long sc = (long)blockIdx.x * 4 + wave; sc < nsc;
          sc += (long)gridDim.x * 4) {
-        const float* bs = bsum + sc * G;
-        const float* bc = bcnt + sc * G;
+        const float* bs = bsum + sc * G * BST;
+        const float* bc = bcnt + sc * G * BST;
         float* pr = proc + sc * G;
         float carry = last_val[sc];
         // After ffill-with-carry, NaNs can only be a PREFIX of the batch
@@ -579,7 +582,7 @@ __global__ __launch_bounds__(256) void window_fill_kernel(
             const int nload = jn + win_buckets - 1;
             // coalesced bucket loads into LDS
             for (int i = lane; i < nload; i += WAVE) {
-                const long idx = (phead + j0 + i) % G;
+                const long idx = ((phead + j0 + i) % G) * BST;
                 ls[i] = bs[idx];
                 lc[i] = bc[idx];
             }
@@ -683,7 +686,7 @@ __global__ __launch_bounds__(256) void window_fill16_kernel(
     const float* __restrict__ bsum, const float* __restrict__ bcnt,
     float* __restrict__ proc, float* __restrict__ last_val,
     int S, int C, int G, long phead_in, int np, int win_buckets,
-    const long long* __restrict__ dstate)
+    const long long* __restrict__ dstate, int BST)
 {
     const long phead = ring_nproc(dstate, phead_in);
     __shared__ float lds_v[4][64];        // per wave: 4 groups x 16 values
@@ -701,7 +704,32 @@ __global__ __launch_bounds__(256) void window_fill16_kernel(
         int nan_prefix = 0;
         if (live) {
             carry = last_val[sc];
-            if (gl < np) {
+            if (gl < np && BST == 2) {
+                // packed (sum,cnt) pairs: ONE f32x2 load and ONE packed add
+                // per bucket — half the loads and adds of the split-array
+                // layout (the whole point of TSKD_PACKED_BUCKETS)
+                const f32x2p_* pk = (const f32x2p_*)(bsum + sc * G * 2);
+                const int start = (int)((phead + gl) % G);
+                const int first = min(win_buckets, G - start);
+                const int rem = win_buckets - first;
+                f32x2p_ a0 = {0.f, 0.f}, a1 = {0.f, 0.f};
+                f32x2p_ a2 = {0.f, 0.f}, a3 = {0.f, 0.f};
+                const f32x2p_* p1 = pk + start;
+                int k = 0;
+                for (; k + 4 <= first; k += 4) {
+                    a0 += p1[k];     a1 += p1[k + 1];
+                    a2 += p1[k + 2]; a3 += p1[k + 3];
+                }
+                for (; k < first; ++k) a0 += p1[k];
+                int j = 0;
+                for (; j + 4 <= rem; j += 4) {
+                    a0 += pk[j];     a1 += pk[j + 1];
+                    a2 += pk[j + 2]; a3 += pk[j + 3];
+                }
+                for (; j < rem; ++j) a0 += pk[j];
+                const f32x2p_ t = (a0 + a1) + (a2 + a3);
+                val = (t.y > 0.f) ? t.x / t.y : NAN;
+            } else if (gl < np) {
                 const float* bs = bsum + sc * G;
                 const float* bc = bcnt + sc * G;
                 // split the (possibly wrapping) 36-bucket window into two
@@ -881,7 +909,8 @@ int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
                               float* bsum, float* bcnt, const int* chan_map,
                               int S, int CIN, int C, int T, int G,
                               int bucket_len, long head,
-                              const long long* dstate, void* stream) {
+                              const long long* dstate, int bst,
+                              void* stream) {
     hipStream_t st = (hipStream_t)stream;
     const int NB = T / bucket_len;
     if (NB <= 0 || S <= 0) return 0;
@@ -907,49 +936,49 @@ int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
         if (mode == 3) {
             hipLaunchKernelGGL(ingest_dense_pair_kernel, dim3(grid), dim3(256),
                                0, st, rp, bsum, bcnt, chan_map, S, CIN, C, T,
-                               G, bucket_len, head, dstate);
+                               G, bucket_len, head, dstate, bst);
             return (int)hipGetLastError();
         }
         if (mode == 0)
             hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, 0>),
                                dim3(grid), dim3(256), 0, st, rp, bsum, bcnt,
                                chan_map, S, CIN, C, T, G, bucket_len, head,
-                               dstate);
+                               dstate, bst);
         else if (mode == 1)
             hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, 1>),
                                dim3(grid), dim3(256), 0, st, rp, bsum, bcnt,
                                chan_map, S, CIN, C, T, G, bucket_len, head,
-                               dstate);
+                               dstate, bst);
         else if (mode == 4)
             hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, 4>),
                                dim3(grid), dim3(256), 0, st, rp, bsum, bcnt,
                                chan_map, S, CIN, C, T, G, bucket_len, head,
-                               dstate);
+                               dstate, bst);
         else if (mode == 5)
             hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, 5>),
                                dim3(grid), dim3(256), 0, st, rp, bsum, bcnt,
                                chan_map, S, CIN, C, T, G, bucket_len, head,
-                               dstate);
+                               dstate, bst);
         else
             hipLaunchKernelGGL((ingest_dense_kernel<unsigned short, 2>),
                                dim3(grid), dim3(256), 0, st, rp, bsum, bcnt,
                                chan_map, S, CIN, C, T, G, bucket_len, head,
-                               dstate);
+                               dstate, bst);
     } else {
         const float* rp = (const float*)raw;
         if (mode >= 3) mode = 1;  // fp32: modes 3-5 are bf16-only
         if (mode == 0)
             hipLaunchKernelGGL((ingest_dense_kernel<float, 0>), dim3(grid),
                                dim3(256), 0, st, rp, bsum, bcnt, chan_map, S,
-                               CIN, C, T, G, bucket_len, head, dstate);
+                               CIN, C, T, G, bucket_len, head, dstate, bst);
         else if (mode == 1)
             hipLaunchKernelGGL((ingest_dense_kernel<float, 1>), dim3(grid),
                                dim3(256), 0, st, rp, bsum, bcnt, chan_map, S,
-                               CIN, C, T, G, bucket_len, head, dstate);
+                               CIN, C, T, G, bucket_len, head, dstate, bst);
         else
             hipLaunchKernelGGL((ingest_dense_kernel<float, 2>), dim3(grid),
                                dim3(256), 0, st, rp, bsum, bcnt, chan_map, S,
-                               CIN, C, T, G, bucket_len, head, dstate);
+                               CIN, C, T, G, bucket_len, head, dstate, bst);
     }
     return (int)hipGetLastError();
 }
@@ -957,28 +986,30 @@ int tskd_preproc_ingest_dense(const void* raw, int raw_is_bf16,
 int tskd_preproc_ingest_events(const int* ev_stream, const int* ev_chan,
                                const long* ev_bucket, const float* ev_val,
                                float* bsum, float* bcnt, int C, int G,
-                               long n_events, long min_bucket, void* stream) {
+                               long n_events, long min_bucket, int bst,
+                               void* stream) {
     if (n_events <= 0) return 0;
     hipLaunchKernelGGL(ingest_events_kernel, dim3(grid_for(n_events, 256)),
                        dim3(256), 0, (hipStream_t)stream, ev_stream, ev_chan,
                        ev_bucket, ev_val, bsum, bcnt, C, G, n_events,
-                       min_bucket);
+                       min_bucket, bst);
     return (int)hipGetLastError();
 }
 
 int tskd_preproc_clear_buckets(float* bsum, float* bcnt, int S, int C, int G,
-                               long head, int nb, void* stream) {
+                               long head, int nb, int bst, void* stream) {
     if (nb <= 0) return 0;
     const long n = (long)S * C * nb;
     hipLaunchKernelGGL(clear_buckets_kernel, dim3(grid_for(n, 256)), dim3(256),
-                       0, (hipStream_t)stream, bsum, bcnt, S, C, G, head, nb);
+                       0, (hipStream_t)stream, bsum, bcnt, S, C, G, head, nb,
+                       bst);
     return (int)hipGetLastError();
 }
 
 int tskd_preproc_window_fill(const float* bsum, const float* bcnt, float* proc,
                              float* last_val, int S, int C, int G, long phead,
                              int np, int win_buckets,
-                             const long long* dstate, void* stream) {
+                             const long long* dstate, int bst, void* stream) {
     if (np <= 0) return 0;
     const char* f16 = getenv("TSKD_FILL16");
     if (np <= 16 && !(f16 && f16[0] == '0')) {
@@ -987,13 +1018,13 @@ int tskd_preproc_window_fill(const float* bsum, const float* bcnt, float* proc,
         hipLaunchKernelGGL(window_fill16_kernel, dim3(grid_for(nsc4, 1)),
                            dim3(256), 0, (hipStream_t)stream, bsum, bcnt,
                            proc, last_val, S, C, G, phead, np, win_buckets,
-                           dstate);
+                           dstate, bst);
         return (int)hipGetLastError();
     }
     const long nsc = (long)S * C * WAVE;  // one wave per (stream, channel)
     hipLaunchKernelGGL(window_fill_kernel, dim3(grid_for(nsc, 256)), dim3(256),
                        0, (hipStream_t)stream, bsum, bcnt, proc, last_val, S,
-                       C, G, phead, np, win_buckets, dstate);
+                       C, G, phead, np, win_buckets, dstate, bst);
     return (int)hipGetLastError();
 }
 
