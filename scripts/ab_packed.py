@@ -17,7 +17,14 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch  # noqa: E402
 
 
-def main(S=16384, rounds=7, steps=8) -> None:
+def main(S=None, rounds=None, steps=None) -> None:
+    import argparse
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--streams", type=int, default=16384)
+    ap.add_argument("--rounds", type=int, default=7)
+    ap.add_argument("--steps", type=int, default=8)
+    a = ap.parse_args()
+    S, rounds, steps = a.streams, a.rounds, a.steps
     from tskd_amd.engine import StreamEngine
     from tskd_amd.models import build_model
     from tskd_amd.ops import MyCNNEngine
@@ -27,7 +34,8 @@ def main(S=16384, rounds=7, steps=8) -> None:
     raw = torch.randn(S, 8, 7500, device="cuda", dtype=torch.bfloat16)
     age = torch.full((S, 1), 65.0, device="cuda")
     engines = {}
-    for tag, env in (("packed", "1"), ("split", "0")):
+    for tag, env in (("split", "0"), ("packed", "1")):  # split FIRST
+
         os.environ["TSKD_PACKED_BUCKETS"] = env
         engines[tag] = StreamEngine(S, 10, ring_grid=2048, fs=125.0,
                                     device="cuda")
