@@ -582,9 +582,15 @@ __global__ __launch_bounds__(256) void window_fill_kernel(
             const int nload = jn + win_buckets - 1;
             // coalesced bucket loads into LDS
             for (int i = lane; i < nload; i += WAVE) {
-                const long idx = ((phead + j0 + i) % G) * BST;
-                ls[i] = bs[idx];
-                lc[i] = bc[idx];
+                const long idx = (phead + j0 + i) % G;
+                if (BST == 2) {  // packed (sum,cnt) pair: one 8 B load
+                    const f32x2p_ v = *(const f32x2p_*)(bs + idx * 2);
+                    ls[i] = v.x;
+                    lc[i] = v.y;
+                } else {
+                    ls[i] = bs[idx];
+                    lc[i] = bc[idx];
+                }
             }
             wsync_();
             // lane j: sliding raw-sample mean of window starting at j0+j
