@@ -102,3 +102,55 @@ class TestFillProperties:
             np.testing.assert_allclose(stitched[split:], whole[split:])
         else:
             np.testing.assert_allclose(stitched, whole)
+
+
+class TestWfdbDecodeProperties:
+    @given(st.integers(1, 4), st.integers(2, 60),
+           st.floats(1.0, 500.0), st.integers(-1000, 1000),
+           st.integers(0, 2**31))
+    @settings(max_examples=25, deadline=None)
+    def test_fmt16_gain_baseline(self, nsig, nsamp, gain, baseline, seed):
+        """(adc - baseline)/gain for arbitrary gains/baselines, multiplexed
+        signals, invalid sentinel -> NaN (independent numpy oracle)."""
+        import os
+        import tempfile
+        from tskd_amd.io import rdrecord
+        rng = np.random.default_rng(seed)
+        adc = rng.integers(-30000, 30000, size=(nsamp, nsig)).astype("<i2")
+        adc[0, 0] = -32768  # invalid sentinel
+        d = tempfile.mkdtemp()
+        adc.tofile(os.path.join(d, "r.dat"))
+        gain_r = round(float(gain), 3)
+        with open(os.path.join(d, "r.hea"), "w") as f:
+            f.write(f"r {nsig} 125 {nsamp}\n")
+            for i in range(nsig):
+                f.write(f"r.dat 16 {gain_r}({baseline})/u 16 0 0 0 0 S{i}\n")
+        rec = rdrecord(os.path.join(d, "r"))
+        want = (adc.astype(np.float64) - baseline) / gain_r
+        want[0, 0] = np.nan
+        np.testing.assert_allclose(rec.p_signal, want, rtol=1e-9)
+
+    @given(st.integers(2, 40), st.integers(0, 2**31))
+    @settings(max_examples=25, deadline=None)
+    def test_fmt212_pairs(self, nsamp, seed):
+        """12-bit packed pairs decode to the independently-unpacked values."""
+        import os
+        import tempfile
+        from tskd_amd.io import rdrecord
+        rng = np.random.default_rng(seed)
+        vals = rng.integers(-2047, 2047, size=nsamp)
+        raw = bytearray()
+        for k in range(0, nsamp - 1, 2):
+            a = int(vals[k]) & 0xFFF
+            b = int(vals[k + 1]) & 0xFFF
+            raw += bytes([a & 0xFF, ((a >> 8) & 0x0F) | ((b >> 8) << 4),
+                          b & 0xFF])
+        if nsamp % 2:
+            a = int(vals[-1]) & 0xFFF
+            raw += bytes([a & 0xFF, (a >> 8) & 0x0F, 0])
+        d = tempfile.mkdtemp()
+        open(os.path.join(d, "r.dat"), "wb").write(bytes(raw))
+        with open(os.path.join(d, "r.hea"), "w") as f:
+            f.write(f"r 1 250 {nsamp}\nr.dat 212 1(0)/u 12 0 0 0 0 ECG\n")
+        rec = rdrecord(os.path.join(d, "r"))
+        np.testing.assert_allclose(rec.p_signal[:, 0], vals.astype(float))
