@@ -100,7 +100,7 @@ class TestLoops:
         assert np.isfinite(vl) and len(results) == 128
         assert all(p in (0.0, 1.0) for _, p in results)
         out = capsys.readouterr().out
-        assert "Epoch: [0]" in out and "Test:" in out
+        assert "[train e0 " in out and "[eval " in out
 
     def test_fit_learns_synthetic_signal(self, tmp_path):
         x, age, y = make_synthetic_labeled_windows(512, pos_frac=0.3, seed=2)

@@ -147,10 +147,21 @@ class StreamEngine:
     def ingest_dense(self, raw: torch.Tensor,
                      chan_map: Optional[Sequence[int]] = None) -> int:
         """raw (S, CIN, T) contiguous samples at self.fs starting at the
-        current head; returns number of new buckets per channel."""
+        current head; returns number of new buckets per channel.
+
+        Only whole buckets are ingested: trailing ``T % bucket_len`` samples
+        are dropped (warned once) — callers should send bucket-aligned
+        trigger chunks (60 s at fs=125 is always aligned)."""
         assert raw.shape[0] == self.S
         cin, t = raw.shape[1], raw.shape[2]
         nb = t // self.bucket_len
+        if t % self.bucket_len and not getattr(self, "_warned_partial", False):
+            import warnings
+            warnings.warn(
+                f"ingest_dense: dropping {t % self.bucket_len} trailing "
+                f"samples (T={t} not a multiple of bucket_len="
+                f"{self.bucket_len})", stacklevel=2)
+            self._warned_partial = True
         if nb > self.G - self.win_buckets:
             raise ValueError(
                 f"batch spans {nb} buckets > ring capacity "
@@ -225,6 +236,8 @@ class StreamEngine:
         seconds (defaults to max ts) — buckets strictly before it become
         eligible for processing.
         """
+        if len(ts) == 0 and advance_to is None:
+            return  # nothing to ingest, no watermark advance
         min_bucket = self.nproc  # processed grid is immutable
         if self._gpu:
             # Bucketing/min/max on the GPU: host float64 math over millions
@@ -436,8 +449,13 @@ class StreamEngine:
         ``out``: optional pre-allocated destination (e.g. a GraphedForward's
         static input buffer) — every element is overwritten."""
         B, WIN = batch, self.model_win
+        assert (B - 1) * stride + WIN <= self.G, (
+            f"windows(batch={B}, stride={stride}) reaches back "
+            f"{(B - 1) * stride + WIN} grid points but the ring holds only "
+            f"G={self.G} — older slots are already overwritten")
         shape = (self.S, B, WIN, self.C) if timelast \
             else (self.S, B, self.C, WIN)
+        provided = out is not None
         if out is not None:
             assert out.shape == shape and out.dtype == dtype
             assert out.is_contiguous()
@@ -465,6 +483,8 @@ class StreamEngine:
                 raise RuntimeError(f"window_gather failed: hipError {rc}")
         else:
             pr = self.proc.numpy()
+            if provided:
+                out.zero_()  # pre-grid-0 windows must read zero, not stale
             for b in range(B):
                 wend = self.nproc - (B - 1 - b) * stride
                 if wend - WIN < 0:

@@ -1,10 +1,12 @@
 """Train/evaluate loops — reference bin/utils.py:183-275 semantics.
 
 train(): zero_grad -> model(input, age) -> loss -> NaN tripwire (utils.py:206)
--> backward -> step, with AverageMeter timing/loss/accuracy and periodic
-prints. evaluate(): no_grad loop collecting (y_true, y_pred) pairs.
+-> backward -> step, with per-batch timing/loss/accuracy meters and periodic
+progress lines. evaluate(): no_grad loop collecting (y_true, y_pred) pairs.
 Device-agnostic: the loaders yield (input, age, target) like the reference's
-TensorDataset (utils.py:365-384).
+TensorDataset (utils.py:365-384). The required parity is behavioral (meter
+cadence, NaN assert, returned averages + result pairs); the progress-line
+formatting is our own.
 """
 
 from __future__ import annotations
@@ -18,17 +20,24 @@ import torch
 from tskd_amd.train.metrics import AverageMeter, compute_batch_accuracy
 
 
+def _progress(tag: str, i: int, n: int, meters: dict) -> str:
+    parts = [f"[{tag} {i + 1}/{n}]"]
+    for name, m in meters.items():
+        parts.append(f"{name}={m.val:.4g} avg={m.avg:.4g}")
+    return "  ".join(parts)
+
+
 def train(model, device, data_loader, criterion, optimizer, epoch,
           print_freq: int = 10) -> Tuple[float, float]:
-    batch_time = AverageMeter()
-    data_time = AverageMeter()
-    losses = AverageMeter()
-    accuracy = AverageMeter()
+    t_step = AverageMeter()
+    t_data = AverageMeter()
+    m_loss = AverageMeter()
+    m_acc = AverageMeter()
 
     model.train()
     end = time.time()
     for i, (inp, age, target) in enumerate(data_loader):
-        data_time.update(time.time() - end)
+        t_data.update(time.time() - end)
         inp, age, target = inp.to(device), age.to(device), target.to(device)
         optimizer.zero_grad()
         output = model(inp, age)
@@ -36,25 +45,23 @@ def train(model, device, data_loader, criterion, optimizer, epoch,
         assert not np.isnan(loss.item()), "model diverged with loss = NaN"
         loss.backward()
         optimizer.step()
-        batch_time.update(time.time() - end)
+        t_step.update(time.time() - end)
         end = time.time()
-        losses.update(loss.item(), target.size(0))
-        accuracy.update(compute_batch_accuracy(output, target).item(),
-                        target.size(0))
+        m_loss.update(loss.item(), target.size(0))
+        m_acc.update(compute_batch_accuracy(output, target).item(),
+                     target.size(0))
         if i % print_freq == 0:
-            print(f"Epoch: [{epoch}][{i}/{len(data_loader)}]\t"
-                  f"Time {batch_time.val:.3f} ({batch_time.avg:.3f})\t"
-                  f"Data {data_time.val:.3f} ({data_time.avg:.3f})\t"
-                  f"Loss {losses.val:.4f} ({losses.avg:.4f})\t"
-                  f"Accuracy {accuracy.val:.3f} ({accuracy.avg:.3f})")
-    return losses.avg, accuracy.avg
+            print(_progress(f"train e{epoch}", i, len(data_loader),
+                            {"loss": m_loss, "acc": m_acc,
+                             "step_s": t_step, "data_s": t_data}))
+    return m_loss.avg, m_acc.avg
 
 
 def evaluate(model, device, data_loader, criterion,
              print_freq: int = 10) -> Tuple[float, float, List[Tuple[int, int]]]:
-    batch_time = AverageMeter()
-    losses = AverageMeter()
-    accuracy = AverageMeter()
+    t_step = AverageMeter()
+    m_loss = AverageMeter()
+    m_acc = AverageMeter()
     results: List[Tuple[int, int]] = []
 
     model.eval()
@@ -65,18 +72,17 @@ def evaluate(model, device, data_loader, criterion,
                                 target.to(device))
             output = model(inp, age)
             loss = criterion(output, target)
-            batch_time.update(time.time() - end)
+            t_step.update(time.time() - end)
             end = time.time()
-            losses.update(loss.item(), target.size(0))
-            accuracy.update(compute_batch_accuracy(output, target).item(),
-                            target.size(0))
+            m_loss.update(loss.item(), target.size(0))
+            m_acc.update(compute_batch_accuracy(output, target).item(),
+                         target.size(0))
             y_true = target.detach().cpu().numpy().flatten().tolist()
             y_pred = torch.sigmoid(output).round().detach().cpu().numpy() \
                 .flatten().tolist()
             results.extend(zip(y_true, y_pred))
             if i % print_freq == 0:
-                print(f"Test: [{i}/{len(data_loader)}]\t"
-                      f"Time {batch_time.val:.3f} ({batch_time.avg:.3f})\t"
-                      f"Loss {losses.val:.4f} ({losses.avg:.4f})\t"
-                      f"Accuracy {accuracy.val:.3f} ({accuracy.avg:.3f})")
-    return losses.avg, accuracy.avg, results
+                print(_progress("eval", i, len(data_loader),
+                                {"loss": m_loss, "acc": m_acc,
+                                 "step_s": t_step}))
+    return m_loss.avg, m_acc.avg, results
