@@ -57,7 +57,8 @@ def run_steps(step_fn, steps, warmup, dist, device):
         dist.barrier()
     torch.cuda.synchronize()
     t1 = time.perf_counter()
-    elapsed = torch.tensor([t1 - t0], device=device, dtype=torch.float64)
+    red_dev = device if (dist and dist.get_backend() == "nccl") else "cpu"
+    elapsed = torch.tensor([t1 - t0], device=red_dev, dtype=torch.float64)
     if dist:
         dist.all_reduce(elapsed, op=dist.ReduceOp.MAX)
     return float(elapsed.item()), lat
@@ -94,12 +95,16 @@ def main() -> None:
     dist = None
     if world > 1:
         import torch.distributed as dist_mod
+        from tskd_amd.parallel.dist import pick_backend
         dist = dist_mod
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        dist.init_process_group("nccl")
+        dist.init_process_group(pick_backend(world))
     assert torch.cuda.is_available(), "bench.py requires an MI355X"
-    torch.cuda.set_device(local_rank)
-    device = torch.device("cuda", local_rank)
+    # one rank per GPU in production; mod by visible count so world=2 on a
+    # 1-GPU box (RCCL-path validation) shares cuda:0
+    dev_idx = local_rank % torch.cuda.device_count()
+    torch.cuda.set_device(dev_idx)
+    device = torch.device("cuda", dev_idx)
 
     from tskd_amd.models import build_model
     from tskd_amd.ops import MyCNNEngine
@@ -206,8 +211,13 @@ def main() -> None:
                 w = se.windows(batch=1, stride=12, dtype=dtype)
                 probs = eng.forward(w, age, apply_sigmoid=True)
             if dist:
-                # predictions to every rank (RCCL all-gather over xGMI)
-                dist.all_gather(gathered, probs.reshape(S).contiguous())
+                if dist.get_backend() == "nccl":
+                    # predictions to every rank (RCCL all-gather over xGMI)
+                    dist.all_gather(gathered, probs.reshape(S).contiguous())
+                else:
+                    # gloo validation topology (ranks sharing one GPU)
+                    from tskd_amd.parallel.dist import all_gather_predictions
+                    all_gather_predictions(probs.reshape(S))
 
         windows_per_step = S
         cfg = {"model": args.variant, "global_batch": S * world,

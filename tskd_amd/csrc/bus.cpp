@@ -523,6 +523,60 @@ public:
         return py::make_tuple(keys, topics, ca, va, ta);
     }
 
+    // The serving ingest edge, fully native: poll + wire parse + key->stream
+    // id mapping + DP shard filter (FNV-1a % world == rank, the same hash
+    // partition_for and tskd_amd.parallel.shard_for_key use) in one pass.
+    // Returns (sid int32[], chan int32[], value float32[], ts float64[],
+    // new_keys [(key, sid), ...]) — the arrays feed torch.from_numpy with no
+    // per-message Python. Replaces the Python loop that was serve.py's
+    // per-message cost (VERDICT r1 "optimize the host ingest edge").
+    py::tuple poll_samples_sid(int max_msgs = 4096, int timeout_ms = 0,
+                               int rank = 0, int world = 1,
+                               int max_streams = 0) {
+        auto msgs = poll(max_msgs, timeout_ms);
+        std::vector<int> sids, chans;
+        std::vector<float> vals;
+        std::vector<double> tss;
+        std::vector<std::pair<std::string, int>> new_keys;
+        sids.reserve(msgs.size());
+        for (auto& m : msgs) {
+            long chan;
+            double val;
+            if (!parse_sample(m.val_.data(), m.val_.size(), &chan, &val))
+                continue;
+            if (world > 1 &&
+                (int)(fnv1a(m.key_) % (uint64_t)world) != rank)
+                continue;  // another rank's patient
+            auto it = sid_.find(m.key_);
+            int sid;
+            if (it == sid_.end()) {
+                if (max_streams > 0 && (int)sid_.size() >= max_streams)
+                    throw std::runtime_error("max_streams exceeded: " +
+                                             m.key_);
+                sid = (int)sid_.size();
+                sid_.emplace(m.key_, sid);
+                new_keys.emplace_back(m.key_, sid);
+            } else {
+                sid = it->second;
+            }
+            sids.push_back(sid);
+            chans.push_back((int)chan);
+            vals.push_back((float)val);
+            tss.push_back((double)m.ts_us / 1e6);
+        }
+        auto sa = py::array_t<int>((py::ssize_t)sids.size());
+        auto ca = py::array_t<int>((py::ssize_t)chans.size());
+        auto va = py::array_t<float>((py::ssize_t)vals.size());
+        auto ta = py::array_t<double>((py::ssize_t)tss.size());
+        std::memcpy(sa.mutable_data(), sids.data(), sids.size() * 4);
+        std::memcpy(ca.mutable_data(), chans.data(), chans.size() * 4);
+        std::memcpy(va.mutable_data(), vals.data(), vals.size() * 4);
+        std::memcpy(ta.mutable_data(), tss.data(), tss.size() * 8);
+        return py::make_tuple(sa, ca, va, ta, new_keys);
+    }
+
+    int n_streams() const { return (int)sid_.size(); }
+
 private:
     static int64_t now_us() {
         struct timespec ts{};

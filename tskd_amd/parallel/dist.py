@@ -23,6 +23,18 @@ import torch
 import torch.distributed as dist
 
 
+def pick_backend(world: int) -> str:
+    """RCCL (backend "nccl" on ROCm) when every rank can own a GPU —
+    the production one-process-per-GPU topology. When ranks outnumber
+    visible devices (world>1 validation on a 1-GPU box), RCCL refuses with
+    "Duplicate GPU detected", so fall back to gloo: compute stays on the
+    GPU, collectives stage through host memory (tiny payloads — see
+    all_gather_predictions)."""
+    if not torch.cuda.is_available():
+        return "gloo"
+    return "nccl" if torch.cuda.device_count() >= world else "gloo"
+
+
 def init_distributed(backend: Optional[str] = None) -> Tuple[int, int]:
     """Initialize from torchrun env; returns (rank, world). No-op world=1."""
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -31,7 +43,7 @@ def init_distributed(backend: Optional[str] = None) -> Tuple[int, int]:
     if not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         if backend is None:
-            backend = "nccl" if torch.cuda.is_available() else "gloo"
+            backend = pick_backend(world)
         dist.init_process_group(backend)
     return dist.get_rank(), dist.get_world_size()
 
@@ -63,6 +75,14 @@ def all_gather_predictions(probs: torch.Tensor,
     if not dist.is_initialized() or dist.get_world_size(group) == 1:
         return probs.unsqueeze(0)
     world = dist.get_world_size(group)
+    backend = dist.get_backend(group)
+    if backend == "gloo" and probs.is_cuda:
+        # gloo has no CUDA all_gather: stage through host (validation-only
+        # topology — ranks sharing one GPU; production uses RCCL direct)
+        host = probs.contiguous().cpu()
+        out = [torch.empty_like(host) for _ in range(world)]
+        dist.all_gather(out, host, group=group)
+        return torch.stack(out).to(probs.device)
     out = [torch.empty_like(probs) for _ in range(world)]
     dist.all_gather(out, probs.contiguous(), group=group)
     return torch.stack(out)
@@ -71,6 +91,8 @@ def all_gather_predictions(probs: torch.Tensor,
 def all_reduce_max(value: float, device=None) -> float:
     if not dist.is_initialized():
         return value
+    if dist.get_backend() == "gloo":
+        device = "cpu"  # gloo reduces host tensors
     t = torch.tensor([value], dtype=torch.float64,
                      device=device or "cpu")
     dist.all_reduce(t, op=dist.ReduceOp.MAX)
