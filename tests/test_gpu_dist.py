@@ -45,10 +45,19 @@ def _run_torchrun(nproc: int, script: str, extra_env=None, timeout=240):
 def test_rccl_world2_allgather_numerics():
     r = _run_torchrun(2, os.path.join(REPO, "scripts", "dist_probe.py"))
     assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
-    lines = [json.loads(ln) for ln in r.stdout.splitlines()
-             if ln.startswith('{"probe"')]
-    assert len(lines) == 2, r.stdout
-    for rec in lines:
+    # both ranks print to the shared stdout; records may share a line —
+    # scan with raw_decode instead of assuming one object per line
+    dec = json.JSONDecoder()
+    recs, pos = [], 0
+    while True:
+        pos = r.stdout.find('{"probe"', pos)
+        if pos < 0:
+            break
+        obj, end = dec.raw_decode(r.stdout[pos:])
+        recs.append(obj)
+        pos += end
+    assert len(recs) == 2, r.stdout
+    for rec in recs:
         assert rec["numerics_ok"], rec
         assert rec["world"] == 2
 
