@@ -1,0 +1,146 @@
+"""Round-2 serving-edge features: native poll_samples_sid (C++ key->sid +
+shard filter), wall-clock event-time fast-forward, force_ready, and the
+serving-latency harness smoke (CPU).
+"""
+import json
+import subprocess
+import sys
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from tskd_amd.bus import Bus, Consumer, Producer
+from tskd_amd.engine import StreamEngine
+from tskd_amd.parallel.dist import shard_for_key
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.fixture
+def bus(tmp_path):
+    return Bus(str(tmp_path / "bus"))
+
+
+class TestPollSamplesSid:
+    def test_dense_sid_assignment_and_arrays(self, bus):
+        bus.create_topic("HR")
+        p = Producer(bus)
+        c = Consumer(bus, starting="earliest")
+        c.subscribe(["HR"])
+        pids = [f"p{i:06d}" for i in range(5)]
+        for rep in range(3):
+            for pid in pids:
+                p.produce("HR", pid, json.dumps([2, 70.0 + rep]),
+                          ts_us=1_000_000 * (rep + 1))
+        sa, ca, va, ta, nk = c.poll_samples_sid(max_msgs=100)
+        assert list(dict(nk).keys()) == pids      # first-seen order
+        assert list(dict(nk).values()) == [0, 1, 2, 3, 4]
+        assert sa.dtype == np.int32 and len(sa) == 15
+        assert (sa[:5] == np.arange(5)).all()
+        assert (ca == 2).all() and va.dtype == np.float32
+        assert ta.dtype == np.float64 and ta[0] == 1.0
+        # second poll: no new keys, ids stable
+        p.produce("HR", pids[3], json.dumps([2, 99.0]), ts_us=4_000_000)
+        sa2, _, _, _, nk2 = c.poll_samples_sid(max_msgs=100)
+        assert nk2 == [] and list(sa2) == [3]
+
+    def test_shard_filter_matches_python_hash(self, bus):
+        bus.create_topic("HR")
+        p = Producer(bus)
+        pids = [f"p{i:06d}" for i in range(64)]
+        for pid in pids:
+            p.produce("HR", pid, json.dumps([0, 1.0]), ts_us=1)
+        world = 4
+        seen = set()
+        for rank in range(world):
+            c = Consumer(bus, starting="earliest")
+            c.subscribe(["HR"])
+            sa, ca, va, ta, nk = c.poll_samples_sid(max_msgs=1000,
+                                                    rank=rank, world=world)
+            keys = [k for k, _ in nk]
+            assert all(shard_for_key(k, world) == rank for k in keys)
+            # dense per-rank ids
+            assert [s for _, s in nk] == list(range(len(keys)))
+            seen.update(keys)
+        assert seen == set(pids)  # partition of the key space
+
+    def test_max_streams_enforced(self, bus):
+        bus.create_topic("HR")
+        p = Producer(bus)
+        for i in range(5):
+            p.produce("HR", f"p{i}", json.dumps([0, 1.0]), ts_us=1)
+        c = Consumer(bus, starting="earliest")
+        c.subscribe(["HR"])
+        with pytest.raises(RuntimeError, match="max_streams"):
+            c.poll_samples_sid(max_msgs=100, max_streams=3)
+
+
+class TestWallClockEventTime:
+    def test_fresh_engine_fast_forwards(self):
+        se = StreamEngine(4, 3, ring_grid=256, device="cpu")
+        now = 1_787_000_000.0  # epoch seconds
+        n = 16
+        se.ingest_events(torch.zeros(n, dtype=torch.long),
+                         torch.zeros(n, dtype=torch.long),
+                         torch.tensor([now + 5 * i for i in range(n)],
+                                      dtype=torch.float64),
+                         torch.arange(n, dtype=torch.float32) + 1,
+                         advance_to=now + 5 * n)
+        # origin skipped ahead: nproc near the wall-clock bucket, bounded
+        # catch-up (no multi-hundred-million-point refill)
+        assert se._origin_nproc > 0
+        # nproc may lag the first-event bucket by up to win-1 (sliding
+        # window not yet complete) but never leads it by more than the ring
+        assert -se.win_buckets <= se.nproc - se._origin_nproc <= se.G
+        assert not se.ready  # < 780 s of data since the stream began
+        se.force_ready()
+        assert se.ready
+        w = se.windows(batch=1, stride=12)
+        assert w.shape == (4, 1, 3, 120)
+        assert torch.isfinite(w).all()
+
+    def test_ready_stays_origin_relative(self):
+        se = StreamEngine(2, 2, ring_grid=512, device="cpu")
+        now = 1_787_000_000.0
+        # feed 800 s of data in 40 s chunks -> ready flips when >= 780 s
+        # (600 s model window + 180 s sliding window) has accumulated
+        became_ready_at = None
+        for step in range(21):
+            t0 = now + 40.0 * step
+            ts = torch.tensor([t0 + 5 * i for i in range(8)],
+                              dtype=torch.float64)
+            se.ingest_events(torch.zeros(8, dtype=torch.long),
+                             torch.zeros(8, dtype=torch.long),
+                             ts, torch.ones(8), advance_to=t0 + 40.0)
+            if se.ready and became_ready_at is None:
+                became_ready_at = 40.0 * (step + 1)
+        assert became_ready_at is not None
+        assert 760.0 <= became_ready_at <= 840.0
+
+    def test_zero_based_times_unaffected(self):
+        se = StreamEngine(2, 2, ring_grid=256, device="cpu")
+        raw = torch.randn(2, 2, 625 * 5)  # 5 buckets at fs=125, 5-s buckets
+        for _ in range(40):
+            se.ingest_dense(raw)
+        assert se._origin_nproc == 0
+        assert se.ready  # 200 buckets > 120 + 36
+
+
+class TestServingLatencyHarness:
+    def test_quick_cpu_run(self):
+        r = subprocess.run(
+            [sys.executable, os.path.join(REPO, "scripts",
+                                          "serving_latency.py"),
+             "--device", "cpu", "--quick"],
+            capture_output=True, text=True, timeout=300, cwd=REPO)
+        assert r.returncode == 0, r.stderr
+        recs = [json.loads(ln) for ln in r.stdout.splitlines()
+                if ln.startswith("{")]
+        assert {x["scenario"] for x in recs} == {"drain", "live"}
+        drain = next(x for x in recs if x["scenario"] == "drain")
+        assert drain["events_drained"] == drain["n_events"]
+        live = next(x for x in recs if x["scenario"] == "live")
+        assert live["n_predictions"] > 0
+        assert live["wire_to_store_ms"]["p50"] is not None

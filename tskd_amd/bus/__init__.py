@@ -118,6 +118,17 @@ class Consumer:
         offsets still advance)."""
         return self._c.poll_samples(max_msgs, timeout_ms)
 
+    def poll_samples_sid(self, max_msgs: int = 4096, timeout_ms: int = 0,
+                         rank: int = 0, world: int = 1,
+                         max_streams: int = 0):
+        """The fully-native serving ingest edge: poll + wire parse + key->
+        dense-stream-id mapping + DP shard filter (FNV-1a % world == rank,
+        same hash as parallel.shard_for_key) in one C++ pass. Returns
+        (sid int32[], chan int32[], value float32[], ts float64[] seconds,
+        new_keys [(key, sid), ...] in sid order)."""
+        return self._c.poll_samples_sid(max_msgs, timeout_ms, rank, world,
+                                        max_streams)
+
     def seek(self, topic: str, partition: int, offset: int) -> None:
         self._c.seek(topic, partition, offset)
 

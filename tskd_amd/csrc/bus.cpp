@@ -586,6 +586,7 @@ private:
     std::shared_ptr<Bus> bus_;
     std::string starting_;
     std::map<std::string, uint64_t> pos_;
+    std::map<std::string, int> sid_;  // key -> dense stream id (serving)
     std::vector<std::pair<std::string, int>> parts_;
 };
 
@@ -630,5 +631,10 @@ PYBIND11_MODULE(_tskd_bus, m) {
              py::arg("timeout_ms") = 0)
         .def("poll_samples", &Consumer::poll_samples,
              py::arg("max_msgs") = 4096, py::arg("timeout_ms") = 0)
-        .def("positions", &Consumer::positions);
+        .def("positions", &Consumer::positions)
+        .def("poll_samples_sid", &Consumer::poll_samples_sid,
+             py::arg("max_msgs") = 4096, py::arg("timeout_ms") = 0,
+             py::arg("rank") = 0, py::arg("world") = 1,
+             py::arg("max_streams") = 0)
+        .def("n_streams", &Consumer::n_streams);
 }
