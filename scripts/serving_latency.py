@@ -53,7 +53,8 @@ def make_server(tmp: str, device: str, max_streams: int,
     return bus, store, srv, cfg
 
 
-def fill_backlog(bus, cfg, n_patients: int, n_events: int) -> float:
+def fill_backlog(bus, cfg, n_patients: int, n_events: int,
+                 base_us: int = 1_000_000) -> float:
     """Pre-produce n_events across patients/channels with monotonically
     increasing event time; returns produce throughput (events/s)."""
     prod = Producer(bus)
@@ -62,7 +63,7 @@ def fill_backlog(bus, cfg, n_patients: int, n_events: int) -> float:
         bus.create_topic(t)
     pids = [f"p{i:06d}" for i in range(n_patients)]
     t0 = time.perf_counter()
-    base = 1_000_000  # event-time us epoch (behind nothing; fresh engine)
+    base = base_us
     k = 0
     for e in range(n_events):
         pid = pids[e % n_patients]
@@ -111,7 +112,12 @@ def scenario_drain(device: str, n_patients: int, n_events: int) -> dict:
             drained += len(sa)
             t_poll += p1 - p0
             t_ingest += p2 - p1
-        # one full trigger (model + persist) on the hot rings
+        # one full trigger (model + persist) on the hot rings — feed one
+        # more grid point of fresh data (ahead of the consumed backlog)
+        # so the trigger actually scores every stream
+        fill_backlog(bus, cfg, n_patients, n_patients * 8,
+                     base_us=int(srv.hwm * 1e6) + 5_000_000)
+        srv.watermark_s = 0.0  # drain everything just produced
         srv.se.force_ready()
         m0 = time.perf_counter()
         n_pred = srv.trigger()
