@@ -95,16 +95,31 @@ def main() -> None:
     ap.add_argument("--reps", type=int, default=60)
     args = ap.parse_args()
     assert torch.cuda.is_available()
-    configs = [
-        ("baseline serial", False, None),
-        ("overlap full-grid (r1 neutral)", True, None),
-        ("overlap grid=8192", True, 8192),
-        ("overlap grid=4096", True, 4096),
-        ("overlap grid=2048", True, 2048),
-        ("overlap grid=1024", True, 1024),
-        ("serial grid=2048 (BW check)", False, 2048),
-        ("baseline serial (repeat)", False, None),
-    ]
+    if os.environ.get("TSKD_AB_FINE"):
+        # round-2 fine sweep after the coarse result (2048 best at +6.5%):
+        # bracket the ingest-BW/model-slot tradeoff
+        configs = [
+            ("baseline serial", False, None),
+            ("overlap grid=1792", True, 1792),
+            ("overlap grid=2048", True, 2048),
+            ("overlap grid=2304", True, 2304),
+            ("overlap grid=2560", True, 2560),
+            ("overlap grid=3072", True, 3072),
+            ("overlap grid=3584", True, 3584),
+            ("overlap grid=2048 (repeat)", True, 2048),
+            ("baseline serial (repeat)", False, None),
+        ]
+    else:
+        configs = [
+            ("baseline serial", False, None),
+            ("overlap full-grid (r1 neutral)", True, None),
+            ("overlap grid=8192", True, 8192),
+            ("overlap grid=4096", True, 4096),
+            ("overlap grid=2048", True, 2048),
+            ("overlap grid=1024", True, 1024),
+            ("serial grid=2048 (BW check)", False, 2048),
+            ("baseline serial (repeat)", False, None),
+        ]
     out = []
     for name, ov, grid in configs:
         rec = run_config(args.streams, args.reps, ov, grid)
