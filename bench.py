@@ -39,10 +39,18 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 import torch  # noqa: E402
 
 
-def run_steps(step_fn, steps, warmup, dist, device):
+def run_steps(step_fn, steps, warmup, dist, device, min_warm_s=6.0):
+    # W contract warmup steps, then keep warming until min_warm_s wall time
+    # has elapsed: stabilizes clocks AND makes the run long enough for the
+    # driver's rocm-smi busy sampler to see it (r1 timed region was 40 ms —
+    # invisible at 1 Hz sampling)
+    w0 = time.perf_counter()
     for _ in range(warmup):
         step_fn()
-    torch.cuda.synchronize()
+    torch.cuda.synchronize()  # drain before wall-clock-paced extension
+    while time.perf_counter() - w0 < min_warm_s:
+        step_fn()
+        torch.cuda.synchronize()
     if dist:
         dist.barrier()
     torch.cuda.synchronize()

@@ -248,3 +248,30 @@ class TestTrainingQuality:
         probs = score_model(trained.cpu(), xv, av)
         rep = classification_metrics(yv, probs)
         assert rep["roc_auc"] > 0.8, rep
+
+
+@pytest.mark.gpu
+class TestBatchAccuracyKernel:
+    """K14 fused metric kernel vs the torch oracle (utils.py:122-134)."""
+
+    def test_matches_torch_oracle(self):
+        from tskd_amd.train.metrics import compute_batch_accuracy
+        torch.manual_seed(0)
+        for n in (1, 63, 64, 257, 100_000):
+            logits = torch.randn(n, device="cuda") * 3
+            target = (torch.rand(n, device="cuda") < 0.4).float()
+            got = compute_batch_accuracy(logits, target)
+            pred = torch.sigmoid(logits).round()
+            want = pred.eq(target).sum().float() * 100.0 / n
+            assert abs(float(got) - float(want)) < 1e-3, (n, got, want)
+
+    def test_zero_logit_rounds_down_and_nan_incorrect(self):
+        from tskd_amd.train.metrics import compute_batch_accuracy
+        # torch: round(sigmoid(0)) = round(0.5) = 0 (half-to-even);
+        # NaN logits never equal the target
+        logits = torch.tensor([0.0, 0.0, float("nan"), 5.0], device="cuda")
+        target = torch.tensor([0.0, 1.0, 0.0, 1.0], device="cuda")
+        got = float(compute_batch_accuracy(logits, target))
+        pred = torch.sigmoid(logits).round()
+        want = float(pred.eq(target).sum().float() * 100.0 / 4)
+        assert abs(got - want) < 1e-6 and abs(got - 50.0) < 1e-6

@@ -773,9 +773,39 @@ int conv_bwd(const float* x, const float* stash, const float* dfeat,
                            dim3(256), 0, s, x, stash, dfeat, wpack, grads, SN);
     return (int)hipGetLastError();
 }
+// K14: fused batch accuracy — sigmoid -> round -> eq -> count
+// in one pass (reference utils.py:122-134). round(sigmoid(x)) == 1 iff
+// x > 0 (torch round-half-to-even sends sigmoid(0)=0.5 to 0), so the
+// sigmoid never needs evaluating; NaN logits count incorrect (NaN.round()
+// != target in torch). Wave shuffle reduction, one atomic per wave.
+__global__ void batch_accuracy_kernel(const float* __restrict__ logits,
+                                      const float* __restrict__ targets,
+                                      float* __restrict__ out, long n) {
+    int correct = 0;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (long)gridDim.x * blockDim.x) {
+        const float v = logits[i];
+        const float pred = (v > 0.0f) ? 1.0f : 0.0f;
+        correct += (!__builtin_isnan(v) && pred == targets[i]) ? 1 : 0;
+    }
+    for (int off = WAVE / 2; off > 0; off >>= 1)
+        correct += __shfl_down(correct, off);
+    if ((threadIdx.x % WAVE) == 0 && correct)
+        atomicAdd(out, (float)correct);
+}
+
 }  // namespace
 
 extern "C" {
+
+int tskd_train_batch_accuracy(const float* logits, const float* targets,
+                              float* out, long n, void* stream) {
+    if (n <= 0) return 0;
+    int grid = (int)min((n + 255) / 256, (long)4096);
+    hipLaunchKernelGGL(batch_accuracy_kernel, dim3(grid), dim3(256), 0,
+                       (hipStream_t)stream, logits, targets, out, n);
+    return (int)hipGetLastError();
+}
 
 int tskd_train_conv_fwd(const float* x, float* feat, float* stash,
                         const float* wpack, int SN, float drop1_p,

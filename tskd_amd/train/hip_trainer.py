@@ -51,6 +51,8 @@ def _load_train_lib() -> ctypes.CDLL:
     lib.tskd_train_adam.argtypes = [P, P, P, P, L, F, F, F, F, I, P]
     lib.tskd_train_stash_sizes.restype = I
     lib.tskd_train_stash_sizes.argtypes = [I, P, P]
+    lib.tskd_train_batch_accuracy.restype = I
+    lib.tskd_train_batch_accuracy.argtypes = [P, P, P, L, P]
     _tlib = lib
     return lib
 
