@@ -165,6 +165,16 @@ class TestDashboard:
             in body
         h = client.get("/health").json()
         assert h["status"] == "ok" and h["predictions"] == 1
+        # SSE push stream: first event carries a full snapshot + predictions
+        with client.stream("GET", "/api/stream?limit=1") as r:
+            assert r.headers["content-type"].startswith("text/event-stream")
+            for line in r.iter_lines():
+                if line.startswith("data: "):
+                    snap = json.loads(line[len("data: "):])
+                    break
+        assert "p000194" in snap["patients"]
+        assert len(snap["raw"]["p000194"]["0"]) == 20
+        assert snap["preds"][0]["patient"] == "p000194"
 
 
 @pytest.mark.gpu

@@ -7,8 +7,13 @@ process consumes the raw channel topics and `call-stream` from the bus into
 bounded in-memory series (guarded by one lock, like the reference's mutex
 :58) and serves:
 
-    GET /                  — HTML dashboard (inline JS canvas sparklines,
-                             1-s raw/processed refresh, 10-s predictions)
+    GET /                  — HTML dashboard (inline JS canvas sparklines)
+    GET /api/stream        — server-sent events: PUSH of raw/processed
+                             snapshots every 1 s + predictions every 10 s
+                             (the reference's Bokeh-websocket-push analog,
+                             plotData.py:224-238); the browser subscribes
+                             with EventSource and falls back to polling
+                             the endpoints below if SSE drops
     GET /api/patients      — known patient ids
     GET /api/raw/<pid>     — per-channel raw series
     GET /api/processed/<pid> — per-channel processed series
@@ -45,25 +50,34 @@ function spark(cv,xs,color){const c=cv.getContext('2d');c.clearRect(0,0,cv.width
  c.strokeStyle=color;c.beginPath();xs.forEach((v,i)=>{const x=i/(xs.length-1||1)*cv.width,
  y=cv.height-8-(v-mn)/sp*(cv.height-16);i?c.lineTo(x,y):c.moveTo(x,y)});c.stroke();
  c.fillStyle='#888';c.fillText(mx.toFixed(1),2,10);c.fillText(mn.toFixed(1),2,cv.height-2)}
-async function tick(){const pats=await j('/api/patients');const root=document.getElementById('root');
- for(const p of pats){let d=document.getElementById('p_'+p);
-  if(!d){d=document.createElement('div');d.id='p_'+p;
-   d.innerHTML=`<h3>${p}</h3><div class="raws"></div><div class="procs"></div>
-   <div>risk: <span class="risk">-</span></div>`;root.appendChild(d)}
-  const raw=await j('/api/raw/'+p),proc=await j('/api/processed/'+p);
-  for(const[name,series]of[['raws',raw],['procs',proc]]){const host=d.getElementsByClassName(name)[0];
-   for(const ch in series){let w=document.getElementById(name+p+ch);
-    if(!w){w=document.createElement('span');w.className='ch';w.id=name+p+ch;
-     w.innerHTML=`<div>${name=='raws'?'raw':'proc'} ch${ch}</div><canvas width=160 height=60></canvas>`;
-     host.appendChild(w)}
-    spark(w.getElementsByTagName('canvas')[0],series[ch],name=='raws'?'#4af':'#fa4')}}}
+function render(p,raw,proc){const root=document.getElementById('root');
+ let d=document.getElementById('p_'+p);
+ if(!d){d=document.createElement('div');d.id='p_'+p;
+  d.innerHTML=`<h3>${p}</h3><div class="raws"></div><div class="procs"></div>
+  <div>risk: <span class="risk">-</span></div>`;root.appendChild(d)}
+ for(const[name,series]of[['raws',raw],['procs',proc]]){const host=d.getElementsByClassName(name)[0];
+  for(const ch in series){let w=document.getElementById(name+p+ch);
+   if(!w){w=document.createElement('span');w.className='ch';w.id=name+p+ch;
+    w.innerHTML=`<div>${name=='raws'?'raw':'proc'} ch${ch}</div><canvas width=160 height=60></canvas>`;
+    host.appendChild(w)}
+   spark(w.getElementsByTagName('canvas')[0],series[ch],name=='raws'?'#4af':'#fa4')}}}
+function showPreds(ps){for(const r of ps){const d=document.getElementById('p_'+r.patient);
+ if(d)d.getElementsByClassName('risk')[0].textContent=
+   r.risk.toFixed(4)+' @ '+r.t.toFixed(0)+'s'}}
+// PUSH path: server-sent events (1-s snapshots + 10-s predictions)
+let pushOk=false;
+try{const es=new EventSource('/api/stream');
+ es.onmessage=(e)=>{pushOk=true;const s=JSON.parse(e.data);
+  for(const p of s.patients)render(p,s.raw[p]||{},s.proc[p]||{});
+  if(s.preds)showPreds(s.preds)};
+ es.onerror=()=>{pushOk=false};}catch(e){pushOk=false}
+// POLL fallback: only active while the push stream is down
+async function tick(){if(!pushOk){const pats=await j('/api/patients');
+ for(const p of pats)render(p,await j('/api/raw/'+p),await j('/api/processed/'+p))}
  setTimeout(tick,1000)}
-async function preds(){const ps=await j('/api/predictions');
- for(const r of ps){const d=document.getElementById('p_'+r.patient);
-  if(d)d.getElementsByClassName('risk')[0].textContent=
-    r.risk.toFixed(4)+' @ '+r.t.toFixed(0)+'s'}
+async function preds(){if(!pushOk)showPreds(await j('/api/predictions'));
  setTimeout(preds,10000)}
-tick();preds();
+setTimeout(tick,1500);setTimeout(preds,1500);
 </script></body></html>"""
 
 
@@ -136,6 +150,41 @@ def build_app(state: DashState):
         from tskd_amd.metrics import prometheus_text
         return PlainTextResponse(prometheus_text([state.pump_timer]),
                                  media_type="text/plain; version=0.0.4")
+
+    @app.get("/api/stream")
+    async def stream(limit: int = 0):
+        # PUSH: SSE snapshots — 1-s raw/processed cadence, predictions on
+        # every 10th event (the reference's two Bokeh periodic callbacks,
+        # plotData.py:216-219, as pushes instead of browser timers).
+        # ?limit=N bounds the stream (testing/curl); 0 = endless.
+        import asyncio
+
+        from fastapi.responses import StreamingResponse
+
+        async def gen():
+            n = 0
+            while limit <= 0 or n < limit:
+                with state.lock:
+                    pats = sorted(set(state.raw) | set(state.proc))
+                    snap = {
+                        "patients": pats,
+                        "raw": {p: {c: list(v)[-160:]
+                                    for c, v in state.raw[p].items()}
+                                for p in pats},
+                        "proc": {p: {c: list(v)[-160:]
+                                     for c, v in state.proc[p].items()}
+                                 for p in pats},
+                    }
+                if n % 10 == 0:
+                    snap["preds"] = [
+                        {"patient": p, "t": t.timestamp(), "risk": r}
+                        for p, t, r in state.store.tail(100)]
+                yield f"data: {json.dumps(snap)}\n\n"
+                n += 1
+                if limit <= 0 or n < limit:
+                    await asyncio.sleep(1.0)
+
+        return StreamingResponse(gen(), media_type="text/event-stream")
 
     @app.get("/api/patients")
     def patients():
