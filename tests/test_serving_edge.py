@@ -144,3 +144,26 @@ class TestServingLatencyHarness:
         live = next(x for x in recs if x["scenario"] == "live")
         assert live["n_predictions"] > 0
         assert live["wire_to_store_ms"]["p50"] is not None
+
+
+@pytest.mark.gpu
+class TestServingLatencyGpu:
+    def test_wire_to_store_quick(self):
+        """Full bus->parse->H2D->ring->model->store path on the GPU: the
+        quick live scenario must score predictions with single-digit-to-
+        low-double-digit-ms wire->store latency (measured p50 ~8 ms at 20k
+        events/s in profiles/r02_serving_capacity.md)."""
+        r = subprocess.run(
+            [sys.executable, os.path.join(REPO, "scripts",
+                                          "serving_latency.py"),
+             "--device", "cuda", "--quick"],
+            capture_output=True, text=True, timeout=420, cwd=REPO)
+        assert r.returncode == 0, r.stderr[-2000:]
+        recs = [json.loads(ln) for ln in r.stdout.splitlines()
+                if ln.startswith("{")]
+        live = next(x for x in recs if x["scenario"] == "live")
+        assert live["n_predictions"] > 0
+        assert live["wire_to_store_ms"]["p50"] < 200.0
+        drain = next(x for x in recs if x["scenario"] == "drain")
+        assert drain["events_drained"] == drain["n_events"]
+        assert drain["n_predictions"] > 0
