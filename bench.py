@@ -40,17 +40,29 @@ import torch  # noqa: E402
 
 
 def run_steps(step_fn, steps, warmup, dist, device, min_warm_s=6.0):
-    # W contract warmup steps, then keep warming until min_warm_s wall time
-    # has elapsed: stabilizes clocks AND makes the run long enough for the
-    # driver's rocm-smi busy sampler to see it (r1 timed region was 40 ms —
-    # invisible at 1 Hz sampling)
+    # W contract warmup steps, then extend warmup to ~min_warm_s wall time:
+    # stabilizes clocks AND makes the run long enough for the driver's
+    # rocm-smi busy sampler to see it (r1 timed region was 40 ms —
+    # invisible at 1 Hz sampling). The extension count must be AGREED
+    # across ranks (steps contain collectives): take the MAX of each
+    # rank's estimate, then run exactly that many everywhere.
     w0 = time.perf_counter()
     for _ in range(warmup):
         step_fn()
-    torch.cuda.synchronize()  # drain before wall-clock-paced extension
-    while time.perf_counter() - w0 < min_warm_s:
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - w0
+    per_step = max(elapsed / max(warmup, 1), 1e-4)
+    n_extra = max(0, int((min_warm_s - elapsed) / per_step) + 1) \
+        if (warmup > 0 and elapsed < min_warm_s) else 0
+    if dist:
+        red_dev = device if dist.get_backend() == "nccl" else "cpu"
+        t = torch.tensor([n_extra], dtype=torch.int64, device=red_dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        n_extra = int(t.item())
+    n_extra = min(n_extra, 100_000)
+    for _ in range(n_extra):
         step_fn()
-        torch.cuda.synchronize()
+    torch.cuda.synchronize()
     if dist:
         dist.barrier()
     torch.cuda.synchronize()
