@@ -39,7 +39,11 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 import torch  # noqa: E402
 
 
-def run_steps(step_fn, steps, warmup, dist, device, min_warm_s=6.0):
+def run_steps(step_fn, steps, warmup, dist, device, min_warm_s=None):
+    if min_warm_s is None:
+        # TSKD_BENCH_MINWARM=0 disables the extension (rocprof runs: the
+        # extra ~3k dispatches overflow the tracer)
+        min_warm_s = float(os.environ.get("TSKD_BENCH_MINWARM", "6"))
     # W contract warmup steps, then extend warmup to ~min_warm_s wall time:
     # stabilizes clocks AND makes the run long enough for the driver's
     # rocm-smi busy sampler to see it (r1 timed region was 40 ms —
