@@ -518,7 +518,8 @@ __global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
 // ---------------------------------------------------------------------------
 // conv backward: dfeat -> conv/bias grads (atomics). One wave per window.
 // ---------------------------------------------------------------------------
-template <class G, int XP = 1>  // XP: lx row pad (bank spread A/B)
+template <class G, int XP = 1, int SLIDE = 0>  // XP: lx row pad (bank
+                                               // spread); SLIDE: r2 scheme
 __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
     const float* __restrict__ x,        // (SN, CIN, L)
     const float* __restrict__ stash,    // (SN, SC_SIZE)
@@ -635,24 +636,59 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
             twsync();
         }
         // conv1 grads: dW1[c][i][k] = sum_s da1[c][s] * x[i][s+k].
-        // 4 independent partial accumulators: keeps 8 LDS loads in flight
-        // per iteration instead of exposing LDS latency every element.
-        for (int o = lane; o < 4 * G::CIN * G::K1; o += WAVE) {
-            const int c = o / (G::CIN * G::K1);
-            const int i = (o / G::K1) % G::CIN;
-            const int k = o % G::K1;
-            const float* da = lda1[wave] + c * G::C1;
-            const float* xr = xw + i * (G::L + XP) + k;
-            float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
-            int s = 0;
-            for (; s + 4 <= G::C1; s += 4) {
-                a0 = fmaf(da[s + 0], xr[s + 0], a0);
-                a1 = fmaf(da[s + 1], xr[s + 1], a1);
-                a2 = fmaf(da[s + 2], xr[s + 2], a2);
-                a3 = fmaf(da[s + 3], xr[s + 3], a3);
+        if (SLIDE && 4 * G::CIN <= WAVE) {
+            // Sliding-register scheme (round 2): lane owns (c, i) with all
+            // K1 accumulators in registers and a K1-wide register window of
+            // x sliding over s. LDS reads drop from 2 per FMA to 2 per
+            // K1 FMAs (1 x + 1 da per s step; the da read is a broadcast
+            // within each c group) — this loop was the kernel's LDS-
+            // throughput bound (r1 PMC: LDS-wait + conflicts).
+            const int c = lane / G::CIN;       // 4 groups
+            const int i = lane % G::CIN;
+            if (lane < 4 * G::CIN) {
+                const float* da = lda1[wave] + c * G::C1;
+                const float* xr = xw + i * (G::L + XP);
+                float acc[G::K1];
+                float w[G::K1];
+                #pragma unroll
+                for (int k = 0; k < G::K1; ++k) {
+                    acc[k] = 0.f;
+                    w[k] = xr[k];
+                }
+                for (int s = 0; s < G::C1; ++s) {
+                    const float d = da[s];
+                    #pragma unroll
+                    for (int k = 0; k < G::K1; ++k)
+                        acc[k] = fmaf(d, w[k], acc[k]);
+                    #pragma unroll
+                    for (int k = 0; k < G::K1 - 1; ++k) w[k] = w[k + 1];
+                    w[G::K1 - 1] = (s + G::K1 < G::L) ? xr[s + G::K1] : 0.f;
+                }
+                #pragma unroll
+                for (int k = 0; k < G::K1; ++k)
+                    gw1[wave][(c * G::CIN + i) * G::K1 + k] += acc[k];
             }
-            for (; s < G::C1; ++s) a0 = fmaf(da[s], xr[s], a0);
-            gw1[wave][o] += (a0 + a1) + (a2 + a3);
+        } else {
+            // 4 independent partial accumulators: keeps 8 LDS loads in
+            // flight per iteration instead of exposing LDS latency every
+            // element.
+            for (int o = lane; o < 4 * G::CIN * G::K1; o += WAVE) {
+                const int c = o / (G::CIN * G::K1);
+                const int i = (o / G::K1) % G::CIN;
+                const int k = o % G::K1;
+                const float* da = lda1[wave] + c * G::C1;
+                const float* xr = xw + i * (G::L + XP) + k;
+                float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+                int s = 0;
+                for (; s + 4 <= G::C1; s += 4) {
+                    a0 = fmaf(da[s + 0], xr[s + 0], a0);
+                    a1 = fmaf(da[s + 1], xr[s + 1], a1);
+                    a2 = fmaf(da[s + 2], xr[s + 2], a2);
+                    a3 = fmaf(da[s + 3], xr[s + 3], a3);
+                }
+                for (; s < G::C1; ++s) a0 = fmaf(da[s], xr[s], a0);
+                gw1[wave][o] += (a0 + a1) + (a2 + a3);
+            }
         }
         #pragma unroll
         for (int c = 0; c < 4; ++c) {
@@ -765,12 +801,27 @@ int conv_bwd(const float* x, const float* stash, const float* dfeat,
     if (const char* e = getenv("TSKD_CONVBWD_GRID")) cap = atoi(e);
     int grid = min((SN + 3) / 4, cap);
     const char* xp = getenv("TSKD_CONVBWD_PAD");
-    if (xp && xp[0] == '0')
-        hipLaunchKernelGGL((train_conv_bwd_kernel<G, 0>), dim3(grid),
-                           dim3(256), 0, s, x, stash, dfeat, wpack, grads, SN);
-    else
-        hipLaunchKernelGGL((train_conv_bwd_kernel<G, 1>), dim3(grid),
-                           dim3(256), 0, s, x, stash, dfeat, wpack, grads, SN);
+    const char* sl = getenv("TSKD_CONVBWD_SLIDE");
+    const bool slide = !(sl && sl[0] == '0');  // r2 default: on
+    if (xp && xp[0] == '0') {
+        if (slide)
+            hipLaunchKernelGGL((train_conv_bwd_kernel<G, 0, 1>), dim3(grid),
+                               dim3(256), 0, s, x, stash, dfeat, wpack,
+                               grads, SN);
+        else
+            hipLaunchKernelGGL((train_conv_bwd_kernel<G, 0, 0>), dim3(grid),
+                               dim3(256), 0, s, x, stash, dfeat, wpack,
+                               grads, SN);
+    } else {
+        if (slide)
+            hipLaunchKernelGGL((train_conv_bwd_kernel<G, 1, 1>), dim3(grid),
+                               dim3(256), 0, s, x, stash, dfeat, wpack,
+                               grads, SN);
+        else
+            hipLaunchKernelGGL((train_conv_bwd_kernel<G, 1, 0>), dim3(grid),
+                               dim3(256), 0, s, x, stash, dfeat, wpack,
+                               grads, SN);
+    }
     return (int)hipGetLastError();
 }
 // K14: fused batch accuracy — sigmoid -> round -> eq -> count
