@@ -112,7 +112,7 @@ __device__ __forceinline__ float drop_mult_(float p, unsigned long long seed,
 // ---------------------------------------------------------------------------
 // conv forward with stash (fp32; one wave per window, 4 waves per block)
 // ---------------------------------------------------------------------------
-template <class G, bool STAGE_X = true>
+template <class G, bool STAGE_X = true, int PAIR = 0>
 __global__ __launch_bounds__(256) void train_conv_fwd_kernel(
     const float* __restrict__ x,      // (SN, CIN, L)
     float* __restrict__ feat,         // (SN, LIN)
@@ -157,6 +157,37 @@ __global__ __launch_bounds__(256) void train_conv_fwd_kernel(
         } else {
             xw = xin;
         }
+        if (PAIR) {
+            // Paired outputs (round 2): one lane computes TWO adjacent s
+            // positions, whose x windows overlap K1-1 of K1+1 values — the
+            // shared x read cuts LDS/L1 traffic in the dominant loop ~1.8x.
+            constexpr int CP = (G::C1 + 1) / 2;
+            for (int o = lane; o < 4 * CP; o += WAVE) {
+                const int c = o / CP, s0 = (o % CP) * 2;
+                float aa = b1[c], ab = b1[c], a2 = 0.f, b2p = 0.f;
+                const float* wr = w1 + c * (G::CIN * G::K1);
+                const bool has_b = s0 + 1 < G::C1;
+                #pragma unroll
+                for (int i = 0; i < G::CIN; ++i) {
+                    float xv[G::K1 + 1];
+                    #pragma unroll
+                    for (int k = 0; k <= G::K1; ++k)
+                        xv[k] = xw[i * G::L + s0 + k];
+                    #pragma unroll
+                    for (int k = 0; k < G::K1; k += 2) {
+                        aa = fmaf(wr[i * G::K1 + k], xv[k], aa);
+                        ab = fmaf(wr[i * G::K1 + k], xv[k + 1], ab);
+                        if (k + 1 < G::K1) {
+                            a2 = fmaf(wr[i * G::K1 + k + 1], xv[k + 1], a2);
+                            b2p = fmaf(wr[i * G::K1 + k + 1], xv[k + 2],
+                                       b2p);
+                        }
+                    }
+                }
+                c1t[c * G::C1 + s0] = tanh_(aa + a2);
+                if (has_b) c1t[c * G::C1 + s0 + 1] = tanh_(ab + b2p);
+            }
+        } else {
         for (int o = lane; o < 4 * G::C1; o += WAVE) {
             const int c = o / G::C1, s = o % G::C1;
             // two independent partial accumulators (even/odd input channel)
@@ -173,6 +204,7 @@ __global__ __launch_bounds__(256) void train_conv_fwd_kernel(
                 }
             }
             c1t[o] = tanh_(acc + acc2);
+        }
         }
         twsync();
         for (int o = lane; o < 4 * G::P1; o += WAVE) {
@@ -763,12 +795,23 @@ int conv_fwd(const float* x, float* feat, float* stash, const float* wpack,
     // here (no transpose cost, and the dependent FMA chain prefers LDS
     // latency) — the OPPOSITE of the inference tlast conv. Default staged.
     const char* sx = getenv("TSKD_TRAIN_STAGE_X");
-    if (sx && sx[0] == '0')
-        hipLaunchKernelGGL((train_conv_fwd_kernel<G, false>), dim3(grid),
+    const char* pp = getenv("TSKD_CONVFWD_PAIR");
+    const bool pair = !(pp && pp[0] == '0');  // r2 default: paired outputs
+    const bool staged = !(sx && sx[0] == '0');
+    if (!staged && !pair)
+        hipLaunchKernelGGL((train_conv_fwd_kernel<G, false, 0>), dim3(grid),
+                           dim3(256), 0, s, x, feat, stash, wpack, SN, d1, d2,
+                           seed);
+    else if (!staged && pair)
+        hipLaunchKernelGGL((train_conv_fwd_kernel<G, false, 1>), dim3(grid),
+                           dim3(256), 0, s, x, feat, stash, wpack, SN, d1, d2,
+                           seed);
+    else if (staged && pair)
+        hipLaunchKernelGGL((train_conv_fwd_kernel<G, true, 1>), dim3(grid),
                            dim3(256), 0, s, x, feat, stash, wpack, SN, d1, d2,
                            seed);
     else
-        hipLaunchKernelGGL((train_conv_fwd_kernel<G, true>), dim3(grid),
+        hipLaunchKernelGGL((train_conv_fwd_kernel<G, true, 0>), dim3(grid),
                            dim3(256), 0, s, x, feat, stash, wpack, SN, d1, d2,
                            seed);
     return (int)hipGetLastError();
