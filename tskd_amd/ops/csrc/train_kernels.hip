@@ -796,7 +796,10 @@ int conv_fwd(const float* x, float* feat, float* stash, const float* wpack,
     // latency) — the OPPOSITE of the inference tlast conv. Default staged.
     const char* sx = getenv("TSKD_TRAIN_STAGE_X");
     const char* pp = getenv("TSKD_CONVFWD_PAIR");
-    const bool pair = !(pp && pp[0] == '0');  // r2 default: paired outputs
+    // measured +6.3% step time at the bench shape (ab_pair.log) — the
+    // fwd chain is latency- not LDS-throughput-bound; default OFF,
+    // kept as an A/B reference
+    const bool pair = pp && pp[0] == '1';
     const bool staged = !(sx && sx[0] == '0');
     if (!staged && !pair)
         hipLaunchKernelGGL((train_conv_fwd_kernel<G, false, 0>), dim3(grid),
