@@ -376,7 +376,7 @@ __global__ void train_loss_kernel(
 // LSTM backward (BPTT reverse scan). One WAVE per block per sequence.
 // LDS: weight copies for column access + lane-private grad rows.
 // ---------------------------------------------------------------------------
-template <class G>
+template <class G, int RACC = 1>  // RACC: grad accumulators in registers
 __global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
     const float* __restrict__ feat,     // (S, B, LIN)
     const float* __restrict__ dlogit,   // (S, B) dL/d z_base
@@ -392,26 +392,39 @@ __global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
     constexpr int P16 = 17;
     __shared__ float lwih1[64 * LIN], lwhh1[64 * P16];
     __shared__ float lwih2[64 * P16], lwhh2[64 * P16];
-    __shared__ float gwih1[64 * LIN], gwhh1[64 * P16], gb1[64];
-    __shared__ float gwih2[64 * P16], gwhh2[64 * P16], gb2[64];
+    // RACC=1 (round 2): the per-lane grad rows live in REGISTERS (the
+    // lane owns its row exclusively — the LDS read-modify-writes were
+    // ~150 LDS ops per scan step AND the arrays cost 18 KB of LDS per
+    // single-wave workgroup, capping occupancy). RACC=0 keeps the r1
+    // LDS-row layout as the A/B reference.
+    __shared__ float gwih1[RACC ? 1 : 64 * LIN];
+    __shared__ float gwhh1[RACC ? 1 : 64 * P16], gb1[RACC ? 1 : 64];
+    __shared__ float gwih2[RACC ? 1 : 64 * P16];
+    __shared__ float gwhh2[RACC ? 1 : 64 * P16], gb2[RACC ? 1 : 64];
     __shared__ float gout[17];
     __shared__ float lda[64];  // per-step activated-gate grads (both layers)
 
     const int lane = threadIdx.x;
     const int unit = lane & 15;
+    float rwih1[LIN], rwhh1[16], rwih2[16], rwhh2[16];  // dead if !RACC
+    float rb1 = 0.f, rb2 = 0.f, rout = 0.f, routb = 0.f;
+    #pragma unroll
+    for (int j = 0; j < LIN; ++j) rwih1[j] = 0.f;
+    #pragma unroll
+    for (int j = 0; j < 16; ++j) rwhh1[j] = rwih2[j] = rwhh2[j] = 0.f;
     for (int i = lane; i < 64 * LIN; i += WAVE) {
         lwih1[i] = wpack[G::OWIH1 + i];
-        gwih1[i] = 0.f;
+        if (!RACC) gwih1[i] = 0.f;
     }
     for (int i = lane; i < 64 * 16; i += WAVE) {
         const int pi = (i / 16) * P16 + (i % 16);
         lwhh1[pi] = wpack[G::OWHH1 + i];
         lwih2[pi] = wpack[G::OWIH2 + i];
         lwhh2[pi] = wpack[G::OWHH2 + i];
-        gwhh1[pi] = gwih2[pi] = gwhh2[pi] = 0.f;
+        if (!RACC) gwhh1[pi] = gwih2[pi] = gwhh2[pi] = 0.f;
     }
     if (lane < 17) gout[lane] = 0.f;
-    gb1[lane] = gb2[lane] = 0.f;
+    if (!RACC) gb1[lane] = gb2[lane] = 0.f;
     const float outw = wpack[G::OOUTW + unit];
     twsync();
 
@@ -447,18 +460,27 @@ __global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
             else if (lane < 48) da2 = dc2 * i2 * (1.f - g2 * g2);
             else da2 = dh2 * tc2 * o2 * (1.f - o2);
             lda[lane] = da2;
-            // weight grads: lane-private LDS rows
+            // weight grads: lane-private rows (registers when RACC)
             #pragma unroll
             for (int j = 0; j < 16; ++j) {
                 const float h1j = sp[G::SL_H1 + j];
                 const float h2pj = spm ? spm[G::SL_H2 + j] : 0.f;
-                gwih2[lane * P16 + j] = fmaf(da2, h1j, gwih2[lane * P16 + j]);
-                gwhh2[lane * P16 + j] = fmaf(da2, h2pj, gwhh2[lane * P16 + j]);
+                if (RACC) {
+                    rwih2[j] = fmaf(da2, h1j, rwih2[j]);
+                    rwhh2[j] = fmaf(da2, h2pj, rwhh2[j]);
+                } else {
+                    gwih2[lane * P16 + j] = fmaf(da2, h1j,
+                                                 gwih2[lane * P16 + j]);
+                    gwhh2[lane * P16 + j] = fmaf(da2, h2pj,
+                                                 gwhh2[lane * P16 + j]);
+                }
             }
-            gb2[lane] += da2;
-            if (lane < 16) gout[lane] = fmaf(dzt, sp[G::SL_H2 + lane],
-                                             gout[lane]);
-            if (lane == 0) gout[16] += dzt;
+            if (RACC) rb2 += da2; else gb2[lane] += da2;
+            if (lane < 16) {
+                if (RACC) rout = fmaf(dzt, sp[G::SL_H2 + lane], rout);
+                else gout[lane] = fmaf(dzt, sp[G::SL_H2 + lane], gout[lane]);
+            }
+            if (lane == 0) { if (RACC) routb += dzt; else gout[16] += dzt; }
             twsync();
             // column reductions: dh1 (into layer1) and dh2_{t-1}
             // lane (u + 16k) sums rows [16k, 16k+16)
@@ -499,16 +521,19 @@ __global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
                 lda[lane] = da1;
                 const float* xt = fs + (long)t * LIN;
                 #pragma unroll
-                for (int j = 0; j < LIN; ++j)
-                    gwih1[lane * LIN + j] = fmaf(da1, xt[j],
-                                                 gwih1[lane * LIN + j]);
+                for (int j = 0; j < LIN; ++j) {
+                    if (RACC) rwih1[j] = fmaf(da1, xt[j], rwih1[j]);
+                    else gwih1[lane * LIN + j] = fmaf(da1, xt[j],
+                                                      gwih1[lane * LIN + j]);
+                }
                 #pragma unroll
                 for (int j = 0; j < 16; ++j) {
                     const float h1pj = spm ? spm[G::SL_H1 + j] : 0.f;
-                    gwhh1[lane * P16 + j] = fmaf(da1, h1pj,
-                                                gwhh1[lane * P16 + j]);
+                    if (RACC) rwhh1[j] = fmaf(da1, h1pj, rwhh1[j]);
+                    else gwhh1[lane * P16 + j] = fmaf(da1, h1pj,
+                                                      gwhh1[lane * P16 + j]);
                 }
-                gb1[lane] += da1;
+                if (RACC) rb1 += da1; else gb1[lane] += da1;
                 dc1n = dc1 * f1;
                 twsync();
                 // dh1_{t-1} and dfeat
@@ -533,6 +558,22 @@ __global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
     }
     // fold the per-block accumulators into global grads
     twsync();
+    if (RACC) {
+        #pragma unroll
+        for (int j = 0; j < LIN; ++j)
+            atomicAdd(&grads[G::OWIH1 + lane * LIN + j], rwih1[j]);
+        #pragma unroll
+        for (int j = 0; j < 16; ++j) {
+            atomicAdd(&grads[G::OWHH1 + lane * 16 + j], rwhh1[j]);
+            atomicAdd(&grads[G::OWIH2 + lane * 16 + j], rwih2[j]);
+            atomicAdd(&grads[G::OWHH2 + lane * 16 + j], rwhh2[j]);
+        }
+        atomicAdd(&grads[G::OBL1 + lane], rb1);
+        atomicAdd(&grads[G::OBL2 + lane], rb2);
+        if (lane < 16) atomicAdd(&grads[G::OOUTW + lane], rout);
+        if (lane == 0) atomicAdd(&grads[G::OOUTB], routb);
+        return;
+    }
     for (int i = lane; i < 64 * LIN; i += WAVE)
         atomicAdd(&grads[G::OWIH1 + i], gwih1[i]);
     for (int i = lane; i < 64 * 16; i += WAVE) {
@@ -834,9 +875,16 @@ int lstm_bwd(const float* feat, const float* dlogit, const float* stash,
              const float* wpack, float* grads, float* dfeat, int S, int B,
              hipStream_t s) {
     if (S <= 0) return 0;
-    hipLaunchKernelGGL((train_lstm_bwd_kernel<G>), dim3(min(S, 32768)),
-                       dim3(WAVE), 0, s, feat, dlogit, stash, wpack, grads,
-                       dfeat, S, B);
+    // TSKD_LSTMBWD_RACC=0 restores the r1 LDS grad-row layout (A/B ref)
+    const char* ra = getenv("TSKD_LSTMBWD_RACC");
+    if (ra && ra[0] == '0')
+        hipLaunchKernelGGL((train_lstm_bwd_kernel<G, 0>),
+                           dim3(min(S, 32768)), dim3(WAVE), 0, s, feat,
+                           dlogit, stash, wpack, grads, dfeat, S, B);
+    else
+        hipLaunchKernelGGL((train_lstm_bwd_kernel<G, 1>),
+                           dim3(min(S, 32768)), dim3(WAVE), 0, s, feat,
+                           dlogit, stash, wpack, grads, dfeat, S, B);
     return (int)hipGetLastError();
 }
 template <class G>
