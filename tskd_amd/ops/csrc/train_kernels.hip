@@ -875,14 +875,17 @@ int lstm_bwd(const float* feat, const float* dlogit, const float* stash,
              const float* wpack, float* grads, float* dfeat, int S, int B,
              hipStream_t s) {
     if (S <= 0) return 0;
-    // TSKD_LSTMBWD_RACC=0 restores the r1 LDS grad-row layout (A/B ref)
+    // TSKD_LSTMBWD_RACC=1 moves the per-lane grad rows to registers —
+    // measured +29% step time at the bench shape (ab_racc.log: the ~75
+    // extra live VGPRs cost more occupancy than the LDS RMWs cost
+    // bandwidth). Default stays the r1 LDS layout; variant kept as A/B.
     const char* ra = getenv("TSKD_LSTMBWD_RACC");
-    if (ra && ra[0] == '0')
-        hipLaunchKernelGGL((train_lstm_bwd_kernel<G, 0>),
+    if (ra && ra[0] == '1')
+        hipLaunchKernelGGL((train_lstm_bwd_kernel<G, 1>),
                            dim3(min(S, 32768)), dim3(WAVE), 0, s, feat,
                            dlogit, stash, wpack, grads, dfeat, S, B);
     else
-        hipLaunchKernelGGL((train_lstm_bwd_kernel<G, 1>),
+        hipLaunchKernelGGL((train_lstm_bwd_kernel<G, 0>),
                            dim3(min(S, 32768)), dim3(WAVE), 0, s, feat,
                            dlogit, stash, wpack, grads, dfeat, S, B);
     return (int)hipGetLastError();
