@@ -187,6 +187,34 @@ __global__ __launch_bounds__(256) void train_conv_fwd_kernel(
                 c1t[c * G::C1 + s0] = tanh_(aa + a2);
                 if (has_b) c1t[c * G::C1 + s0 + 1] = tanh_(ab + b2p);
             }
+        } else if (PAIR == 2) {
+            // 4 independent accumulators (i parity × k parity): halves the
+            // dependent-FMA chain depth vs the 2-accumulator default
+            for (int o = lane; o < 4 * G::C1; o += WAVE) {
+                const int c = o / G::C1, s = o % G::C1;
+                float a0 = b1[c], a1 = 0.f, a2 = 0.f, a3 = 0.f;
+                const float* wr = w1 + c * (G::CIN * G::K1);
+                #pragma unroll
+                for (int i = 0; i < G::CIN; i += 2) {
+                    #pragma unroll
+                    for (int k = 0; k < G::K1; k += 2) {
+                        a0 = fmaf(wr[i * G::K1 + k],
+                                  xw[i * G::L + s + k], a0);
+                        if (k + 1 < G::K1)
+                            a1 = fmaf(wr[i * G::K1 + k + 1],
+                                      xw[i * G::L + s + k + 1], a1);
+                        if (i + 1 < G::CIN) {
+                            a2 = fmaf(wr[(i + 1) * G::K1 + k],
+                                      xw[(i + 1) * G::L + s + k], a2);
+                            if (k + 1 < G::K1)
+                                a3 = fmaf(wr[(i + 1) * G::K1 + k + 1],
+                                          xw[(i + 1) * G::L + s + k + 1],
+                                          a3);
+                        }
+                    }
+                }
+                c1t[o] = tanh_((a0 + a1) + (a2 + a3));
+            }
         } else {
         for (int o = lane; o < 4 * G::C1; o += WAVE) {
             const int c = o / G::C1, s = o % G::C1;
@@ -836,28 +864,38 @@ int conv_fwd(const float* x, float* feat, float* stash, const float* wpack,
     // here (no transpose cost, and the dependent FMA chain prefers LDS
     // latency) — the OPPOSITE of the inference tlast conv. Default staged.
     const char* sx = getenv("TSKD_TRAIN_STAGE_X");
+    // TSKD_CONVFWD_PAIR: 0 (default) = 2-accumulator loop; 1 = paired
+    // outputs (measured +6.3%, A/B ref); 2 = 4-accumulator chain split
     const char* pp = getenv("TSKD_CONVFWD_PAIR");
-    // measured +6.3% step time at the bench shape (ab_pair.log) — the
-    // fwd chain is latency- not LDS-throughput-bound; default OFF,
-    // kept as an A/B reference
-    const bool pair = pp && pp[0] == '1';
+    const int pair = pp ? atoi(pp) : 0;
     const bool staged = !(sx && sx[0] == '0');
-    if (!staged && !pair)
-        hipLaunchKernelGGL((train_conv_fwd_kernel<G, false, 0>), dim3(grid),
-                           dim3(256), 0, s, x, feat, stash, wpack, SN, d1, d2,
-                           seed);
-    else if (!staged && pair)
-        hipLaunchKernelGGL((train_conv_fwd_kernel<G, false, 1>), dim3(grid),
-                           dim3(256), 0, s, x, feat, stash, wpack, SN, d1, d2,
-                           seed);
-    else if (staged && pair)
-        hipLaunchKernelGGL((train_conv_fwd_kernel<G, true, 1>), dim3(grid),
-                           dim3(256), 0, s, x, feat, stash, wpack, SN, d1, d2,
-                           seed);
-    else
-        hipLaunchKernelGGL((train_conv_fwd_kernel<G, true, 0>), dim3(grid),
-                           dim3(256), 0, s, x, feat, stash, wpack, SN, d1, d2,
-                           seed);
+    if (!staged) {
+        if (pair == 1)
+            hipLaunchKernelGGL((train_conv_fwd_kernel<G, false, 1>),
+                               dim3(grid), dim3(256), 0, s, x, feat, stash,
+                               wpack, SN, d1, d2, seed);
+        else if (pair == 2)
+            hipLaunchKernelGGL((train_conv_fwd_kernel<G, false, 2>),
+                               dim3(grid), dim3(256), 0, s, x, feat, stash,
+                               wpack, SN, d1, d2, seed);
+        else
+            hipLaunchKernelGGL((train_conv_fwd_kernel<G, false, 0>),
+                               dim3(grid), dim3(256), 0, s, x, feat, stash,
+                               wpack, SN, d1, d2, seed);
+    } else {
+        if (pair == 1)
+            hipLaunchKernelGGL((train_conv_fwd_kernel<G, true, 1>),
+                               dim3(grid), dim3(256), 0, s, x, feat, stash,
+                               wpack, SN, d1, d2, seed);
+        else if (pair == 2)
+            hipLaunchKernelGGL((train_conv_fwd_kernel<G, true, 2>),
+                               dim3(grid), dim3(256), 0, s, x, feat, stash,
+                               wpack, SN, d1, d2, seed);
+        else
+            hipLaunchKernelGGL((train_conv_fwd_kernel<G, true, 0>),
+                               dim3(grid), dim3(256), 0, s, x, feat, stash,
+                               wpack, SN, d1, d2, seed);
+    }
     return (int)hipGetLastError();
 }
 template <class G>
