@@ -182,6 +182,7 @@ def scenario_live(device: str, n_patients: int, rate_eps: int,
         time.sleep(min(3.0, 8 * trigger_period_s))
         srv.trigger()
         srv.se.force_ready()
+        srv.stage_profile = True  # poll/ingest/model/persist decomposition
         lats = []
         for i in range(n_triggers):
             t_start = time.perf_counter()
@@ -215,6 +216,9 @@ def scenario_live(device: str, n_patients: int, rate_eps: int,
                 "max": round(lats[-1], 3) if lats else None,
                 "n": len(lats),
             },
+            "stage_p50_ms": {
+                k: (round(statistics.median(v), 3) if v else None)
+                for k, v in srv.stage_ms.items()},
             "note": ("latency = store-insert wall time minus produce wall "
                      "time of the newest consumed sample (event time == "
                      "wall time); excludes the semantic watermark delay "
