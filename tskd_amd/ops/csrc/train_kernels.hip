@@ -404,8 +404,12 @@ __global__ void train_loss_kernel(
 // LSTM backward (BPTT reverse scan). One WAVE per block per sequence.
 // LDS: weight copies for column access + lane-private grad rows.
 // ---------------------------------------------------------------------------
-template <class G, int RACC = 1>  // RACC: grad accumulators in registers
-__global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
+template <class G, int RACC = 1, int WPB = 1>
+// RACC: grad accumulators in registers (A/B ref; measured +29%)
+// WPB: sequences (waves) per block sharing ONE read-only weight copy —
+//      the 19 KB LDS weight stage was per-sequence at WPB=1, capping
+//      occupancy at ~4 single-wave workgroups/CU
+__global__ __launch_bounds__(WAVE * WPB) void train_lstm_bwd_kernel(
     const float* __restrict__ feat,     // (S, B, LIN)
     const float* __restrict__ dlogit,   // (S, B) dL/d z_base
     const float* __restrict__ stash,    // (S, B, SL_SIZE)
@@ -420,43 +424,56 @@ __global__ __launch_bounds__(WAVE) void train_lstm_bwd_kernel(
     constexpr int P16 = 17;
     __shared__ float lwih1[64 * LIN], lwhh1[64 * P16];
     __shared__ float lwih2[64 * P16], lwhh2[64 * P16];
-    // RACC=1 (round 2): the per-lane grad rows live in REGISTERS (the
-    // lane owns its row exclusively — the LDS read-modify-writes were
-    // ~150 LDS ops per scan step AND the arrays cost 18 KB of LDS per
-    // single-wave workgroup, capping occupancy). RACC=0 keeps the r1
-    // LDS-row layout as the A/B reference.
-    __shared__ float gwih1[RACC ? 1 : 64 * LIN];
-    __shared__ float gwhh1[RACC ? 1 : 64 * P16], gb1[RACC ? 1 : 64];
-    __shared__ float gwih2[RACC ? 1 : 64 * P16];
-    __shared__ float gwhh2[RACC ? 1 : 64 * P16], gb2[RACC ? 1 : 64];
-    __shared__ float gout[17];
-    __shared__ float lda[64];  // per-step activated-gate grads (both layers)
+    // RACC=1: the per-lane grad rows live in REGISTERS. RACC=0 keeps the
+    // r1 LDS-row layout (per wave when WPB > 1).
+    __shared__ float sgwih1[WPB][RACC ? 1 : 64 * LIN];
+    __shared__ float sgwhh1[WPB][RACC ? 1 : 64 * P16];
+    __shared__ float sgb1[WPB][RACC ? 1 : 64];
+    __shared__ float sgwih2[WPB][RACC ? 1 : 64 * P16];
+    __shared__ float sgwhh2[WPB][RACC ? 1 : 64 * P16];
+    __shared__ float sgb2[WPB][RACC ? 1 : 64];
+    __shared__ float sgout[WPB][17];
+    __shared__ float slda[WPB][64];  // per-step activated-gate grads
 
-    const int lane = threadIdx.x;
+    const int lane = threadIdx.x % WAVE;
+    const int wv = threadIdx.x / WAVE;
     const int unit = lane & 15;
+    float* gwih1 = sgwih1[wv];
+    float* gwhh1 = sgwhh1[wv];
+    float* gb1 = sgb1[wv];
+    float* gwih2 = sgwih2[wv];
+    float* gwhh2 = sgwhh2[wv];
+    float* gb2 = sgb2[wv];
+    float* gout = sgout[wv];
+    float* lda = slda[wv];
     float rwih1[LIN], rwhh1[16], rwih2[16], rwhh2[16];  // dead if !RACC
     float rb1 = 0.f, rb2 = 0.f, rout = 0.f, routb = 0.f;
     #pragma unroll
     for (int j = 0; j < LIN; ++j) rwih1[j] = 0.f;
     #pragma unroll
     for (int j = 0; j < 16; ++j) rwhh1[j] = rwih2[j] = rwhh2[j] = 0.f;
-    for (int i = lane; i < 64 * LIN; i += WAVE) {
+    for (int i = threadIdx.x; i < 64 * LIN; i += WAVE * WPB)
         lwih1[i] = wpack[G::OWIH1 + i];
-        if (!RACC) gwih1[i] = 0.f;
-    }
-    for (int i = lane; i < 64 * 16; i += WAVE) {
+    for (int i = threadIdx.x; i < 64 * 16; i += WAVE * WPB) {
         const int pi = (i / 16) * P16 + (i % 16);
         lwhh1[pi] = wpack[G::OWHH1 + i];
         lwih2[pi] = wpack[G::OWIH2 + i];
         lwhh2[pi] = wpack[G::OWHH2 + i];
-        if (!RACC) gwhh1[pi] = gwih2[pi] = gwhh2[pi] = 0.f;
+    }
+    if (!RACC) {
+        for (int i = lane; i < 64 * LIN; i += WAVE) gwih1[i] = 0.f;
+        for (int i = lane; i < 64 * 16; i += WAVE) {
+            const int pi = (i / 16) * P16 + (i % 16);
+            gwhh1[pi] = gwih2[pi] = gwhh2[pi] = 0.f;
+        }
+        gb1[lane] = gb2[lane] = 0.f;
     }
     if (lane < 17) gout[lane] = 0.f;
-    if (!RACC) gb1[lane] = gb2[lane] = 0.f;
     const float outw = wpack[G::OOUTW + unit];
-    twsync();
+    if (WPB > 1) __syncthreads(); else twsync();
 
-    for (long seq = blockIdx.x; seq < S; seq += gridDim.x) {
+    for (long seq = blockIdx.x * WPB + wv; seq < S;
+         seq += (long)gridDim.x * WPB) {
         const float* fs = feat + seq * (long)B * LIN;
         const float* dz = dlogit + seq * (long)B;
         const float* ss = stash + seq * (long)B * G::SL_SIZE;
@@ -917,8 +934,23 @@ int lstm_bwd(const float* feat, const float* dlogit, const float* stash,
     // measured +29% step time at the bench shape (ab_racc.log: the ~75
     // extra live VGPRs cost more occupancy than the LDS RMWs cost
     // bandwidth). Default stays the r1 LDS layout; variant kept as A/B.
+    // TSKD_LSTMBWD_WPB=2 packs two sequences (waves) per block sharing
+    // one LDS weight copy (halves the per-sequence weight-stage cost).
     const char* ra = getenv("TSKD_LSTMBWD_RACC");
-    if (ra && ra[0] == '1')
+    const char* wb = getenv("TSKD_LSTMBWD_WPB");
+    const bool racc = ra && ra[0] == '1';
+    const int wpb = wb ? atoi(wb) : 1;
+    if (wpb == 2) {
+        const int grid = min((S + 1) / 2, 32768);
+        if (racc)
+            hipLaunchKernelGGL((train_lstm_bwd_kernel<G, 1, 2>), dim3(grid),
+                               dim3(2 * WAVE), 0, s, feat, dlogit, stash,
+                               wpack, grads, dfeat, S, B);
+        else
+            hipLaunchKernelGGL((train_lstm_bwd_kernel<G, 0, 2>), dim3(grid),
+                               dim3(2 * WAVE), 0, s, feat, dlogit, stash,
+                               wpack, grads, dfeat, S, B);
+    } else if (racc)
         hipLaunchKernelGGL((train_lstm_bwd_kernel<G, 1>),
                            dim3(min(S, 32768)), dim3(WAVE), 0, s, feat,
                            dlogit, stash, wpack, grads, dfeat, S, B);
