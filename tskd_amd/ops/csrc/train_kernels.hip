@@ -647,8 +647,11 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
 {
     __shared__ float lw2[21];  // conv2 weights + bias
     // row stride L+1: L=120 dwords is 24 mod 32 banks (partial conflicts
-    // on the stride-L xw reads in the conv1-grad loop); +1 spreads banks
-    __shared__ float lx[4][G::CIN * (G::L + XP)];
+    // on the stride-L xw reads in the conv1-grad loop); +1 spreads banks.
+    // SLIDE=2 reads x straight from global (the sliding loop touches each
+    // x element once, L1-streamed) — the 19 KB stage is then dead weight
+    __shared__ float lx[SLIDE == 2 ? 1 : 4]
+                       [SLIDE == 2 ? 1 : G::CIN * (G::L + XP)];
     __shared__ float lp1[4][4 * G::P1];
     __shared__ float lda1[4][4 * G::C1];
     __shared__ float lda2[4][G::C2];
@@ -678,9 +681,10 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
         const float* m2s = st + G::SC_M2;
         const float* dfw = dfeat + win * G::LIN;
         const float* xin = x + win * (G::CIN * G::L);
-        float* xw = lx[wave];
-        for (int i = lane; i < G::CIN * G::L; i += WAVE)
-            xw[i + XP * (i / G::L)] = xin[i];  // padded-row store
+        float* xw = lx[SLIDE == 2 ? 0 : wave];
+        if (SLIDE != 2)
+            for (int i = lane; i < G::CIN * G::L; i += WAVE)
+                xw[i + XP * (i / G::L)] = xin[i];  // padded-row store
         // recompute the (dropout-masked) pool1 output = conv2's input
         for (int o = lane; o < 4 * G::P1; o += WAVE) {
             const int c = o / G::P1, q = o % G::P1;
@@ -765,7 +769,8 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
             const int i = lane % G::CIN;
             if (lane < 4 * G::CIN) {
                 const float* da = lda1[wave] + c * G::C1;
-                const float* xr = xw + i * (G::L + XP);
+                const float* xr = (SLIDE == 2) ? (xin + i * G::L)
+                                               : (xw + i * (G::L + XP));
                 float acc[G::K1];
                 float w[G::K1];
                 #pragma unroll
@@ -969,9 +974,15 @@ int conv_bwd(const float* x, const float* stash, const float* dfeat,
     int grid = min((SN + 3) / 4, cap);
     const char* xp = getenv("TSKD_CONVBWD_PAD");
     const char* sl = getenv("TSKD_CONVBWD_SLIDE");
-    const bool slide = !(sl && sl[0] == '0');  // r2 default: on
+    // 0 = r1 loop; 1 (default) = sliding-register (-22.1%); 2 = sliding
+    // with global-x reads (no LDS stage)
+    const int slide = sl ? atoi(sl) : 1;
     if (xp && xp[0] == '0') {
-        if (slide)
+        if (slide == 2)
+            hipLaunchKernelGGL((train_conv_bwd_kernel<G, 0, 2>), dim3(grid),
+                               dim3(256), 0, s, x, stash, dfeat, wpack,
+                               grads, SN);
+        else if (slide == 1)
             hipLaunchKernelGGL((train_conv_bwd_kernel<G, 0, 1>), dim3(grid),
                                dim3(256), 0, s, x, stash, dfeat, wpack,
                                grads, SN);
@@ -980,7 +991,11 @@ int conv_bwd(const float* x, const float* stash, const float* dfeat,
                                dim3(256), 0, s, x, stash, dfeat, wpack,
                                grads, SN);
     } else {
-        if (slide)
+        if (slide == 2)
+            hipLaunchKernelGGL((train_conv_bwd_kernel<G, 1, 2>), dim3(grid),
+                               dim3(256), 0, s, x, stash, dfeat, wpack,
+                               grads, SN);
+        else if (slide == 1)
             hipLaunchKernelGGL((train_conv_bwd_kernel<G, 1, 1>), dim3(grid),
                                dim3(256), 0, s, x, stash, dfeat, wpack,
                                grads, SN);
