@@ -974,9 +974,10 @@ int conv_bwd(const float* x, const float* stash, const float* dfeat,
     int grid = min((SN + 3) / 4, cap);
     const char* xp = getenv("TSKD_CONVBWD_PAD");
     const char* sl = getenv("TSKD_CONVBWD_SLIDE");
-    // 0 = r1 loop; 1 (default) = sliding-register (-22.1%); 2 = sliding
-    // with global-x reads (no LDS stage)
-    const int slide = sl ? atoi(sl) : 1;
+    // 0 = r1 loop; 1 = sliding-register (-22.1% vs 0, ab_slide.log);
+    // 2 (default) = sliding with global-x reads, no LDS stage (-4.5%
+    // further, ab_slide2.log)
+    const int slide = sl ? atoi(sl) : 2;
     if (xp && xp[0] == '0') {
         if (slide == 2)
             hipLaunchKernelGGL((train_conv_bwd_kernel<G, 0, 2>), dim3(grid),
