@@ -650,8 +650,8 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
     // on the stride-L xw reads in the conv1-grad loop); +1 spreads banks.
     // SLIDE=2 reads x straight from global (the sliding loop touches each
     // x element once, L1-streamed) — the 19 KB stage is then dead weight
-    __shared__ float lx[SLIDE == 2 ? 1 : 4]
-                       [SLIDE == 2 ? 1 : G::CIN * (G::L + XP)];
+    __shared__ float lx[SLIDE >= 2 ? 1 : 4]
+                       [SLIDE >= 2 ? 1 : G::CIN * (G::L + XP)];
     __shared__ float lp1[4][4 * G::P1];
     __shared__ float lda1[4][4 * G::C1];
     __shared__ float lda2[4][G::C2];
@@ -681,8 +681,8 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
         const float* m2s = st + G::SC_M2;
         const float* dfw = dfeat + win * G::LIN;
         const float* xin = x + win * (G::CIN * G::L);
-        float* xw = lx[SLIDE == 2 ? 0 : wave];
-        if (SLIDE != 2)
+        float* xw = lx[SLIDE >= 2 ? 0 : wave];
+        if (SLIDE < 2)
             for (int i = lane; i < G::CIN * G::L; i += WAVE)
                 xw[i + XP * (i / G::L)] = xin[i];  // padded-row store
         // recompute the (dropout-masked) pool1 output = conv2's input
@@ -769,23 +769,53 @@ __global__ __launch_bounds__(256) void train_conv_bwd_kernel(
             const int i = lane % G::CIN;
             if (lane < 4 * G::CIN) {
                 const float* da = lda1[wave] + c * G::C1;
-                const float* xr = (SLIDE == 2) ? (xin + i * G::L)
+                const float* xr = (SLIDE >= 2) ? (xin + i * G::L)
                                                : (xw + i * (G::L + XP));
                 float acc[G::K1];
-                float w[G::K1];
                 #pragma unroll
-                for (int k = 0; k < G::K1; ++k) {
-                    acc[k] = 0.f;
-                    w[k] = xr[k];
-                }
-                for (int s = 0; s < G::C1; ++s) {
-                    const float d = da[s];
+                for (int k = 0; k < G::K1; ++k) acc[k] = 0.f;
+                if (SLIDE == 3) {
+                    // K1-deep double-buffered prefetch: the 1-ahead shift
+                    // (below) exposes an L1-latency wait EVERY step (PMC:
+                    // 76% SQ_WAIT_ANY); batching K1 independent loads per
+                    // K1 steps amortizes it K1x.
+                    float w[2 * G::K1];
                     #pragma unroll
-                    for (int k = 0; k < G::K1; ++k)
-                        acc[k] = fmaf(d, w[k], acc[k]);
+                    for (int k = 0; k < 2 * G::K1; ++k)
+                        w[k] = (k < G::L) ? xr[k] : 0.f;
+                    for (int s0 = 0; s0 < G::C1; s0 += G::K1) {
+                        #pragma unroll
+                        for (int t = 0; t < G::K1; ++t) {
+                            const int s = s0 + t;
+                            if (s < G::C1) {
+                                const float d = da[s];
+                                #pragma unroll
+                                for (int k = 0; k < G::K1; ++k)
+                                    acc[k] = fmaf(d, w[t + k], acc[k]);
+                            }
+                        }
+                        #pragma unroll
+                        for (int k = 0; k < G::K1; ++k) w[k] = w[k + G::K1];
+                        #pragma unroll
+                        for (int t = 0; t < G::K1; ++t) {
+                            const int idx = s0 + 2 * G::K1 + t;
+                            w[G::K1 + t] = (idx < G::L) ? xr[idx] : 0.f;
+                        }
+                    }
+                } else {
+                    float w[G::K1];
                     #pragma unroll
-                    for (int k = 0; k < G::K1 - 1; ++k) w[k] = w[k + 1];
-                    w[G::K1 - 1] = (s + G::K1 < G::L) ? xr[s + G::K1] : 0.f;
+                    for (int k = 0; k < G::K1; ++k) w[k] = xr[k];
+                    for (int s = 0; s < G::C1; ++s) {
+                        const float d = da[s];
+                        #pragma unroll
+                        for (int k = 0; k < G::K1; ++k)
+                            acc[k] = fmaf(d, w[k], acc[k]);
+                        #pragma unroll
+                        for (int k = 0; k < G::K1 - 1; ++k) w[k] = w[k + 1];
+                        w[G::K1 - 1] = (s + G::K1 < G::L) ? xr[s + G::K1]
+                                                          : 0.f;
+                    }
                 }
                 #pragma unroll
                 for (int k = 0; k < G::K1; ++k)
