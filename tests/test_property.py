@@ -182,3 +182,56 @@ class TestEngineIncrementalProperty:
         np.testing.assert_allclose(
             inc.proc[:, :, :inc.nproc].numpy(),
             one.proc[:, :, :one.nproc].numpy(), rtol=1e-5, atol=1e-6)
+
+
+class TestPollSamplesSidEquivalence:
+    """The one-pass native edge (poll_samples_sid) must agree exactly with
+    poll_samples + the Python key->sid/shard mapping it replaced."""
+
+    @given(st.lists(st.tuples(st.integers(0, 30),      # patient index
+                              st.integers(0, 9),       # channel
+                              st.floats(-500, 500, allow_nan=False),
+                              st.booleans()),           # parseable?
+                    min_size=0, max_size=120),
+           st.integers(1, 4))                           # world
+    @settings(max_examples=25, deadline=None)
+    def test_matches_python_mapping(self, msgs, world):
+        import tempfile
+
+        from tskd_amd.bus import Bus, Consumer, Producer
+        from tskd_amd.parallel.dist import shard_for_key
+        with tempfile.TemporaryDirectory() as d:
+            bus = Bus(d)
+            bus.create_topic("T")
+            prod = Producer(bus)
+            for i, (pi, ch, val, ok) in enumerate(msgs):
+                pid = f"p{pi:06d}"
+                body = f"[{ch}, {val!r}]" if ok else "not json"
+                prod.produce("T", pid, body, ts_us=1000 * (i + 1))
+            for rank in range(world):
+                ca = Consumer(bus, starting="earliest")
+                ca.subscribe(["T"])
+                keys, _t, chans, vals, tss = ca.poll_samples(max_msgs=1000)
+                # reference mapping in python
+                sid_map, exp = {}, []
+                for k, c, v, t in zip(keys, chans, vals, tss):
+                    if shard_for_key(k, world) != rank:
+                        continue
+                    if k not in sid_map:
+                        sid_map[k] = len(sid_map)
+                    exp.append((sid_map[k], int(c), float(v), float(t)))
+                cb = Consumer(bus, starting="earliest")
+                cb.subscribe(["T"])
+                sa, cb_, va, ta, nk = cb.poll_samples_sid(
+                    max_msgs=1000, rank=rank, world=world)
+                got = list(zip(sa.tolist(), cb_.tolist(), va.tolist(),
+                               ta.tolist()))
+                assert [(s, c) for s, c, _, _ in got] == \
+                    [(s, c) for s, c, _, _ in exp]
+                np.testing.assert_allclose([v for *_, v, _ in got],
+                                           [v for *_, v, _ in exp],
+                                           rtol=1e-6)
+                np.testing.assert_allclose([t for *_, t in got],
+                                           [t for *_, t in exp], rtol=0,
+                                           atol=0)
+                assert dict(nk) == sid_map
