@@ -39,6 +39,11 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 import torch  # noqa: E402
 
 
+def _sync():
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+
+
 def run_steps(step_fn, steps, warmup, dist, device, min_warm_s=None):
     if min_warm_s is None:
         # TSKD_BENCH_MINWARM=0 disables the extension (rocprof runs: the
@@ -53,7 +58,7 @@ def run_steps(step_fn, steps, warmup, dist, device, min_warm_s=None):
     w0 = time.perf_counter()
     for _ in range(warmup):
         step_fn()
-    torch.cuda.synchronize()
+    _sync()
     elapsed = time.perf_counter() - w0
     per_step = max(elapsed / max(warmup, 1), 1e-4)
     n_extra = max(0, int((min_warm_s - elapsed) / per_step) + 1) \
@@ -66,20 +71,20 @@ def run_steps(step_fn, steps, warmup, dist, device, min_warm_s=None):
     n_extra = min(n_extra, 100_000)
     for _ in range(n_extra):
         step_fn()
-    torch.cuda.synchronize()
+    _sync()
     if dist:
         dist.barrier()
-    torch.cuda.synchronize()
+    _sync()
     lat = []
     t0 = time.perf_counter()
     for _ in range(steps):
         s0 = time.perf_counter()
         step_fn()
-        torch.cuda.synchronize()
+        _sync()
         lat.append(time.perf_counter() - s0)
     if dist:
         dist.barrier()
-    torch.cuda.synchronize()
+    _sync()
     t1 = time.perf_counter()
     red_dev = device if (dist and dist.get_backend() == "nccl") else "cpu"
     elapsed = torch.tensor([t1 - t0], device=red_dev, dtype=torch.float64)
@@ -204,7 +209,7 @@ def main() -> None:
                                          timelast=True, capture=False)
                 while se.nproc < se.head - se.win_buckets + 1 or se.nproc == 0:
                     se.ingest_dense(raw, chan_map=chan_map)
-                torch.cuda.synchronize()
+                _sync()
                 tg = TriggerGraph(se, raw, chan_map, graphed, stride=12,
                                   overlap=args.overlap)
             except Exception as e:  # pragma: no cover - fallback safety
@@ -220,10 +225,10 @@ def main() -> None:
         if tg is not None:
             lp = []
             for _ in range(20):
-                torch.cuda.synchronize()
+                _sync()
                 t0 = time.perf_counter()
                 tg.replay()
-                torch.cuda.synchronize()
+                _sync()
                 lp.append((time.perf_counter() - t0) * 1e3)
             trigger_latency_ms = statistics.median(lp)
 

@@ -163,9 +163,15 @@ def main() -> None:
         torch.manual_seed(0)
         tr = MyCNNHipTrainer(build_model("MyCNN5"), device="cuda",
                              lr=1e-5, pos_weight=pos_w)
+        # mini-batch parity with the torch trainer: one optimizer step per
+        # 1-sequence chunk (same number of gradient steps per epoch)
         losses = []
         for ep in range(args.epochs):
-            losses.append(tr.step(xs, as_, ys))
+            ep_loss = 0.0
+            for ci in range(xs.shape[0]):
+                ep_loss += tr.step(xs[ci:ci + 1], as_[ci:ci + 1],
+                                   ys[ci:ci + 1])
+            losses.append(ep_loss / xs.shape[0])
         hip_model = tr.export_model().cpu().eval()
         hip_ckpt = os.path.join(args.out_dir, "MyCNN5_p000194_hip.pth")
         from tskd_amd.models import save_checkpoint

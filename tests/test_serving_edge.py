@@ -167,3 +167,56 @@ class TestServingLatencyGpu:
         drain = next(x for x in recs if x["scenario"] == "drain")
         assert drain["events_drained"] == drain["n_events"]
         assert drain["n_predictions"] > 0
+
+
+def _bench_warmup_worker(rank, world, port, out_dir):
+    import time
+
+    import torch.distributed as dist
+    os.environ.update({"MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+                       "RANK": str(rank), "WORLD_SIZE": str(world)})
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    sys.path.insert(0, REPO)
+    from bench import run_steps
+    calls = [0]
+
+    def step():
+        calls[0] += 1
+        # a collective per step + rank-dependent speed: if the warmup
+        # extension were wall-clock-paced per rank (the r2 deadlock bug),
+        # ranks would issue different collective counts and hang here
+        t = torch.ones(1)
+        dist.all_reduce(t)
+        time.sleep(0.001 * (1 + 4 * rank))
+
+    elapsed, lat = run_steps(step, steps=3, warmup=2, dist=dist,
+                             device="cpu", min_warm_s=0.2)
+    with open(os.path.join(out_dir, f"warm{rank}.txt"), "w") as f:
+        f.write(f"{calls[0]} {elapsed:.4f} {len(lat)}")
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+class TestBenchWarmupAgreement:
+    def test_ranks_agree_on_extension_count(self, tmp_path):
+        """Regression: the SMI-visibility warmup extension must run the
+        SAME number of steps on every rank (steps contain collectives)."""
+        import socket
+        import torch.multiprocessing as mp
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+        ctx = mp.get_context("spawn")
+        ps = [ctx.Process(target=_bench_warmup_worker,
+                          args=(r, 2, port, str(tmp_path)))
+              for r in range(2)]
+        for p in ps:
+            p.start()
+        for p in ps:
+            p.join(120)
+            assert p.exitcode == 0, p.exitcode
+        c0, e0, l0 = open(tmp_path / "warm0.txt").read().split()
+        c1, e1, l1 = open(tmp_path / "warm1.txt").read().split()
+        assert c0 == c1          # identical collective counts
+        assert l0 == l1 == "3"   # timed steps as contracted
+        assert int(c0) > 5       # warmup actually extended past W=2
