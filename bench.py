@@ -109,9 +109,9 @@ def main() -> None:
     p.add_argument("--variant", default="MyCNN5")
     p.add_argument("--overlap", action="store_true",
                    help="[pipeline] two-stream ingest/model software "
-                        "pipelining (measured NEUTRAL: both stages fill "
-                        "all 256 CUs, so streams cannot co-schedule — "
-                        "kept as an A/B reference)")
+                        "pipelining (measured NEUTRAL in r1 at full grids "
+                        "and NULL in the r2 CU-partition sweep with capped "
+                        "ingest grids — kept as an A/B reference)")
     p.add_argument("--graph", action="store_true", default=True,
                    help="hipGraph-capture the model forward (pipeline mode)")
     p.add_argument("--no-graph", dest="graph", action="store_false")
