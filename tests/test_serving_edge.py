@@ -220,3 +220,21 @@ class TestBenchWarmupAgreement:
         assert c0 == c1          # identical collective counts
         assert l0 == l1 == "3"   # timed steps as contracted
         assert int(c0) > 5       # warmup actually extended past W=2
+
+
+class TestSignalSurvey:
+    def test_survey_committed_record(self, capsys):
+        """Signal-availability survey (explore_fantasia analog) over the
+        committed real p000194 numerics record."""
+        from tskd_amd.cli.surveysignals import main
+        main(["--wavef-path", os.path.join(REPO, "data", "waveform"),
+              "--json"])
+        rec = json.loads(capsys.readouterr().out)
+        assert rec["n_records"] == 1
+        assert rec["channel_counts"]["HR"] == 1
+        r = rec["records"]["p000194-2112-05-23-14-34n"]
+        assert r["sig_len"] == 1625
+        assert abs(r["fs_hz"] - 1 / 60) < 1e-6
+        # the reference's channel-name mismatch shows up here: record says
+        # NBPSys, config says "NBP Sys"
+        assert "NBPSys" in r["channels"] and r["channels"]["HR"] > 0
