@@ -235,6 +235,8 @@ def main() -> None:
     ap.add_argument("--json", default=None)
     ap.add_argument("--quick", action="store_true",
                     help="small shapes (CPU CI smoke)")
+    ap.add_argument("--big-live", action="store_true",
+                    help="one extra live scenario at 16384 patients")
     args = ap.parse_args()
     dev = args.device
     out = []
@@ -251,6 +253,10 @@ def main() -> None:
                                  n_triggers=120, trigger_period_s=0.25))
         out.append(scenario_live(dev, n_patients=1024, rate_eps=100_000,
                                  n_triggers=120, trigger_period_s=0.25))
+        if args.big_live:
+            out.append(scenario_live(dev, n_patients=16384,
+                                     rate_eps=100_000, n_triggers=120,
+                                     trigger_period_s=0.25))
     for rec in out:
         print(json.dumps(rec), flush=True)
     if args.json:
