@@ -204,11 +204,27 @@ class FusedServer:
             if np_new <= 0 or not self.se.ready or not self.pids:
                 return 0
             dtype = torch.bfloat16 if self.device != "cpu" else torch.float32
-            w = self.se.windows(batch=1, stride=12, dtype=dtype,
-                                timelast=self.device != "cpu")
             n_active = len(self.pids)
-            age = torch.tensor([[self.ages.get(p)] for p in self.pids],
-                               device=w.device)
+            # cached device buffers: the age column only changes when the
+            # patient set grows, and the gather kernel overwrites every
+            # window element — rebuilding these per trigger cost ~13 ms of
+            # Python at 16k live patients (serving_latency_big.json)
+            if getattr(self, "_age_n", 0) != n_active:
+                self._age_dev = torch.tensor(
+                    [[self.ages.get(p)] for p in self.pids],
+                    device=self.se.device)
+                self._age_n = n_active
+            if getattr(self, "_win_buf", None) is None or \
+                    self._win_buf.dtype != dtype:
+                tl = self.device != "cpu"
+                shape = (self.se.S, 1, self.se.model_win, self.se.C) if tl \
+                    else (self.se.S, 1, self.se.C, self.se.model_win)
+                self._win_buf = torch.zeros(shape, dtype=dtype,
+                                            device=self.se.device)
+            w = self.se.windows(batch=1, stride=12, dtype=dtype,
+                                timelast=self.device != "cpu",
+                                out=self._win_buf)
+            age = self._age_dev
             probs_all = self.me.forward(w[:n_active].contiguous()
                                         if n_active < self.se.S else w,
                                         age, apply_sigmoid=True)
