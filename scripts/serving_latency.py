@@ -140,11 +140,14 @@ def scenario_drain(device: str, n_patients: int, n_events: int) -> dict:
 
 
 def scenario_live(device: str, n_patients: int, rate_eps: int,
-                  n_triggers: int, trigger_period_s: float) -> dict:
+                  n_triggers: int, trigger_period_s: float,
+                  poll_thread: bool = False) -> dict:
     """Producer thread at a target event rate with event-time = wall time;
     server triggers on a cadence; wire->store latency per trigger."""
     with tempfile.TemporaryDirectory() as tmp:
         bus, store, srv, cfg = make_server(tmp, device, n_patients)
+        if poll_thread:
+            srv.start_poll_thread()
         topics = [cfg.topic_for_channel(c) for c in cfg.channel_names[:8]]
         for t in topics:
             bus.create_topic(t)
@@ -198,6 +201,7 @@ def scenario_live(device: str, n_patients: int, rate_eps: int,
                 time.sleep(dt)
         stop.set()
         th.join(timeout=2)
+        srv.stop_poll_thread()
         lats = [x for x in lats if x >= 0]
         lats.sort()
 
@@ -206,6 +210,7 @@ def scenario_live(device: str, n_patients: int, rate_eps: int,
                 else None
         return {
             "scenario": "live", "device": device,
+            "poll_thread": poll_thread,
             "n_patients": n_patients, "target_rate_eps": rate_eps,
             "produced_events": produced[0],
             "n_triggers": n_triggers,
@@ -253,10 +258,17 @@ def main() -> None:
                                  n_triggers=120, trigger_period_s=0.25))
         out.append(scenario_live(dev, n_patients=1024, rate_eps=100_000,
                                  n_triggers=120, trigger_period_s=0.25))
+        out.append(scenario_live(dev, n_patients=1024, rate_eps=100_000,
+                                 n_triggers=120, trigger_period_s=0.25,
+                                 poll_thread=True))
         if args.big_live:
             out.append(scenario_live(dev, n_patients=16384,
                                      rate_eps=100_000, n_triggers=120,
                                      trigger_period_s=0.25))
+            out.append(scenario_live(dev, n_patients=16384,
+                                     rate_eps=100_000, n_triggers=120,
+                                     trigger_period_s=0.25,
+                                     poll_thread=True))
     for rec in out:
         print(json.dumps(rec), flush=True)
     if args.json:

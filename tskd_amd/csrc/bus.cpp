@@ -461,6 +461,10 @@ public:
     }
 
     std::vector<Message> poll(int max_msgs = 256, int timeout_ms = 0) {
+        // pure C++ until the pybind wrapper converts the return value:
+        // release the GIL for the whole scan so a background poll thread
+        // (serve.py start_poll_thread) truly overlaps Python/GPU work
+        py::gil_scoped_release rel_scan;
         std::vector<Message> out;
         const int64_t deadline = now_us() + (int64_t)timeout_ms * 1000;
         while (true) {
@@ -485,10 +489,7 @@ public:
                 if ((int)out.size() >= max_msgs) return out;
             }
             if (!out.empty() || now_us() >= deadline) return out;
-            {
-                py::gil_scoped_release rel;
-                usleep(1000);
-            }
+            usleep(1000);  // GIL already released above
         }
     }
 
@@ -539,30 +540,34 @@ public:
         std::vector<double> tss;
         std::vector<std::pair<std::string, int>> new_keys;
         sids.reserve(msgs.size());
-        for (auto& m : msgs) {
-            long chan;
-            double val;
-            if (!parse_sample(m.val_.data(), m.val_.size(), &chan, &val))
-                continue;
-            if (world > 1 &&
-                (int)(fnv1a(m.key_) % (uint64_t)world) != rank)
-                continue;  // another rank's patient
-            auto it = sid_.find(m.key_);
-            int sid;
-            if (it == sid_.end()) {
-                if (max_streams > 0 && (int)sid_.size() >= max_streams)
-                    throw std::runtime_error("max_streams exceeded: " +
-                                             m.key_);
-                sid = (int)sid_.size();
-                sid_.emplace(m.key_, sid);
-                new_keys.emplace_back(m.key_, sid);
-            } else {
-                sid = it->second;
+        {
+            py::gil_scoped_release rel_parse;  // pure C++ parse/map pass
+            for (auto& m : msgs) {
+                long chan;
+                double val;
+                if (!parse_sample(m.val_.data(), m.val_.size(), &chan,
+                                  &val))
+                    continue;
+                if (world > 1 &&
+                    (int)(fnv1a(m.key_) % (uint64_t)world) != rank)
+                    continue;  // another rank's patient
+                auto it = sid_.find(m.key_);
+                int sid;
+                if (it == sid_.end()) {
+                    if (max_streams > 0 && (int)sid_.size() >= max_streams)
+                        throw std::runtime_error("max_streams exceeded: " +
+                                                 m.key_);
+                    sid = (int)sid_.size();
+                    sid_.emplace(m.key_, sid);
+                    new_keys.emplace_back(m.key_, sid);
+                } else {
+                    sid = it->second;
+                }
+                sids.push_back(sid);
+                chans.push_back((int)chan);
+                vals.push_back((float)val);
+                tss.push_back((double)m.ts_us / 1e6);
             }
-            sids.push_back(sid);
-            chans.push_back((int)chan);
-            vals.push_back((float)val);
-            tss.push_back((double)m.ts_us / 1e6);
         }
         auto sa = py::array_t<int>((py::ssize_t)sids.size());
         auto ca = py::array_t<int>((py::ssize_t)chans.size());
