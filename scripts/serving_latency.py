@@ -245,32 +245,34 @@ def main() -> None:
     args = ap.parse_args()
     dev = args.device
     out = []
+    def run(r):
+        out.append(r)
+        print(json.dumps(r), flush=True)  # stream results scenario-by-scenario
+        return r
+
     if args.quick:
-        out.append(scenario_drain(dev, n_patients=32, n_events=20_000))
+        run(scenario_drain(dev, n_patients=32, n_events=20_000))
         # >= 30 triggers x 0.25 s: the measurement window must span at
         # least one 5-s bucket boundary or no trigger produces new grid
         # points (and therefore no predictions) — phase-dependent flake
-        out.append(scenario_live(dev, n_patients=32, rate_eps=2_000,
-                                 n_triggers=30, trigger_period_s=0.25))
+        run(scenario_live(dev, n_patients=32, rate_eps=2_000,
+                          n_triggers=30, trigger_period_s=0.25))
     else:
-        out.append(scenario_drain(dev, n_patients=1024, n_events=2_000_000))
-        out.append(scenario_live(dev, n_patients=1024, rate_eps=20_000,
-                                 n_triggers=120, trigger_period_s=0.25))
-        out.append(scenario_live(dev, n_patients=1024, rate_eps=100_000,
-                                 n_triggers=120, trigger_period_s=0.25))
-        out.append(scenario_live(dev, n_patients=1024, rate_eps=100_000,
-                                 n_triggers=120, trigger_period_s=0.25,
-                                 poll_thread=True))
+        run(scenario_drain(dev, n_patients=1024, n_events=2_000_000))
+        run(scenario_live(dev, n_patients=1024, rate_eps=20_000,
+                          n_triggers=120, trigger_period_s=0.25))
+        run(scenario_live(dev, n_patients=1024, rate_eps=100_000,
+                          n_triggers=120, trigger_period_s=0.25))
+        run(scenario_live(dev, n_patients=1024, rate_eps=100_000,
+                          n_triggers=120, trigger_period_s=0.25,
+                          poll_thread=True))
         if args.big_live:
-            out.append(scenario_live(dev, n_patients=16384,
-                                     rate_eps=100_000, n_triggers=120,
-                                     trigger_period_s=0.25))
-            out.append(scenario_live(dev, n_patients=16384,
-                                     rate_eps=100_000, n_triggers=120,
-                                     trigger_period_s=0.25,
-                                     poll_thread=True))
-    for rec in out:
-        print(json.dumps(rec), flush=True)
+            run(scenario_live(dev, n_patients=16384,
+                              rate_eps=100_000, n_triggers=120,
+                              trigger_period_s=0.25))
+            run(scenario_live(dev, n_patients=16384,
+                              rate_eps=100_000, n_triggers=120,
+                              trigger_period_s=0.25, poll_thread=True))
     if args.json:
         with open(args.json, "w") as f:
             json.dump(out, f, indent=1)
