@@ -3,6 +3,7 @@
 import json
 import multiprocessing as mp
 import os
+import time
 
 import pytest
 
@@ -265,6 +266,38 @@ class TestConcurrentStress:
         for m in c.poll(max_msgs=1024, timeout_ms=100):
             got.append(int(m.value))
         assert got == list(range(300))
+
+
+class TestConcurrentGrowAndPoll:
+    def test_grow_remap_under_gil_free_poll(self, bus):
+        """Regression (round 2): poll releases the GIL, so a producer
+        thread can grow+remap the partition mmap WHILE the consumer is
+        mid-scan — without PartMap's in-process op lock this segfaulted
+        (observed once on a GPU box). 8 KB values force several 2x file
+        growths past the 1 MiB initial mapping during a live poll loop."""
+        import threading
+        bus.create_topic("big")
+        p = Producer(bus)
+        c = Consumer(bus, starting="earliest")
+        c.subscribe(["big"])
+        N = 1200
+        payload = "x" * 8192
+
+        def producer():
+            for i in range(N):
+                p.produce("big", f"k{i % 7}", f"[{i % 10}, {float(i)}]"
+                          + payload[: 8000], ts_us=i + 1)
+
+        th = threading.Thread(target=producer)
+        th.start()
+        got = 0
+        deadline = time.time() + 60
+        while got < N and time.time() < deadline:
+            sa, ca, va, ta, nk = c.poll_samples_sid(max_msgs=4096,
+                                                    timeout_ms=50)
+            got += len(sa)
+        th.join()
+        assert got == N  # every message survived concurrent growth
 
 
 class TestTcpRelay:

@@ -115,12 +115,23 @@ public:
     PartHeader* hdr() { return reinterpret_cast<PartHeader*>(base_); }
     uint8_t* data() { return static_cast<uint8_t*>(base_) + HDR_SIZE; }
 
+    // Intra-process serialization note: maybe_remap() munmaps the old
+    // mapping, so every method that touches base_ takes op_mtx_. Before
+    // round 2 the Python GIL provided this serialization implicitly;
+    // poll() now releases the GIL (background poll thread), so producer
+    // and consumer threads of ONE process can be inside the same PartMap
+    // concurrently — without this lock a grow-triggered remap in append()
+    // unmaps the region a reader is dereferencing (observed as a flaky
+    // segfault in the live serving study). Cross-process safety is
+    // unchanged (process-shared robust mutex + atomic committed counter).
     uint64_t committed() {
+        std::lock_guard<std::mutex> g(op_mtx_);
         maybe_remap();
         return hdr()->committed.load(std::memory_order_acquire);
     }
 
     uint64_t trimmed() {
+        std::lock_guard<std::mutex> g(op_mtx_);
         maybe_remap();
         return hdr()->trimmed.load(std::memory_order_acquire);
     }
@@ -132,6 +143,8 @@ public:
     // Offsets stay ABSOLUTE and stable; reads below the trim point are a
     // consumer-visible gap (Kafka-retention semantics).
     uint64_t trim(uint64_t before_off) {
+        std::lock_guard<std::mutex> g(op_mtx_);
+        maybe_remap();
         auto* h = hdr();
         int rc = pthread_mutex_lock(&h->mtx);
         if (rc == EOWNERDEAD) pthread_mutex_consistent(&h->mtx);
@@ -159,6 +172,8 @@ public:
     }
 
     void append(const std::string& key, const std::string& val, int64_t ts_us) {
+        std::lock_guard<std::mutex> g(op_mtx_);
+        maybe_remap();
         auto* h = hdr();
         int rc = pthread_mutex_lock(&h->mtx);
         if (rc == EOWNERDEAD) pthread_mutex_consistent(&h->mtx);
@@ -187,7 +202,10 @@ public:
     // Read one message at byte offset; returns next offset or 0 if none.
     bool read(uint64_t off, std::string& key, std::string& val, uint64_t& seq,
               int64_t& ts_us, uint64_t& next) {
-        const uint64_t lim = committed();
+        std::lock_guard<std::mutex> g(op_mtx_);
+        maybe_remap();
+        const uint64_t lim =
+            hdr()->committed.load(std::memory_order_acquire);
         if (off + sizeof(MsgHeader) > lim) return false;
         // below the retention trim point: caller must skip forward
         if (off < hdr()->trimmed.load(std::memory_order_acquire)) return false;
@@ -207,7 +225,10 @@ public:
         return true;
     }
 
-    void sync() { msync(base_, mapped_, MS_SYNC); }
+    void sync() {
+        std::lock_guard<std::mutex> g(op_mtx_);
+        msync(base_, mapped_, MS_SYNC);
+    }
 
 private:
     void remap() {
@@ -232,6 +253,7 @@ private:
     void* base_ = nullptr;
     size_t mapped_ = 0;
     off_t file_size_ = 0;
+    std::mutex op_mtx_;  // in-process serialization (see committed())
 };
 
 uint64_t fnv1a(const std::string& s) {
