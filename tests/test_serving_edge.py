@@ -290,3 +290,39 @@ class TestPollThread:
         assert results["inline"]["pids"] == results["thread"]["pids"]
         assert results["inline"]["n"] > 0
         assert results["inline"]["preds"] == results["thread"]["preds"]
+
+
+class TestServeCliMain:
+    def test_main_poll_thread_flow(self, tmp_path, monkeypatch):
+        """serve.main() end-to-end through the CLI arg path: tmp bus with a
+        replayed backlog, --poll-thread, --max-triggers, prediction store
+        written, clean shutdown (poll thread joined, deferred flush)."""
+        import time
+
+        from tskd_amd.cli import serve
+        from tskd_amd.config import get_global_config
+        from tskd_amd.store import PredictionStore
+        cfg = get_global_config()
+        bus = Bus(str(tmp_path / "bus"))
+        topics = [cfg.topic_for_channel(c) for c in cfg.channel_names[:4]]
+        for t in topics:
+            bus.create_topic(t)
+        prod = Producer(bus)
+        for step in range(200):
+            for pi in range(4):
+                for ch in range(4):
+                    prod.produce(topics[ch], f"p{pi:06d}",
+                                 json.dumps([ch, 60.0 + step % 7]),
+                                 ts_us=int(step * 5e6))
+        store_path = str(tmp_path / "pred.log")
+        serve.main(["--bus-dir", str(tmp_path / "bus"),
+                    "--store-path", store_path,
+                    "--model-path", str(tmp_path / "nonexistent.pth"),
+                    "--device", "cpu", "--max-streams", "8",
+                    "--starting", "earliest", "--poll-thread",
+                    "--pipelined", "--speed", "1e6",
+                    "--max-triggers", "30"])
+        store = PredictionStore(store_path)
+        assert store.count() > 0
+        pids = {r[0] for r in store.tail(100)}
+        assert pids == {f"p{i:06d}" for i in range(4)}
