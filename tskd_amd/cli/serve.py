@@ -114,17 +114,23 @@ class FusedServer:
 
     def start_poll_thread(self, interval_s: float = 0.01) -> None:
         import threading as _threading
+        self._poll_error: Optional[BaseException] = None
 
         def loop():
-            while not self._poll_stop.is_set():
-                chunk = self.consumer.poll_samples_sid(
-                    max_msgs=131072, timeout_ms=0, rank=self.rank,
-                    world=self.world, max_streams=self.max_streams)
-                if len(chunk[0]) or chunk[4]:
-                    with self._poll_lock:
-                        self._poll_pending.append(chunk)
-                else:
-                    self._poll_stop.wait(interval_s)
+            try:
+                while not self._poll_stop.is_set():
+                    chunk = self.consumer.poll_samples_sid(
+                        max_msgs=131072, timeout_ms=0, rank=self.rank,
+                        world=self.world, max_streams=self.max_streams)
+                    if len(chunk[0]) or chunk[4]:
+                        with self._poll_lock:
+                            self._poll_pending.append(chunk)
+                    else:
+                        self._poll_stop.wait(interval_s)
+            except BaseException as e:  # surface at the next trigger —
+                # a silently-dead poll thread would stall serving
+                self._poll_error = e
+                log.error("poll thread died: %s", e)
 
         self._poll_thread = _threading.Thread(target=loop, daemon=True)
         self._poll_thread.start()
@@ -142,6 +148,8 @@ class FusedServer:
             return self.consumer.poll_samples_sid(
                 max_msgs=131072, timeout_ms=0, rank=self.rank,
                 world=self.world, max_streams=self.max_streams)
+        if getattr(self, "_poll_error", None) is not None:
+            raise RuntimeError("poll thread died") from self._poll_error
         with self._poll_lock:
             chunks, self._poll_pending = self._poll_pending, []
         if not chunks:
